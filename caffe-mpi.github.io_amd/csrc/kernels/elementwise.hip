@@ -1,0 +1,932 @@
+// elementwise.hip — gfx950 kernels for the HBM-bound hot-path ops:
+// im2col/col2im (batched, [K][Nimg*Spad] col layout), pooling, fused
+// BatchNorm (3 kernels/direction vs the reference's ~10-launch GEMV chain,
+// batch_norm_layer.hpp:96-120), ReLU, LRN, dropout, softmax+NLL, fused SGD
+// update (sgd_solver.cu:10-20 semantics), eltwise and synthetic-data fills.
+//
+// Design rules (cdna_hip_programming.md): grid-stride loops capped at
+// 2048 blocks ×256 threads (G11), coalesced unit-stride innermost access,
+// no atomics on the backward scatter paths (gather formulation instead —
+// deterministic, matching the reference's no-atomics choice,
+// conv_layer.cu:43-48).
+#include <hip/hip_runtime.h>
+
+#include "../math.hpp"
+
+namespace camd {
+namespace gpu {
+
+constexpr int TPB = 256;
+static inline int nblocks(long n, int per_thread = 1) {
+  long b = (n + (long)TPB * per_thread - 1) / ((long)TPB * per_thread);
+  return (int)std::min<long>(b, 2048);
+}
+#define GRID_STRIDE(i, n)                                        \
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < (n); \
+       i += (long)gridDim.x * blockDim.x)
+
+// ------------------------------------------------------------ im2col
+__global__ void k_im2col_b(const float* __restrict__ x, int Nimg, int C,
+                           int H, int W, int kh, int kw, int ph, int pw,
+                           int sh, int sw, int dh, int dw, int OH, int OW,
+                           long Spad, float* __restrict__ col) {
+  const long S = (long)OH * OW;
+  const long cols = (long)Nimg * Spad;
+  const long total = (long)C * kh * kw * cols;
+  GRID_STRIDE(idx, total) {
+    const long kck = idx / cols;        // row of col = (c, i, j)
+    const long cidx = idx - kck * cols; // column = n*Spad + sp
+    const int n = (int)(cidx / Spad);
+    const long sp = cidx - (long)n * Spad;
+    float v = 0.f;
+    if (sp < S) {
+      const int c = (int)(kck / (kh * kw));
+      const int ki = (int)(kck / kw) % kh;
+      const int kj = (int)(kck % kw);
+      const int oh = (int)(sp / OW), ow = (int)(sp % OW);
+      const int h = oh * sh - ph + ki * dh;
+      const int w = ow * sw - pw + kj * dw;
+      if (h >= 0 && h < H && w >= 0 && w < W)
+        v = x[(((long)n * C + c) * H + h) * W + w];
+    }
+    col[idx] = v;
+  }
+}
+
+void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
+                    int W, int kh, int kw, int ph, int pw, int sh, int sw,
+                    int dh, int dw, int OH, int OW, long Spad, float* col) {
+  const long total = (long)C * kh * kw * Nimg * Spad;
+  PerfScope perf("im2col", s, 0,
+                 8.0 * total);  // ~1 read + 1 write per element
+  hipLaunchKernelGGL(k_im2col_b, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
+                     x, Nimg, C, H, W, kh, kw, ph, pw, sh, sw, dh, dw, OH,
+                     OW, Spad, col);
+}
+
+// gather col2im (reference im2col.cu:256-295 pattern — no atomics)
+__global__ void k_col2im_b(const float* __restrict__ col, int Nimg, int C,
+                           int H, int W, int kh, int kw, int ph, int pw,
+                           int sh, int sw, int dh, int dw, int OH, int OW,
+                           long Spad, float* __restrict__ dx) {
+  const long total = (long)Nimg * C * H * W;
+  const long cols = (long)Nimg * Spad;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const int c = (int)((idx / ((long)W * H)) % C);
+    const int n = (int)(idx / ((long)W * H * C));
+    float acc = 0.f;
+    for (int i = 0; i < kh; ++i) {
+      int hk = h + ph - i * dh;
+      if (hk < 0 || hk % sh) continue;
+      hk /= sh;
+      if (hk >= OH) continue;
+      for (int j = 0; j < kw; ++j) {
+        int wk = w + pw - j * dw;
+        if (wk < 0 || wk % sw) continue;
+        wk /= sw;
+        if (wk >= OW) continue;
+        const long row = ((long)c * kh + i) * kw + j;
+        acc += col[row * cols + (long)n * Spad + (long)hk * OW + wk];
+      }
+    }
+    dx[idx] = acc;
+  }
+}
+
+void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
+                    int W, int kh, int kw, int ph, int pw, int sh, int sw,
+                    int dh, int dw, int OH, int OW, long Spad, float* dx) {
+  const long total = (long)Nimg * C * H * W;
+  PerfScope perf("col2im", s, 0, 8.0 * total * kh * kw / (sh * sw));
+  hipLaunchKernelGGL(k_col2im_b, dim3(nblocks(total, 2)), dim3(TPB), 0, s,
+                     dcol, Nimg, C, H, W, kh, kw, ph, pw, sh, sw, dh, dw,
+                     OH, OW, Spad, dx);
+}
+
+// y[N][C][S] -> out[C][N*Spad] (zero pad)
+__global__ void k_nchw_to_cpad(const float* __restrict__ y, int Nimg, int C,
+                               long S, long Spad, float* __restrict__ out) {
+  const long total = (long)C * Nimg * Spad;
+  GRID_STRIDE(idx, total) {
+    const long c = idx / ((long)Nimg * Spad);
+    const long rem = idx - c * (long)Nimg * Spad;
+    const int n = (int)(rem / Spad);
+    const long sp = rem - (long)n * Spad;
+    out[idx] = sp < S ? y[((long)n * C + c) * S + sp] : 0.f;
+  }
+}
+
+void nchw_to_cpad(hipStream_t s, const float* y, int Nimg, int C, long S,
+                  long Spad, float* out) {
+  const long total = (long)C * Nimg * Spad;
+  PerfScope perf("transpose", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_nchw_to_cpad, dim3(nblocks(total, 4)), dim3(TPB), 0,
+                     s, y, Nimg, C, S, Spad, out);
+}
+
+// ------------------------------------------------------------ relu
+__global__ void k_relu_fwd(const float* __restrict__ x, long n, float slope,
+                           float* __restrict__ y) {
+  GRID_STRIDE(i, n) {
+    const float v = x[i];
+    y[i] = v > 0.f ? v : slope * v;
+  }
+}
+void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y) {
+  PerfScope perf("relu", s, 0, 8.0 * n);
+  hipLaunchKernelGGL(k_relu_fwd, dim3(nblocks(n, 8)), dim3(TPB), 0, s, x, n,
+                     slope, y);
+}
+
+__global__ void k_relu_bwd(const float* __restrict__ x,
+                           const float* __restrict__ dy, long n, float slope,
+                           float* __restrict__ dx) {
+  GRID_STRIDE(i, n) {
+    dx[i] = dy[i] * (x[i] > 0.f ? 1.f : slope);
+  }
+}
+void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
+              float slope, float* dx) {
+  PerfScope perf("relu", s, 0, 12.0 * n);
+  hipLaunchKernelGGL(k_relu_bwd, dim3(nblocks(n, 8)), dim3(TPB), 0, s, x,
+                     dy, n, slope, dx);
+}
+
+// ------------------------------------------------------------ pooling
+__global__ void k_pool_max_fwd(const float* __restrict__ x, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
+                               int sh, int sw, int OH, int OW,
+                               float* __restrict__ y, int* __restrict__ mask) {
+  const long total = (long)N * C * OH * OW;
+  GRID_STRIDE(idx, total) {
+    const int ow = (int)(idx % OW);
+    const int oh = (int)((idx / OW) % OH);
+    const long nc = idx / ((long)OW * OH);
+    const float* xp = x + nc * H * W;
+    int hs = oh * sh - ph, ws = ow * sw - pw;
+    const int he = min(hs + kh, H), we = min(ws + kw, W);
+    hs = max(hs, 0);
+    ws = max(ws, 0);
+    float best = -3.402823466e38f;
+    int bi = -1;
+    for (int h = hs; h < he; ++h)
+      for (int w = ws; w < we; ++w) {
+        const float v = xp[h * W + w];
+        if (v > best) {  // strict >, first max wins (pooling_layer.cpp:166)
+          best = v;
+          bi = h * W + w;
+        }
+      }
+    y[idx] = best;
+    mask[idx] = bi;
+  }
+}
+void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* y, int* mask) {
+  const long total = (long)N * C * OH * OW;
+  PerfScope perf("pool", s, 0, 4.0 * total * (kh * kw + 2));
+  hipLaunchKernelGGL(k_pool_max_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y,
+                     mask);
+}
+
+// deterministic gather backward: each input element scans the <= ceil(k/s)^2
+// windows that can contain it and sums where mask points at it
+__global__ void k_pool_max_bwd(const float* __restrict__ dy,
+                               const int* __restrict__ mask, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
+                               int sh, int sw, int OH, int OW,
+                               float* __restrict__ dx) {
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int me = h * W + w;
+    const int ph0 = (h + ph < kh) ? 0 : (h + ph - kh) / sh + 1;
+    const int ph1 = min((h + ph) / sh + 1, OH);
+    const int pw0 = (w + pw < kw) ? 0 : (w + pw - kw) / sw + 1;
+    const int pw1 = min((w + pw) / sw + 1, OW);
+    float acc = 0.f;
+    const float* dyp = dy + nc * OH * OW;
+    const int* mp = mask + nc * OH * OW;
+    for (int a = ph0; a < ph1; ++a)
+      for (int b = pw0; b < pw1; ++b)
+        if (mp[a * OW + b] == me) acc += dyp[a * OW + b];
+    dx[idx] = acc;
+  }
+}
+void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
+                  int C, int H, int W, int kh, int kw, int ph, int pw,
+                  int sh, int sw, int OH, int OW, float* dx) {
+  const long total = (long)N * C * H * W;
+  PerfScope perf("pool", s, 0, 12.0 * total);
+  hipLaunchKernelGGL(k_pool_max_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, dy, mask, N, C, H, W, kh, kw, ph, pw, sh, sw, OH,
+                     OW, dx);
+}
+
+__global__ void k_pool_ave_fwd(const float* __restrict__ x, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
+                               int sh, int sw, int OH, int OW,
+                               float* __restrict__ y) {
+  const long total = (long)N * C * OH * OW;
+  GRID_STRIDE(idx, total) {
+    const int ow = (int)(idx % OW);
+    const int oh = (int)((idx / OW) % OH);
+    const long nc = idx / ((long)OW * OH);
+    const float* xp = x + nc * H * W;
+    int hs = oh * sh - ph, ws = ow * sw - pw;
+    int he = min(hs + kh, H + ph), we = min(ws + kw, W + pw);
+    const int ps = (he - hs) * (we - ws);  // padded size (:201)
+    hs = max(hs, 0);
+    ws = max(ws, 0);
+    he = min(he, H);
+    we = min(we, W);
+    float acc = 0.f;
+    for (int h = hs; h < he; ++h)
+      for (int w = ws; w < we; ++w) acc += xp[h * W + w];
+    y[idx] = acc / ps;
+  }
+}
+void pool_ave_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* y) {
+  const long total = (long)N * C * OH * OW;
+  PerfScope perf("pool", s, 0, 4.0 * total * (kh * kw + 1));
+  hipLaunchKernelGGL(k_pool_ave_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y);
+}
+
+__global__ void k_pool_ave_bwd(const float* __restrict__ dy, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
+                               int sh, int sw, int OH, int OW,
+                               float* __restrict__ dx) {
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int ph0 = (h + ph < kh) ? 0 : (h + ph - kh) / sh + 1;
+    const int ph1 = min((h + ph) / sh + 1, OH);
+    const int pw0 = (w + pw < kw) ? 0 : (w + pw - kw) / sw + 1;
+    const int pw1 = min((w + pw) / sw + 1, OW);
+    float acc = 0.f;
+    const float* dyp = dy + nc * OH * OW;
+    for (int a = ph0; a < ph1; ++a)
+      for (int b = pw0; b < pw1; ++b) {
+        int hs = a * sh - ph, ws = b * sw - pw;
+        const int he = min(hs + kh, H + ph), we = min(ws + kw, W + pw);
+        const int ps = (he - hs) * (we - ws);
+        acc += dyp[a * OW + b] / ps;
+      }
+    dx[idx] = acc;
+  }
+}
+void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* dx) {
+  const long total = (long)N * C * H * W;
+  PerfScope perf("pool", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_pool_ave_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, dy, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
+}
+
+// ------------------------------------------------------------ batchnorm
+// partials layout: double2[C][nb] {sum, sumsq} (fwd) / {sum_dy, sum_dyxn}
+// (bwd).  Deterministic: fixed block→slice mapping, in-block tree reduce.
+int bn_blocks_per_channel(int N, long S) {
+  const long per_c = (long)N * S;
+  long nb = (per_c + TPB * 16 - 1) / (TPB * 16);
+  return (int)std::max<long>(1, std::min<long>(nb, 64));
+}
+
+__global__ void k_bn_fwd_stats(const float* __restrict__ x, int N, int C,
+                               long S, int nb, double2* __restrict__ out) {
+  const int c = blockIdx.x % C;
+  const int slice = blockIdx.x / C;
+  const long per_c = (long)N * S;
+  const long lo = per_c * slice / nb, hi = per_c * (slice + 1) / nb;
+  double s1 = 0, s2 = 0;
+  for (long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const int n = (int)(i / S);
+    const long sp = i - (long)n * S;
+    const double v = x[((long)n * C + c) * S + sp];
+    s1 += v;
+    s2 += v * v;
+  }
+  __shared__ double sh1[TPB], sh2[TPB];
+  sh1[threadIdx.x] = s1;
+  sh2[threadIdx.x] = s2;
+  __syncthreads();
+  for (int off = TPB / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sh1[threadIdx.x] += sh1[threadIdx.x + off];
+      sh2[threadIdx.x] += sh2[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
+}
+
+void bn_fwd_stats(hipStream_t s, const float* x, int N, int C, long S,
+                  int nb, void* partials) {
+  PerfScope perf("bn", s, 0, 4.0 * N * C * S);
+  hipLaunchKernelGGL(k_bn_fwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, N, C,
+                     S, nb, (double2*)partials);
+}
+
+__global__ void k_bn_fwd_finalize(const double2* __restrict__ partials,
+                                  int nb, int C, long NS, float eps,
+                                  float* __restrict__ mean,
+                                  float* __restrict__ var,
+                                  float* __restrict__ inv_std) {
+  GRID_STRIDE(c, (long)C) {
+    double s1 = 0, s2 = 0;
+    for (int b = 0; b < nb; ++b) {
+      s1 += partials[c * nb + b].x;
+      s2 += partials[c * nb + b].y;
+    }
+    const double m = s1 / NS;
+    const double v = s2 / NS - m * m;  // E[x^2]-m^2 in double ≈ E[(x-m)^2]
+    mean[c] = (float)m;
+    var[c] = (float)v;
+    inv_std[c] = (float)(1.0 / sqrt(v + (double)eps));
+  }
+}
+void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
+                     long NS, float eps, float* mean, float* var,
+                     float* inv_std) {
+  hipLaunchKernelGGL(k_bn_fwd_finalize, dim3(1), dim3(TPB), 0, s,
+                     (const double2*)partials, nb, C, NS, eps, mean, var,
+                     inv_std);
+}
+
+__global__ void k_bn_fwd_norm(const float* __restrict__ x,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ inv_std,
+                              const float* __restrict__ scale,
+                              const float* __restrict__ bias, int sb, int N,
+                              int C, long S, float* __restrict__ y) {
+  const long total = (long)N * C * S;
+  GRID_STRIDE(i, total) {
+    const int c = (int)((i / S) % C);
+    const float v = (x[i] - mean[c]) * inv_std[c];
+    y[i] = sb ? v * scale[c] + bias[c] : v;
+  }
+}
+void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
+                 const float* inv_std, const float* scale, const float* bias,
+                 int sb, int N, int C, long S, float* y) {
+  const long total = (long)N * C * S;
+  PerfScope perf("bn", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_bn_fwd_norm, dim3(nblocks(total, 8)), dim3(TPB), 0,
+                     s, x, mean, inv_std, scale, bias, sb, N, C, S, y);
+}
+
+__global__ void k_bn_moving_avg(const float* __restrict__ mean,
+                                const float* __restrict__ var, int C,
+                                float maf, int copy_only,
+                                float* __restrict__ gmean,
+                                float* __restrict__ gvar) {
+  GRID_STRIDE(c, (long)C) {
+    if (copy_only) {
+      gmean[c] = mean[c];
+      gvar[c] = var[c];
+    } else {  // batch_norm_layer.cpp:200-207
+      gmean[c] = (1.f - maf) * mean[c] + maf * gmean[c];
+      gvar[c] = (1.f - maf) * var[c] + maf * gvar[c];
+    }
+  }
+}
+void bn_moving_avg(hipStream_t s, const float* mean, const float* var, int C,
+                   float maf, int copy_only, float* gmean, float* gvar) {
+  hipLaunchKernelGGL(k_bn_moving_avg, dim3(1), dim3(TPB), 0, s, mean, var,
+                     C, maf, copy_only, gmean, gvar);
+}
+
+__global__ void k_bn_fwd_test(const float* __restrict__ x,
+                              const float* __restrict__ gmean,
+                              const float* __restrict__ gvar,
+                              const float* __restrict__ scale,
+                              const float* __restrict__ bias, int sb, int N,
+                              int C, long S, float eps,
+                              float* __restrict__ y) {
+  const long total = (long)N * C * S;
+  GRID_STRIDE(i, total) {
+    const int c = (int)((i / S) % C);
+    const float inv = rsqrtf(gvar[c] + eps);
+    const float v = (x[i] - gmean[c]) * inv;
+    y[i] = sb ? v * scale[c] + bias[c] : v;
+  }
+}
+void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,
+                 const float* gvar, const float* scale, const float* bias,
+                 int sb, int N, int C, long S, float eps, float* y) {
+  const long total = (long)N * C * S;
+  PerfScope perf("bn", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_bn_fwd_test, dim3(nblocks(total, 8)), dim3(TPB), 0,
+                     s, x, gmean, gvar, scale, bias, sb, N, C, S, eps, y);
+}
+
+__global__ void k_bn_bwd_stats(const float* __restrict__ x,
+                               const float* __restrict__ dy,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std, int N,
+                               int C, long S, int nb,
+                               double2* __restrict__ out) {
+  const int c = blockIdx.x % C;
+  const int slice = blockIdx.x / C;
+  const long per_c = (long)N * S;
+  const long lo = per_c * slice / nb, hi = per_c * (slice + 1) / nb;
+  const float m = mean[c], inv = inv_std[c];
+  double s_dy = 0, s_dyxn = 0;
+  for (long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
+    const int n = (int)(i / S);
+    const long sp = i - (long)n * S;
+    const long off = ((long)n * C + c) * S + sp;
+    const double d = dy[off];
+    s_dy += d;
+    s_dyxn += d * (double)((x[off] - m) * inv);
+  }
+  __shared__ double sh1[TPB], sh2[TPB];
+  sh1[threadIdx.x] = s_dy;
+  sh2[threadIdx.x] = s_dyxn;
+  __syncthreads();
+  for (int off = TPB / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sh1[threadIdx.x] += sh1[threadIdx.x + off];
+      sh2[threadIdx.x] += sh2[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
+}
+void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
+                  const float* mean, const float* inv_std, int N, int C,
+                  long S, int nb, void* partials) {
+  PerfScope perf("bn", s, 0, 8.0 * N * C * S);
+  hipLaunchKernelGGL(k_bn_bwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, dy,
+                     mean, inv_std, N, C, S, nb, (double2*)partials);
+}
+
+__global__ void k_bn_bwd_finalize(const double2* __restrict__ partials,
+                                  int nb, int C, long NS,
+                                  const float* __restrict__ scale, int sb,
+                                  float* __restrict__ dscale,
+                                  float* __restrict__ dbias,
+                                  float* __restrict__ m_dy,
+                                  float* __restrict__ m_dyxn) {
+  GRID_STRIDE(c, (long)C) {
+    double s_dy = 0, s_dyxn = 0;
+    for (int b = 0; b < nb; ++b) {
+      s_dy += partials[c * nb + b].x;
+      s_dyxn += partials[c * nb + b].y;
+    }
+    if (sb) {
+      dscale[c] = (float)s_dyxn;  // Σ dy·x̂ (batch_norm_layer.cpp:322-330)
+      dbias[c] = (float)s_dy;     // Σ dy
+    }
+    const double sc = sb ? (double)scale[c] : 1.0;
+    m_dy[c] = (float)(sc * s_dy / NS);
+    m_dyxn[c] = (float)(sc * s_dyxn / NS);
+  }
+}
+void bn_bwd_finalize(hipStream_t s, const void* partials, int nb, int C,
+                     long NS, const float* scale, int sb, float* dscale,
+                     float* dbias, float* m_dy, float* m_dyxn) {
+  hipLaunchKernelGGL(k_bn_bwd_finalize, dim3(1), dim3(TPB), 0, s,
+                     (const double2*)partials, nb, C, NS, scale, sb, dscale,
+                     dbias, m_dy, m_dyxn);
+}
+
+__global__ void k_bn_bwd_apply(const float* __restrict__ x,
+                               const float* __restrict__ dy,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ inv_std,
+                               const float* __restrict__ scale, int sb,
+                               const float* __restrict__ m_dy,
+                               const float* __restrict__ m_dyxn, int N,
+                               int C, long S, float* __restrict__ dx) {
+  const long total = (long)N * C * S;
+  GRID_STRIDE(i, total) {
+    const int c = (int)((i / S) % C);
+    const float xn = (x[i] - mean[c]) * inv_std[c];
+    const float d = dy[i] * (sb ? scale[c] : 1.f);
+    dx[i] = (d - m_dy[c] - m_dyxn[c] * xn) * inv_std[c];
+  }
+}
+void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
+                  const float* mean, const float* inv_std,
+                  const float* scale, int sb, const float* m_dy,
+                  const float* m_dyxn, int N, int C, long S, float* dx) {
+  const long total = (long)N * C * S;
+  PerfScope perf("bn", s, 0, 12.0 * total);
+  hipLaunchKernelGGL(k_bn_bwd_apply, dim3(nblocks(total, 8)), dim3(TPB), 0,
+                     s, x, dy, mean, inv_std, scale, sb, m_dy, m_dyxn, N, C,
+                     S, dx);
+}
+
+// ------------------------------------------------------------ LRN
+// one thread per (n, s); running cross-channel window (lrn_layer.cu:9-60
+// shape, restated)
+__global__ void k_lrn_fwd(const float* __restrict__ x, int N, int C, long S,
+                          int size, float aos, float beta, float k,
+                          float* __restrict__ scale, float* __restrict__ y) {
+  const long total = (long)N * S;
+  const int pre = (size - 1) / 2;
+  GRID_STRIDE(idx, total) {
+    const int n = (int)(idx / S);
+    const long sp = idx - (long)n * S;
+    const float* xp = x + (long)n * C * S + sp;
+    float* scp = scale + (long)n * C * S + sp;
+    float* yp = y + (long)n * C * S + sp;
+    float acc = 0.f;
+    for (int c = 0; c < size - pre && c < C; ++c) {
+      const float v = xp[(long)c * S];
+      acc += v * v;
+    }
+    for (int c = 0; c < C; ++c) {
+      if (c > 0) {
+        const int head = c + size - 1 - pre;
+        if (head < C) {
+          const float v = xp[(long)head * S];
+          acc += v * v;
+        }
+        const int tail = c - 1 - pre;
+        if (tail >= 0) {
+          const float v = xp[(long)tail * S];
+          acc -= v * v;
+        }
+      }
+      const float sc = k + aos * acc;
+      scp[(long)c * S] = sc;
+      yp[(long)c * S] = xp[(long)c * S] * __powf(sc, -beta);
+    }
+  }
+}
+void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+             int size, float alpha, float beta, float k, float* scale,
+             float* y) {
+  const long S = (long)H * W;
+  PerfScope perf("lrn", s, 0, 12.0 * N * C * S);
+  hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, N,
+                     C, S, size, alpha / size, beta, k, scale, y);
+}
+
+__global__ void k_lrn_bwd(const float* __restrict__ x,
+                          const float* __restrict__ y,
+                          const float* __restrict__ dy,
+                          const float* __restrict__ scale, int N, int C,
+                          long S, int size, float cr, float beta,
+                          float* __restrict__ dx) {
+  const long total = (long)N * S;
+  const int pre = (size - 1) / 2;
+  GRID_STRIDE(idx, total) {
+    const int n = (int)(idx / S);
+    const long sp = idx - (long)n * S;
+    const long base = (long)n * C * S + sp;
+    // ratio(c) = dy*y/scale; window for dx[c]: cc in [c-(size-1-pre), c+pre]
+    float acc = 0.f;
+    const int lo0 = -(size - 1 - pre);
+    for (int cc = lo0; cc <= pre - 1; ++cc)
+      if (cc >= 0 && cc < C) {
+        const long i = base + (long)cc * S;
+        acc += dy[i] * y[i] / scale[i];
+      }
+    for (int c = 0; c < C; ++c) {
+      const int head = c + pre;
+      if (head >= 0 && head < C) {
+        const long i = base + (long)head * S;
+        acc += dy[i] * y[i] / scale[i];
+      }
+      const long i = base + (long)c * S;
+      dx[i] = dy[i] * __powf(scale[i], -beta) - cr * x[i] * acc;
+      const int tail = c + lo0;
+      if (tail >= 0 && tail < C) {
+        const long j = base + (long)tail * S;
+        acc -= dy[j] * y[j] / scale[j];
+      }
+    }
+  }
+}
+void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
+             const float* scale, int N, int C, int H, int W, int size,
+             float alpha, float beta, float* dx) {
+  const long S = (long)H * W;
+  PerfScope perf("lrn", s, 0, 20.0 * N * C * S);
+  hipLaunchKernelGGL(k_lrn_bwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, y,
+                     dy, scale, N, C, S, size, 2.f * alpha * beta / size,
+                     beta, dx);
+}
+
+// ------------------------------------------------------------ softmax
+// one wave per sample row (C up to 1000): wave-parallel max/sum reductions
+__global__ void k_softmax_fwd(const float* __restrict__ x, int outer, int C,
+                              int inner, float* __restrict__ y) {
+  const long rows = (long)outer * inner;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wpb = blockDim.x >> 6;
+  for (long r = (long)blockIdx.x * wpb + wave; r < rows;
+       r += (long)gridDim.x * wpb) {
+    const int o = (int)(r / inner);
+    const int sp = (int)(r - (long)o * inner);
+    const float* xp = x + (long)o * C * inner + sp;
+    float* yp = y + (long)o * C * inner + sp;
+    float mx = -3.402823466e38f;
+    for (int c = lane; c < C; c += 64)
+      mx = fmaxf(mx, xp[(long)c * inner]);
+    for (int off = 32; off > 0; off >>= 1)
+      mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+    float sum = 0.f;
+    for (int c = lane; c < C; c += 64) {
+      const float e = __expf(xp[(long)c * inner] - mx);
+      yp[(long)c * inner] = e;
+      sum += e;
+    }
+    for (int off = 32; off > 0; off >>= 1) sum += __shfl_xor(sum, off, 64);
+    const float inv = 1.f / sum;
+    for (int c = lane; c < C; c += 64) yp[(long)c * inner] *= inv;
+  }
+}
+void softmax_fwd(hipStream_t s, const float* x, int outer, int C, int inner,
+                 float* prob) {
+  const long rows = (long)outer * inner;
+  PerfScope perf("softmax", s, 0, 12.0 * rows * C);
+  const int blocks = (int)std::min<long>((rows + 3) / 4, 2048);
+  hipLaunchKernelGGL(k_softmax_fwd, dim3(blocks), dim3(TPB), 0, s, x, outer,
+                     C, inner, prob);
+}
+
+// two-stage deterministic loss sum: block partials then final add
+__global__ void k_sm_loss(const float* __restrict__ prob,
+                          const float* __restrict__ label, int outer, int C,
+                          int inner, float norm, float* __restrict__ loss) {
+  const long rows = (long)outer * inner;
+  double acc = 0;
+  for (long r = blockIdx.x * (long)blockDim.x + threadIdx.x; r < rows;
+       r += (long)gridDim.x * blockDim.x) {
+    const int o = (int)(r / inner);
+    const int sp = (int)(r - (long)o * inner);
+    const int lv = (int)label[(long)o * inner + sp];
+    const float p =
+        fmaxf(prob[((long)o * C + lv) * inner + sp], 1.175494351e-38f);
+    acc -= (double)__logf(p);
+  }
+  __shared__ double sh[TPB];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = TPB / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    // single block launch: write final
+    loss[blockIdx.x] = (float)(sh[0] / norm);
+  }
+}
+void softmaxloss_fwd(hipStream_t s, const float* prob, const float* label,
+                     int outer, int C, int inner, float* loss_out) {
+  PerfScope perf("softmax", s, 0, 8.0 * outer * inner);
+  hipLaunchKernelGGL(k_sm_loss, dim3(1), dim3(TPB), 0, s, prob, label,
+                     outer, C, inner, (float)((long)outer * inner),
+                     loss_out);
+}
+
+__global__ void k_sm_loss_bwd(const float* __restrict__ prob,
+                              const float* __restrict__ label, int outer,
+                              int C, int inner, float scale,
+                              float* __restrict__ dx) {
+  const long total = (long)outer * C * inner;
+  GRID_STRIDE(i, total) {
+    const long o = i / ((long)C * inner);
+    const long rem = i - o * C * inner;
+    const int c = (int)(rem / inner);
+    const int sp = (int)(rem - (long)c * inner);
+    const int lv = (int)label[o * inner + sp];
+    dx[i] = (prob[i] - (c == lv ? 1.f : 0.f)) * scale;
+  }
+}
+void softmaxloss_bwd(hipStream_t s, const float* prob, const float* label,
+                     int outer, int C, int inner, float scale, float* dx) {
+  const long total = (long)outer * C * inner;
+  PerfScope perf("softmax", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_sm_loss_bwd, dim3(nblocks(total, 4)), dim3(TPB), 0,
+                     s, prob, label, outer, C, inner, scale, dx);
+}
+
+// ------------------------------------------------------- row/col sums
+__global__ void k_rowsum(const float* __restrict__ A, long M, long N,
+                         float* __restrict__ out) {
+  for (long m = blockIdx.x; m < M; m += gridDim.x) {
+    const float* a = A + m * N;
+    double acc = 0;
+    for (long n = threadIdx.x; n < N; n += blockDim.x) acc += a[n];
+    __shared__ double sh[TPB];
+    sh[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = TPB / 2; off > 0; off >>= 1) {
+      if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) out[m] = (float)sh[0];
+    __syncthreads();
+  }
+}
+void rowsum(hipStream_t s, const float* A, long M, long N, float* out) {
+  PerfScope perf("reduce", s, 0, 4.0 * M * N);
+  hipLaunchKernelGGL(k_rowsum, dim3((int)std::min<long>(M, 2048)),
+                     dim3(TPB), 0, s, A, M, N, out);
+}
+
+__global__ void k_colsum(const float* __restrict__ A, long M, long N,
+                         float* __restrict__ out) {
+  GRID_STRIDE(n, N) {
+    double acc = 0;
+    for (long m = 0; m < M; ++m) acc += A[m * N + n];
+    out[n] = (float)acc;
+  }
+}
+void colsum(hipStream_t s, const float* A, long M, long N, float* out) {
+  PerfScope perf("reduce", s, 0, 4.0 * M * N);
+  hipLaunchKernelGGL(k_colsum, dim3(nblocks(N)), dim3(TPB), 0, s, A, M, N,
+                     out);
+}
+
+// ------------------------------------------------------------ eltwise etc.
+__global__ void k_axpby(long n, float a, const float* __restrict__ x,
+                        float b, float* __restrict__ y) {
+  GRID_STRIDE(i, n) y[i] = a * x[i] + b * y[i];
+}
+void axpby(hipStream_t s, long n, float a, const float* x, float b,
+           float* y) {
+  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  hipLaunchKernelGGL(k_axpby, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, a, x,
+                     b, y);
+}
+
+__global__ void k_copy(long n, const float* __restrict__ x,
+                       float* __restrict__ y) {
+  GRID_STRIDE(i, n) y[i] = x[i];
+}
+void copy(hipStream_t s, long n, const float* x, float* y) {
+  if (x == y) return;
+  PerfScope perf("eltwise", s, 0, 8.0 * n);
+  hipLaunchKernelGGL(k_copy, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, x, y);
+}
+
+__global__ void k_set(long n, float v, float* __restrict__ y) {
+  GRID_STRIDE(i, n) y[i] = v;
+}
+void set_const(hipStream_t s, long n, float v, float* y) {
+  hipLaunchKernelGGL(k_set, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, v, y);
+}
+
+__global__ void k_add3(long n, const float* __restrict__ a,
+                       const float* __restrict__ b, float* __restrict__ y) {
+  GRID_STRIDE(i, n) y[i] = a[i] + b[i];
+}
+void add3(hipStream_t s, long n, const float* a, const float* b, float* y) {
+  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  hipLaunchKernelGGL(k_add3, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, a, b,
+                     y);
+}
+
+__global__ void k_acc(long n, const float* __restrict__ x,
+                      float* __restrict__ y) {
+  GRID_STRIDE(i, n) y[i] += x[i];
+}
+void acc(hipStream_t s, long n, const float* x, float* y) {
+  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  hipLaunchKernelGGL(k_acc, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, x, y);
+}
+
+__global__ void k_concat_fwd(const float* __restrict__ x, int N, int Cs,
+                             long S, int Cd, int c_off,
+                             float* __restrict__ y) {
+  const long total = (long)N * Cs * S;
+  GRID_STRIDE(i, total) {
+    const int n = (int)(i / ((long)Cs * S));
+    const long rem = i - (long)n * Cs * S;
+    y[((long)n * Cd + c_off) * S + rem] = x[i];
+  }
+}
+void concat_fwd(hipStream_t s, const float* x, int N, int Cs, long S, int Cd,
+                int c_off, float* y) {
+  const long total = (long)N * Cs * S;
+  PerfScope perf("concat", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_concat_fwd, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
+                     x, N, Cs, S, Cd, c_off, y);
+}
+
+__global__ void k_concat_bwd(const float* __restrict__ dy, int N, int Cs,
+                             long S, int Cd, int c_off,
+                             float* __restrict__ dx) {
+  const long total = (long)N * Cs * S;
+  GRID_STRIDE(i, total) {
+    const int n = (int)(i / ((long)Cs * S));
+    const long rem = i - (long)n * Cs * S;
+    dx[i] = dy[((long)n * Cd + c_off) * S + rem];
+  }
+}
+void concat_bwd(hipStream_t s, const float* dy, int N, int Cs, long S,
+                int Cd, int c_off, float* dx) {
+  const long total = (long)N * Cs * S;
+  PerfScope perf("concat", s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_concat_bwd, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
+                     dy, N, Cs, S, Cd, c_off, dx);
+}
+
+// ------------------------------------------------------------ dropout/rng
+__device__ __forceinline__ uint64_t d_splitmix64(uint64_t x) {
+  x += 0x9E3779B97F4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+__device__ __forceinline__ float d_u01(uint64_t h) {
+  return (float)((h >> 11) * (1.0 / 9007199254740992.0));
+}
+
+__global__ void k_dropout_fwd(const float* __restrict__ x, long n,
+                              uint64_t key, float threshold, float scale,
+                              float* __restrict__ y,
+                              uint8_t* __restrict__ mask) {
+  GRID_STRIDE(i, n) {
+    const uint8_t m = d_u01(d_splitmix64(key ^ (uint64_t)i)) >= threshold;
+    mask[i] = m;
+    y[i] = x[i] * m * scale;
+  }
+}
+void dropout_fwd(hipStream_t s, const float* x, long n, uint64_t seed,
+                 uint64_t counter, float threshold, float scale, float* y,
+                 uint8_t* mask) {
+  PerfScope perf("dropout", s, 0, 9.0 * n);
+  (void)counter;  // caller passes the pre-mixed key via `seed`
+  hipLaunchKernelGGL(k_dropout_fwd, dim3(nblocks(n, 4)), dim3(TPB), 0, s, x,
+                     n, seed, threshold, scale, y, mask);
+}
+
+__global__ void k_dropout_bwd(const float* __restrict__ dy,
+                              const uint8_t* __restrict__ mask, long n,
+                              float scale, float* __restrict__ dx) {
+  GRID_STRIDE(i, n) dx[i] = dy[i] * mask[i] * scale;
+}
+void dropout_bwd(hipStream_t s, const float* dy, const uint8_t* mask, long n,
+                 float scale, float* dx) {
+  PerfScope perf("dropout", s, 0, 9.0 * n);
+  hipLaunchKernelGGL(k_dropout_bwd, dim3(nblocks(n, 4)), dim3(TPB), 0, s,
+                     dy, mask, n, scale, dx);
+}
+
+// ------------------------------------------------------------ SGD
+__global__ void k_sgd(long n, float* __restrict__ g, float* __restrict__ w,
+                      float* __restrict__ h, float mom, float lr,
+                      float decay, float gscale) {
+  GRID_STRIDE(i, n) {
+    float gi = g[i] * gscale + decay * w[i];
+    gi = h[i] = mom * h[i] + lr * gi;
+    w[i] -= gi;
+    g[i] = 0.f;
+  }
+}
+void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
+                float mom, float lr, float decay, float gscale) {
+  PerfScope perf("sgd", s, 0, 20.0 * n);
+  hipLaunchKernelGGL(k_sgd, dim3(nblocks(n, 4)), dim3(TPB), 0, s, n, g, w,
+                     h, mom, lr, decay, gscale);
+}
+
+// ------------------------------------------------------------ synthetic
+__global__ void k_fill_uniform(long n, uint64_t key, float lo, float hi,
+                               float* __restrict__ y) {
+  GRID_STRIDE(i, n)
+  y[i] = lo + (hi - lo) * d_u01(d_splitmix64(key ^ (uint64_t)i));
+}
+void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
+                  float lo, float hi, float* y) {
+  PerfScope perf("data", s, 0, 4.0 * n);
+  (void)counter;
+  hipLaunchKernelGGL(k_fill_uniform, dim3(nblocks(n, 4)), dim3(TPB), 0, s,
+                     n, seed, lo, hi, y);
+}
+
+__global__ void k_fill_labels(long n, uint64_t key, int classes,
+                              float* __restrict__ y) {
+  GRID_STRIDE(i, n)
+  y[i] = (float)(d_splitmix64(key ^ 0xABCDull ^ (uint64_t)i) %
+                 (uint64_t)classes);
+}
+void fill_labels(hipStream_t s, long n, uint64_t seed, uint64_t counter,
+                 int classes, float* y) {
+  (void)counter;
+  hipLaunchKernelGGL(k_fill_labels, dim3(nblocks(n)), dim3(TPB), 0, s, n,
+                     seed, classes, y);
+}
+
+}  // namespace gpu
+}  // namespace camd

@@ -1,0 +1,164 @@
+// math.hpp — CPU math for the engine's CPU mode and launcher declarations
+// for the hand-written gfx950 HIP kernels (implemented in kernels/*.hip).
+// Replaces the reference's util/math_functions.{cpp,cu} + util/im2col.* —
+// note we deliberately do NOT reproduce the reference's per-call stream
+// synchronization (math_functions.cu:26); all launches are async on the
+// caller's stream.
+#pragma once
+
+#include "core.hpp"
+
+namespace camd {
+
+// ------------------------------------------------------------- CPU math
+namespace cpu {
+// Row-major C[M][N] = alpha*op(A)*op(B) + beta*C (fp32 accumulation, blocked)
+void gemm(bool transA, bool transB, long M, long N, long K, float alpha,
+          const float* A, const float* B, float beta, float* C);
+void axpy(long n, float a, const float* x, float* y);
+void axpby(long n, float a, const float* x, float b, float* y);
+void scal(long n, float a, float* x);
+void im2col(const float* im, int C, int H, int W, int kh, int kw, int ph,
+            int pw, int sh, int sw, int dh, int dw, float* col);
+void col2im(const float* col, int C, int H, int W, int kh, int kw, int ph,
+            int pw, int sh, int sw, int dh, int dw, float* im);
+}  // namespace cpu
+
+inline int conv_out_dim(int in, int k, int pad, int stride, int dil) {
+  return (in + 2 * pad - (dil * (k - 1) + 1)) / stride + 1;
+}
+void pool_out_dim(int H, int W, int kh, int kw, int ph, int pw, int sh,
+                  int sw, int* OH, int* OW);  // ceil + clip, pooling_layer.cpp:86
+
+// ------------------------------------------------------------ GPU launchers
+namespace gpu {
+
+// Epilogue for the MFMA GEMM (bias/ReLU fusion + the conv NCHW scatter that
+// replaces a separate bias/copy pass).
+struct GemmEpi {
+  // when spad > 0: C column c maps to image n = c / spad, pixel s = c % spad;
+  // columns with s >= S are padding and are dropped; output element is
+  // C[n * n_stride + row * S + s] (NCHW scatter; the caller pre-offsets C by
+  // the group's channel base and passes n_stride = Cout_total * S).
+  long spad = 0;
+  long S = 0;
+  long n_stride = 0;
+  const float* bias = nullptr;
+  bool bias_per_col = false;  // IP: bias indexed by column, conv: by row
+  bool relu = false;
+};
+
+void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
+          float alpha, const float* A, long lda, const float* B, long ldb,
+          float beta, float* C, long ldc, const GemmEpi* epi = nullptr);
+
+// col[K][Nimg*Spad] from x[Nimg][C][H][W]; pad columns zero-filled.
+void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
+                    int W, int kh, int kw, int ph, int pw, int sh, int sw,
+                    int dh, int dw, int OH, int OW, long Spad, float* col);
+// dx[Nimg][C][H][W] from dcol[K][Nimg*Spad] (gather, deterministic)
+void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
+                    int W, int kh, int kw, int ph, int pw, int sh, int sw,
+                    int dh, int dw, int OH, int OW, long Spad, float* dx);
+// out[C][Nimg*Spad] from y[Nimg][C][S] (pad columns zeroed)
+void nchw_to_cpad(hipStream_t s, const float* y, int Nimg, int C, long S,
+                  long Spad, float* out);
+
+void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y);
+void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
+              float slope, float* dx);
+
+void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* y, int* mask);
+void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
+                  int C, int H, int W, int kh, int kw, int ph, int pw,
+                  int sh, int sw, int OH, int OW, float* dx);
+void pool_ave_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* y);
+void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
+                  int kh, int kw, int ph, int pw, int sh, int sw, int OH,
+                  int OW, float* dx);
+
+// Fused BatchNorm (replaces the reference's ~10-launch GEMV chain,
+// batch_norm_layer.hpp:96-120, with 3 kernels fwd / 3 bwd).
+// partials: double2[nb*C] workspace. mean/var/inv_std: float[C] device.
+int bn_blocks_per_channel(int N, long S);
+void bn_fwd_stats(hipStream_t s, const float* x, int N, int C, long S,
+                  int nb, void* partials);
+void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
+                     long NS, float eps, float* mean, float* var,
+                     float* inv_std);
+void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
+                 const float* inv_std, const float* scale, const float* bias,
+                 int scale_bias, int N, int C, long S, float* y);
+void bn_moving_avg(hipStream_t s, const float* mean, const float* var, int C,
+                   float maf, int copy_only, float* gmean, float* gvar);
+void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,
+                 const float* gvar, const float* scale, const float* bias,
+                 int scale_bias, int N, int C, long S, float eps, float* y);
+void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
+                  const float* mean, const float* inv_std, int N, int C,
+                  long S, int nb, void* partials);
+// writes dscale/dbias (if scale_bias) and the per-channel m_dy/m_dyxn terms
+void bn_bwd_finalize(hipStream_t s, const void* partials, int nb, int C,
+                     long NS, const float* scale, int scale_bias,
+                     float* dscale, float* dbias, float* m_dy,
+                     float* m_dyxn);
+void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
+                  const float* mean, const float* inv_std,
+                  const float* scale, int scale_bias, const float* m_dy,
+                  const float* m_dyxn, int N, int C, long S, float* dx);
+
+void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
+             int size, float alpha, float beta, float k, float* scale,
+             float* y);
+void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
+             const float* scale, int N, int C, int H, int W, int size,
+             float alpha, float beta, float* dx);
+
+void softmax_fwd(hipStream_t s, const float* x, int outer, int C, int inner,
+                 float* prob);
+// loss_out: device float[1]; norm = outer*inner (VALID, no ignore label)
+void softmaxloss_fwd(hipStream_t s, const float* prob, const float* label,
+                     int outer, int C, int inner, float* loss_out);
+void softmaxloss_bwd(hipStream_t s, const float* prob, const float* label,
+                     int outer, int C, int inner, float scale, float* dx);
+
+// out[m] = Σ_n A[m][n] (deterministic block reduce) — conv bias grad
+void rowsum(hipStream_t s, const float* A, long M, long N, float* out);
+// out[n] = Σ_m A[m][n] — IP bias grad
+void colsum(hipStream_t s, const float* A, long M, long N, float* out);
+
+void axpby(hipStream_t s, long n, float a, const float* x, float b, float* y);
+void copy(hipStream_t s, long n, const float* x, float* y);
+void set_const(hipStream_t s, long n, float v, float* y);
+void add3(hipStream_t s, long n, const float* a, const float* b, float* y);
+// y[n0..] accumulate: y += x
+void acc(hipStream_t s, long n, const float* x, float* y);
+
+// channel-block copy for Concat: src[N][Cs][S] <-> dst[N][Cd][S] at offset
+void concat_fwd(hipStream_t s, const float* x, int N, int Cs, long S,
+                int Cd, int c_off, float* y);
+void concat_bwd(hipStream_t s, const float* dy, int N, int Cs, long S,
+                int Cd, int c_off, float* dx);
+
+void dropout_fwd(hipStream_t s, const float* x, long n, uint64_t seed,
+                 uint64_t counter, float threshold, float scale, float* y,
+                 uint8_t* mask);
+void dropout_bwd(hipStream_t s, const float* dy, const uint8_t* mask, long n,
+                 float scale, float* dx);
+
+// fused SGD update (reference sgd_solver.cu:10-20 + the 1/nranks scale of
+// net.cpp:910): g = g*gscale + decay*w; h = mom*h + lr*g; w -= h; g = 0.
+void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
+                float mom, float lr, float decay, float gscale);
+
+void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
+                  float lo, float hi, float* y);
+void fill_labels(hipStream_t s, long n, uint64_t seed, uint64_t counter,
+                 int classes, float* y);
+
+}  // namespace gpu
+}  // namespace camd

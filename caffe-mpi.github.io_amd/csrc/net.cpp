@@ -1,0 +1,267 @@
+#include "net.hpp"
+
+#include <set>
+
+namespace camd {
+
+namespace {
+// does this layer message run in `phase`? (include/exclude NetStateRule
+// subset — only `phase` is used by the four model sets)
+bool layer_in_phase(const PMsgPtr& lm, Phase phase) {
+  const char* want = phase == Phase::TRAIN ? "TRAIN" : "TEST";
+  auto incs = lm->subs("include");
+  if (!incs.empty()) {
+    for (auto& r : incs)
+      if (r->str("phase") == want) return true;
+    return false;
+  }
+  for (auto& r : lm->subs("exclude"))
+    if (r->str("phase") == want) return false;
+  return true;
+}
+}  // namespace
+
+Net::Net(const PMsgPtr& net_param, Phase phase, int batch_override)
+    : phase_(phase) {
+  init(net_param, batch_override);
+}
+
+// reference src/caffe/util/insert_splits.cpp behaviour: any blob consumed by
+// more than one layer below its producer gets a Split layer so every
+// consumer owns a private diff that Split's backward sums.
+void Net::insert_splits(std::vector<PMsgPtr>& msgs) {
+  // count consumptions of each (versioned) blob name
+  std::map<std::string, int> last_producer;  // name -> producing msg idx
+  std::map<std::string, int> consumers;      // versioned key -> count
+  auto vkey = [&](const std::string& b) {
+    return b + "#" + std::to_string(last_producer.count(b)
+                                        ? last_producer[b]
+                                        : -1);
+  };
+  std::vector<std::vector<std::string>> bkeys(msgs.size());
+  for (size_t i = 0; i < msgs.size(); ++i) {
+    for (auto& b : msgs[i]->strs("bottom")) {
+      const std::string k = vkey(b);
+      consumers[k]++;
+      bkeys[i].push_back(k);
+    }
+    for (auto& t : msgs[i]->strs("top")) last_producer[t] = (int)i;
+  }
+  // rewire
+  std::map<std::string, int> split_next;  // key -> next split-top index
+  std::map<std::string, std::string> split_base;
+  std::vector<PMsgPtr> out;
+  last_producer.clear();
+  for (size_t i = 0; i < msgs.size(); ++i) {
+    // rewrite bottoms that consume split blobs
+    auto& m = msgs[i];
+    size_t bi = 0;
+    for (auto& f : m->fields) {
+      if (f.first != "bottom") continue;
+      const std::string k = bkeys[i][bi++];
+      auto it = split_base.find(k);
+      if (it != split_base.end())
+        f.second.scalar =
+            it->second + "_split_" + std::to_string(split_next[k]++);
+    }
+    out.push_back(m);
+    // after this layer, split any of its tops with >1 consumers
+    for (auto& t : m->strs("top")) {
+      last_producer[t] = (int)i;
+      const std::string k = t + "#" + std::to_string(i);
+      if (consumers[k] > 1) {
+        auto sp = std::make_shared<PMsg>();
+        auto addf = [&](const std::string& n, const std::string& v) {
+          PVal pv;
+          pv.kind = PVal::SCALAR;
+          pv.scalar = v;
+          sp->fields.emplace_back(n, pv);
+        };
+        addf("name", t + "_" + m->str("name") + "_split");
+        addf("type", "Split");
+        addf("bottom", t);
+        for (int c = 0; c < consumers[k]; ++c)
+          addf("top", t + "_split_" + std::to_string(c));
+        out.push_back(sp);
+        split_base[k] = t;
+        split_next[k] = 0;
+      }
+    }
+  }
+  msgs = std::move(out);
+}
+
+void Net::init(const PMsgPtr& msg, int batch_override) {
+  name_ = msg->str("name");
+  std::vector<PMsgPtr> lmsgs;
+  for (auto& lm : msg->subs("layer"))
+    if (layer_in_phase(lm, phase_)) lmsgs.push_back(lm);
+  CHECK_(!lmsgs.empty()) << "net has no layers for this phase";
+  insert_splits(lmsgs);
+
+  for (auto& lm : lmsgs) {
+    auto layer = create_layer(lm);
+    layer->set_phase(phase_);
+    // wire bottoms
+    std::vector<Blob*> bottom, top;
+    for (auto& b : lm->strs("bottom")) {
+      auto it = blob_map_.find(b);
+      CHECK_(it != blob_map_.end())
+          << "layer " << layer->name() << ": unknown bottom blob " << b;
+      bottom.push_back(it->second.get());
+    }
+    for (auto& t : lm->strs("top")) {
+      auto it = blob_map_.find(t);
+      if (it != blob_map_.end()) {
+        // in-place: top name equals an existing blob (must be a bottom)
+        top.push_back(it->second.get());
+      } else {
+        auto nb = std::make_shared<Blob>();
+        blob_map_[t] = nb;
+        top.push_back(nb.get());
+      }
+    }
+    if (batch_override > 0 && layer->type() == "Data") {
+      // per-rank batch override (reference divides the prototxt batch
+      // across GPUs, parallel.cpp:284-348)
+      layer->SetUp(bottom, top);
+      static_cast<DataLayer*>(layer.get())->batch_ = batch_override;
+      layer->Reshape(bottom, top);
+    } else {
+      layer->SetUp(bottom, top);
+    }
+    // loss weights
+    auto lw = lm->nums("loss_weight");
+    for (size_t i = 0; i < top.size(); ++i) {
+      float w = i < lw.size() ? (float)lw[i]
+                              : (i == 0 ? layer->default_loss_weight() : 0.f);
+      layer->set_loss((int)i, w);
+    }
+    layers_.push_back(layer);
+    bottoms_.push_back(bottom);
+    tops_.push_back(top);
+  }
+
+  // need-backward / propagate-down flags (net.cpp Init bottom-up logic)
+  std::map<Blob*, bool> blob_needs;
+  layer_need_bwd_.resize(layers_.size(), false);
+  prop_down_.resize(layers_.size());
+  for (size_t i = 0; i < layers_.size(); ++i) {
+    bool need = !layers_[i]->blobs().empty();
+    prop_down_[i].resize(bottoms_[i].size(), false);
+    for (size_t b = 0; b < bottoms_[i].size(); ++b) {
+      prop_down_[i][b] = blob_needs[bottoms_[i][b]];
+      need = need || prop_down_[i][b];
+    }
+    if (layers_[i]->type() == "Data" || layers_[i]->type() == "Accuracy")
+      need = false;
+    layer_need_bwd_[i] = need;
+    for (auto* t : tops_[i]) blob_needs[t] = need;
+  }
+
+  // learnable params in backward-completion order (reverse layer order)
+  for (int i = (int)layers_.size() - 1; i >= 0; --i) {
+    auto& lb = layers_[i]->blobs();
+    for (size_t j = 0; j < lb.size(); ++j) {
+      if (layers_[i]->skip_apply_update((int)j)) continue;
+      LParam p;
+      p.blob = lb[j].get();
+      p.layer = layers_[i].get();
+      p.blob_idx = (int)j;
+      p.lr_mult = layers_[i]->lr_mult((int)j);
+      p.decay_mult = layers_[i]->decay_mult((int)j);
+      p.count = p.blob->count();
+      p.offset = arena_count_;
+      arena_count_ += (long)padded(p.count);
+      params_.push_back(p);
+    }
+  }
+  if (Engine::get().mode == Mode::GPU) setup_arena();
+}
+
+void Net::setup_arena() {
+  if (params_.empty()) return;
+  Engine& E = Engine::get();
+  diff_arena_ =
+      (float*)E.dalloc.alloc(sizeof(float) * (size_t)arena_count_);
+  HIP_CHECK(hipMemsetAsync(diff_arena_, 0,
+                           sizeof(float) * (size_t)arena_count_, E.stream));
+  for (auto& p : params_)
+    p.blob->diff_mem().set_gpu_view(diff_arena_ + p.offset);
+  layer_events_.resize(layers_.size(), nullptr);
+  for (auto& ev : layer_events_)
+    HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+}
+
+void Net::Forward() {
+  for (size_t i = 0; i < layers_.size(); ++i)
+    layers_[i]->Forward(bottoms_[i], tops_[i]);
+}
+
+void Net::Backward(ReduceHook* hook) {
+  Engine& E = Engine::get();
+  // param ids grouped by layer, in backward order == ascending arena offset
+  size_t pidx = 0;
+  hipEvent_t last_ev = nullptr;
+  for (int i = (int)layers_.size() - 1; i >= 0; --i) {
+    if (layer_need_bwd_[i])
+      layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+    if (hook) {
+      // emit this layer's params (consecutive in params_, ascending offset)
+      const size_t start = pidx;
+      while (pidx < params_.size() &&
+             params_[pidx].layer == layers_[i].get())
+        ++pidx;
+      if (pidx > start) {
+        hipEvent_t ev = nullptr;
+        if (E.mode == Mode::GPU) {
+          ev = layer_events_[i];
+          HIP_CHECK(hipEventRecord(ev, E.stream));
+          last_ev = ev;
+        }
+        for (size_t k = start; k < pidx; ++k) hook->param_ready((int)k, ev);
+      }
+    }
+  }
+  if (hook) {
+    hipEvent_t ev = last_ev;
+    if (E.mode == Mode::GPU && !ev) {
+      ev = layer_events_[0];
+      HIP_CHECK(hipEventRecord(ev, E.stream));
+    }
+    hook->iteration_end(ev);
+  }
+}
+
+float Net::loss() {
+  Engine::get().sync();
+  float total = 0.f;
+  for (size_t i = 0; i < layers_.size(); ++i)
+    for (size_t t = 0; t < tops_[i].size(); ++t) {
+      const float w = layers_[i]->loss((int)t);
+      if (w != 0.f && layers_[i]->type() != "Accuracy")
+        total += w * tops_[i][t]->cpu_data()[0];
+    }
+  return total;
+}
+
+std::vector<std::string> Net::blob_names() const {
+  std::vector<std::string> out;
+  for (auto& kv : blob_map_) out.push_back(kv.first);
+  return out;
+}
+
+void Net::ShareTrainedLayersWith(Net& other) {
+  std::map<std::string, Layer*> by_name;
+  for (auto& l : other.layers_) by_name[l->name()] = l.get();
+  for (auto& l : layers_) {
+    auto it = by_name.find(l->name());
+    if (it == by_name.end()) continue;
+    auto& src = it->second->blobs();
+    auto& dst = l->blobs();
+    for (size_t i = 0; i < dst.size() && i < src.size(); ++i)
+      dst[i]->ShareData(*src[i]);
+  }
+}
+
+}  // namespace camd

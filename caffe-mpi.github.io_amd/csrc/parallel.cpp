@@ -1,0 +1,84 @@
+// parallel.cpp — RCCL communicator (the reference's NCCL call-site set is
+// tiny: ncclGetUniqueId/ncclCommInitRank parallel.cpp:42-45,:166-169,
+// ncclBcast :217-222, ncclAllReduce :233-239/:248-250, ncclCommDestroy :140).
+// One process per GPU; the unique id travels over torch.distributed's store
+// in bench.py (the reference used MPI_Bcast; single-node needs no MPI).
+#include <rccl/rccl.h>
+
+#include "solver.hpp"
+
+namespace camd {
+
+#define NCCL_CHECK(x)                                                  \
+  do {                                                                 \
+    ncclResult_t r_ = (x);                                             \
+    if (r_ != ncclSuccess)                                             \
+      CAMD_FATAL << "RCCL error: " << ncclGetErrorString(r_)           \
+                 << " at " #x;                                         \
+  } while (0)
+
+void rccl_unique_id(void* out) {
+  static_assert(sizeof(ncclUniqueId) == 128, "ncclUniqueId size");
+  ncclUniqueId id;
+  NCCL_CHECK(ncclGetUniqueId(&id));
+  memcpy(out, &id, sizeof(id));
+}
+
+namespace {
+
+class RcclComm : public Comm {
+ public:
+  RcclComm(int rank, int world, const void* uid) : world_(world) {
+    ncclUniqueId id;
+    memcpy(&id, uid, sizeof(id));
+    NCCL_CHECK(ncclCommInitRank(&comm_, world, id, rank));
+  }
+  ~RcclComm() override { ncclCommDestroy(comm_); }
+  void allreduce(float* buf, long count, hipStream_t s) override {
+    NCCL_CHECK(ncclAllReduce(buf, buf, (size_t)count, ncclFloat, ncclSum,
+                             comm_, s));
+  }
+  void bcast(float* buf, long count, int root, hipStream_t s) override {
+    NCCL_CHECK(
+        ncclBcast(buf, (size_t)count, ncclFloat, root, comm_, s));
+  }
+  int world() const override { return world_; }
+
+ private:
+  ncclComm_t comm_;
+  int world_;
+};
+
+class CallbackComm : public Comm {
+ public:
+  CallbackComm(HostAllreduceFn fn, void* ud, int world)
+      : fn_(fn), ud_(ud), world_(world) {}
+  void allreduce(float* buf, long count, hipStream_t) override {
+    fn_(buf, count, ud_);
+  }
+  void bcast(float* buf, long count, int /*root*/, hipStream_t) override {
+    // CPU test comm: callback semantics give every rank identical data via
+    // identical seeds; bcast is the identity here
+    (void)buf;
+    (void)count;
+  }
+  int world() const override { return world_; }
+
+ private:
+  HostAllreduceFn fn_;
+  void* ud_;
+  int world_;
+};
+
+}  // namespace
+
+std::unique_ptr<Comm> make_rccl_comm(int rank, int world, const void* uid) {
+  return std::unique_ptr<Comm>(new RcclComm(rank, world, uid));
+}
+
+std::unique_ptr<Comm> make_callback_comm(HostAllreduceFn fn, void* ud,
+                                         int world) {
+  return std::unique_ptr<Comm>(new CallbackComm(fn, ud, world));
+}
+
+}  // namespace camd

@@ -1,0 +1,202 @@
+#include "solver.hpp"
+
+#include <cmath>
+
+namespace camd {
+
+Solver::Solver(const PMsgPtr& sp, int batch_override) : param_(sp) {
+  PMsgPtr net_msg;
+  if (sp->has("net")) {
+    net_msg = parse_prototxt_file(sp->str("net"));
+  } else if (sp->sub("net_param")) {
+    net_msg = sp->sub("net_param");
+  } else {
+    CAMD_FATAL << "solver has no net";
+  }
+  // solver-pinned seed (models use random_seed; `caffe time` pins 1371)
+  Engine& E = Engine::get();
+  if (sp->has("random_seed")) {
+    E.seed = (uint64_t)sp->inum("random_seed");
+    E.cpu_rng.seed(E.seed + (uint64_t)E.rank);  // +rank: parallel.cpp:179-187
+  }
+  net_.reset(new Net(net_msg, Phase::TRAIN, batch_override));
+  weight_decay_ = (float)sp->num("weight_decay", 0.0);
+  const long nparams = (long)net_->learnable_params().size();
+  const long buckets = sp->inum("reduce_buckets", 6);  // caffe.proto:140
+  bucket_budget_ =
+      std::max<long>(1, net_->learnable_count() / std::max<long>(1, buckets));
+  if (E.mode == Mode::GPU && net_->learnable_count() > 0) {
+    history_ =
+        (float*)E.dalloc.alloc(sizeof(float) * net_->learnable_count());
+    HIP_CHECK(hipMemsetAsync(history_, 0,
+                             sizeof(float) * net_->learnable_count(),
+                             E.stream));
+  } else if (nparams > 0) {
+    host_history_.assign(net_->learnable_count(), 0.f);
+  }
+}
+
+Solver::~Solver() {
+  if (history_)
+    Engine::get().dalloc.release(history_,
+                                 sizeof(float) * net_->learnable_count());
+}
+
+float Solver::GetLearningRate() const {
+  // sgd_solver.cpp:24-66 (+ rampup :27-33)
+  const double base_lr = param_->num("base_lr");
+  const long rampup = param_->inum("rampup_interval", 0);
+  if (iter_ < rampup) {
+    const double r0 = param_->num("rampup_lr", 0.0);
+    return (float)(r0 + (base_lr - r0) * ((double)iter_ / rampup));
+  }
+  const std::string policy = param_->str("lr_policy", "fixed");
+  if (policy == "fixed") return (float)base_lr;
+  if (policy == "step") {
+    const long step = iter_ / std::max<long>(1, param_->inum("stepsize", 1));
+    return (float)(base_lr * std::pow(param_->num("gamma", 0.1), step));
+  }
+  if (policy == "exp")
+    return (float)(base_lr * std::pow(param_->num("gamma", 0.1), iter_));
+  if (policy == "inv")
+    return (float)(base_lr *
+                   std::pow(1.0 + param_->num("gamma") * iter_,
+                            -param_->num("power")));
+  if (policy == "multistep") {
+    auto sv = param_->inums("stepvalue");
+    while (current_step_ < (int)sv.size() && iter_ >= sv[current_step_])
+      ++current_step_;
+    return (float)(base_lr *
+                   std::pow(param_->num("gamma", 0.1), current_step_));
+  }
+  if (policy == "poly") {
+    const double min_lr = param_->num("min_lr", 0.0);
+    return (float)(min_lr +
+                   (base_lr - min_lr) *
+                       std::pow(1.0 - (double)iter_ /
+                                          param_->num("max_iter", 1),
+                                param_->num("power")));
+  }
+  if (policy == "sigmoid")
+    return (float)(base_lr /
+                   (1.0 + std::exp(-param_->num("gamma") *
+                                   (double)(iter_ -
+                                            param_->inum("stepsize")))));
+  CAMD_FATAL << "unknown lr_policy " << policy;
+}
+
+void Solver::bcast_weights() {
+  if (!comm_ || comm_->world() < 2) return;
+  Engine& E = Engine::get();
+  for (auto& p : net_->learnable_params()) {
+    float* ptr = E.mode == Mode::GPU ? p.blob->mutable_gpu_data()
+                                     : p.blob->mutable_cpu_data();
+    comm_->bcast(ptr, p.count, 0, E.stream);
+  }
+  E.sync();
+}
+
+void Reducer::start_iteration() {
+  bucket_start_ = bucket_end_ = 0;
+}
+
+void Reducer::param_ready(int param_id, hipEvent_t done) {
+  auto& params = solver_->net().learnable_params();
+  CHECK_EQ_((long)param_id, bucket_end_) << "params must arrive in order";
+  bucket_end_ = param_id + 1;
+  const auto& first = params[bucket_start_];
+  const auto& last = params[bucket_end_ - 1];
+  const long span = last.offset + (long)padded(last.count) - first.offset;
+  if (span >= solver_->bucket_budget_) flush(done);
+}
+
+void Reducer::flush(hipEvent_t ev) {
+  if (bucket_end_ == bucket_start_) return;
+  Solver& S = *solver_;
+  Engine& E = Engine::get();
+  auto& params = S.net().learnable_params();
+  const auto& first = params[bucket_start_];
+  const auto& last = params[bucket_end_ - 1];
+  const long count = last.offset + (long)padded(last.count) - first.offset;
+  if (E.mode == Mode::GPU) {
+    if (ev) HIP_CHECK(hipStreamWaitEvent(E.comm_stream, ev, 0));
+    if (S.comm_ && S.comm_->world() > 1) {
+      PerfScope ps("allreduce", E.comm_stream, 0, 2.0 * count * 4);
+      S.comm_->allreduce(S.net().diff_arena() + first.offset, count,
+                         E.comm_stream);
+    }
+    for (long k = bucket_start_; k < bucket_end_; ++k) {
+      const auto& p = params[k];
+      gpu::sgd_update(E.comm_stream, p.count,
+                      S.net().diff_arena() + p.offset,
+                      p.blob->mutable_gpu_data(), S.history() + p.offset,
+                      S.cur_mom_, S.cur_lr_ * p.lr_mult,
+                      S.weight_decay_ * p.decay_mult, S.grad_scale_);
+    }
+  } else {
+    if (S.comm_ && S.comm_->world() > 1) {
+      // CPU buckets: gather the param diffs into one flat range is
+      // unnecessary — host diffs are per-blob; reduce each param
+      for (long k = bucket_start_; k < bucket_end_; ++k)
+        S.comm_->allreduce(params[k].blob->mutable_cpu_diff(),
+                           params[k].count, nullptr);
+    }
+    for (long k = bucket_start_; k < bucket_end_; ++k) {
+      const auto& p = params[k];
+      float* g = p.blob->mutable_cpu_diff();
+      float* w = p.blob->mutable_cpu_data();
+      float* h = S.host_history().data() + p.offset;
+      const float lr = S.cur_lr_ * p.lr_mult;
+      const float decay = S.weight_decay_ * p.decay_mult;
+      for (long i = 0; i < p.count; ++i) {
+        float gi = g[i] * S.grad_scale_ + decay * w[i];
+        gi = h[i] = S.cur_mom_ * h[i] + lr * gi;
+        w[i] -= gi;
+        g[i] = 0.f;
+      }
+    }
+  }
+  bucket_start_ = bucket_end_;
+}
+
+void Reducer::iteration_end(hipEvent_t backward_done) {
+  flush(backward_done);
+  Engine& E = Engine::get();
+  if (E.mode == Mode::GPU) {
+    if (!comm_done_ev_)
+      HIP_CHECK(
+          hipEventCreateWithFlags(&comm_done_ev_, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(comm_done_ev_, E.comm_stream));
+    // next iteration's forward must see the updated weights
+    HIP_CHECK(hipStreamWaitEvent(E.stream, comm_done_ev_, 0));
+  }
+}
+
+void Solver::Step(int iters) {
+  Engine& E = Engine::get();
+  const long display = param_->inum("display", 0);
+  for (int i = 0; i < iters; ++i) {
+    cur_lr_ = GetLearningRate();
+    cur_mom_ = GetMomentum();
+    grad_scale_ = comm_ && comm_->world() > 1
+                      ? 1.f / (float)comm_->world()
+                      : 1.f;
+    reducer_.start_iteration();
+    net_->Forward();
+    net_->Backward(&reducer_);
+    ++iter_;
+    if (display > 0 && iter_ % display == 0) {
+      const float l = net_->loss();
+      fprintf(stderr, "[caffe_amd] Iteration %ld, loss = %g, lr = %g\n",
+              iter_, l, cur_lr_);
+    }
+  }
+  (void)E;
+}
+
+std::shared_ptr<Solver> create_solver_from_file(const std::string& path,
+                                                int batch_override) {
+  return std::make_shared<Solver>(parse_prototxt_file(path), batch_override);
+}
+
+}  // namespace camd
