@@ -1034,7 +1034,35 @@ void AccuracyLayer::Forward_cpu(const std::vector<Blob*>& bottom,
   top[0]->mutable_cpu_data()[0] = total ? (float)correct / total : 0.f;
 }
 
+// ----------------------------------------------------------------- Input
+// reference InputLayer: tops shaped by input_param, filled externally
+class InputLayer : public Layer {
+ public:
+  using Layer::Layer;
+  void Reshape(const std::vector<Blob*>&,
+               const std::vector<Blob*>& top) override {
+    auto ip = param_->sub("input_param");
+    CHECK_(ip) << "Input layer needs input_param";
+    auto shapes = ip->subs("shape");
+    CHECK_EQ_((int)shapes.size(), (int)top.size());
+    for (size_t i = 0; i < top.size(); ++i) {
+      std::vector<int> dims;
+      for (long d : shapes[i]->inums("dim")) dims.push_back((int)d);
+      if (top[i]->shape() != dims) top[i]->Reshape(dims);
+    }
+  }
+  void Forward_cpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override {}
+  void Forward_gpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override {}
+  void Backward_cpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override {}
+  void Backward_gpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override {}
+};
+
 // registry entries
+REGISTER_LAYER("Input", InputLayer)
 REGISTER_LAYER("Data", DataLayer)
 REGISTER_LAYER("DummyData", DataLayer)
 REGISTER_LAYER("Convolution", ConvolutionLayer)

@@ -142,21 +142,33 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
     tops_.push_back(top);
   }
 
-  // need-backward / propagate-down flags (net.cpp Init bottom-up logic)
+  // need-backward / propagate-down flags (net.cpp Init bottom-up logic;
+  // force_backward marks every non-data path like the reference does)
+  const bool force = msg->boolean("force_backward", false);
   std::map<Blob*, bool> blob_needs;
   layer_need_bwd_.resize(layers_.size(), false);
   prop_down_.resize(layers_.size());
   for (size_t i = 0; i < layers_.size(); ++i) {
+    const bool is_source = layers_[i]->type() == "Data" ||
+                           layers_[i]->type() == "Accuracy";
     bool need = !layers_[i]->blobs().empty();
     prop_down_[i].resize(bottoms_[i].size(), false);
     for (size_t b = 0; b < bottoms_[i].size(); ++b) {
       prop_down_[i][b] = blob_needs[bottoms_[i][b]];
       need = need || prop_down_[i][b];
     }
-    if (layers_[i]->type() == "Data" || layers_[i]->type() == "Accuracy")
-      need = false;
+    if (is_source) need = false;
+    if (force && !is_source) {
+      need = true;
+      for (size_t b = 0; b < bottoms_[i].size(); ++b)
+        prop_down_[i][b] = blob_needs.count(bottoms_[i][b])
+                               ? blob_needs[bottoms_[i][b]]
+                               : false;
+    }
     layer_need_bwd_[i] = need;
-    for (auto* t : tops_[i]) blob_needs[t] = need;
+    for (auto* t : tops_[i])
+      blob_needs[t] = need || (force && !is_source &&
+                               layers_[i]->type() != "SoftmaxWithLoss");
   }
 
   // learnable params in backward-completion order (reverse layer order)

@@ -1,0 +1,307 @@
+"""Layer-level parity checks: engine (CPU or GPU mode) vs oracle/.
+
+Each check builds a one-layer net through the engine's own prototxt path,
+drives it with seeded inputs, and compares every output and gradient with
+the oracle at 1e-4 relative fp32 (the reference's own tolerance class,
+test_convolution_layer.cpp:247).  Used by test_engine_cpu.py (CPU mode,
+runs everywhere) and test_gpu_parity.py (@gpu, MI355X).
+"""
+import numpy as np
+
+from engine_util import TOL, relerr, run_layer
+from oracle import oracle as orc
+
+rng = np.random.default_rng(20250915)
+
+
+def check_conv(mode, N=2, C=8, H=13, W=13, Co=16, k=3, s=1, p=1, grp=1,
+               bias=True):
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w = (rng.standard_normal((Co, C // grp, k, k)) * 0.2).astype(np.float32)
+    b = rng.standard_normal(Co).astype(np.float32) if bias else None
+    body = f"""layer {{
+  name: "conv"
+  type: "Convolution"
+  bottom: "in0"
+  top: "out"
+  convolution_param {{
+    num_output: {Co}
+    kernel_size: {k}
+    stride: {s}
+    pad: {p}
+    group: {grp}
+    bias_term: {"true" if bias else "false"}
+  }}
+}}"""
+    y_ref = orc.conv_fwd(x, w, b, pad=(p, p), stride=(s, s), group=grp)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer(mode, [(N, C, H, W)], body, [x],
+                       params=[w, b] if bias else [w], top_diff=dy)
+    assert relerr(y, y_ref) < TOL, f"conv fwd {relerr(y, y_ref)}"
+    dx_ref, dw_ref, db_ref = orc.conv_bwd(x, w, dy, pad=(p, p),
+                                          stride=(s, s), group=grp,
+                                          want_db=bias)
+    assert relerr(net.blob("in0", diff=True), dx_ref) < TOL, "conv dx"
+    assert relerr(net.param(0, diff=True),
+                  dw_ref.ravel()) < TOL, "conv dw"
+    if bias:
+        assert relerr(net.param(1, diff=True), db_ref) < TOL, "conv db"
+
+
+def check_ip(mode, M=4, K=32, Nout=10):
+    x = rng.standard_normal((M, K)).astype(np.float32)
+    w = rng.standard_normal((Nout, K)).astype(np.float32) * 0.2
+    w = w.astype(np.float32)
+    b = rng.standard_normal(Nout).astype(np.float32)
+    body = f"""layer {{
+  name: "ip"
+  type: "InnerProduct"
+  bottom: "in0"
+  top: "out"
+  inner_product_param {{ num_output: {Nout} }}
+}}"""
+    y_ref = orc.ip_fwd(x, w, b)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer(mode, [(M, K)], body, [x], params=[w, b],
+                       top_diff=dy)
+    assert relerr(y, y_ref) < TOL
+    dx_ref, dw_ref, db_ref = orc.ip_bwd(x, w, dy)
+    assert relerr(net.blob("in0", diff=True), dx_ref) < TOL
+    assert relerr(net.param(0, diff=True), dw_ref.ravel()) < TOL
+    assert relerr(net.param(1, diff=True), db_ref) < TOL
+
+
+def check_pool(mode, pool="MAX", N=2, C=4, H=13, W=13, k=3, s=2, p=0):
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    body = f"""layer {{
+  name: "pool"
+  type: "Pooling"
+  bottom: "in0"
+  top: "out"
+  pooling_param {{ pool: {pool} kernel_size: {k} stride: {s} pad: {p} }}
+}}"""
+    if pool == "MAX":
+        y_ref, mask = orc.pool_max_fwd(x, k, k, p, p, s, s)
+    else:
+        y_ref = orc.pool_ave_fwd(x, k, k, p, p, s, s)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer(mode, [(N, C, H, W)], body, [x], top_diff=None)
+    assert relerr(y, y_ref) < TOL
+    # pooling has no params; drive backward through an explicit top diff —
+    # pooling's prop_down is False in a net without params below, so check
+    # via a conv underneath instead? Simpler: pooling alone won't backprop.
+    # Covered at full-net level; here forward only.
+    if pool == "MAX":
+        _ = mask
+
+
+def check_pool_bwd(mode, pool="MAX", N=2, C=4, H=13, W=13, k=3, s=2):
+    # pool preceded by a 1x1 conv so prop_down[0] is true
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w = np.eye(C, dtype=np.float32).reshape(C, C, 1, 1).copy()
+    body = f"""layer {{
+  name: "c1"
+  type: "Convolution"
+  bottom: "in0"
+  top: "mid"
+  convolution_param {{ num_output: {C} kernel_size: 1 bias_term: false }}
+}}
+layer {{
+  name: "pool"
+  type: "Pooling"
+  bottom: "mid"
+  top: "out"
+  pooling_param {{ pool: {pool} kernel_size: {k} stride: {s} }}
+}}"""
+    if pool == "MAX":
+        y_ref, mask = orc.pool_max_fwd(x, k, k, 0, 0, s, s)
+    else:
+        y_ref = orc.pool_ave_fwd(x, k, k, 0, 0, s, s)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer(mode, [(N, C, H, W)], body, [x], params=[w],
+                       top_diff=dy)
+    assert relerr(y, y_ref) < TOL
+    if pool == "MAX":
+        dx_ref = orc.pool_max_bwd(dy, mask, H, W)
+    else:
+        dx_ref = orc.pool_ave_bwd(dy, H, W, k, k, 0, 0, s, s)
+    # dx lands in "mid"'s diff == conv's top diff; identity conv passes it on
+    assert relerr(net.blob("mid", diff=True), dx_ref) < TOL
+
+
+def check_bn(mode, N=4, C=6, H=5, W=5, scale_bias=True):
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    sc = (rng.standard_normal(C) + 1.5).astype(np.float32)
+    bi = rng.standard_normal(C).astype(np.float32)
+    eps = 1e-4
+    body = f"""layer {{
+  name: "bn"
+  type: "BatchNorm"
+  bottom: "in0"
+  top: "out"
+  batch_norm_param {{
+    moving_average_fraction: 0.9
+    eps: {eps}
+    scale_bias: {"true" if scale_bias else "false"}
+  }}
+}}"""
+    y_ref, mean, var, inv_std, xnorm = orc.bn_fwd_train(
+        x, eps, sc if scale_bias else None, bi if scale_bias else None)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    # learnable params of BN are [scale, bias] (stats blobs skipped)
+    params = [sc, bi] if scale_bias else None
+    net, y = run_layer(mode, [(N, C, H, W)], body, [x], params=params,
+                       top_diff=dy)
+    assert relerr(y, y_ref) < TOL, f"bn fwd {relerr(y, y_ref)}"
+    dx_ref, dsc_ref, dbi_ref = orc.bn_bwd(xnorm, dy, inv_std,
+                                          sc if scale_bias else None)
+    assert relerr(net.blob("in0", diff=True), dx_ref) < TOL, "bn dx"
+    if scale_bias:
+        # learnable params (arena order): scale=idx 0, bias=idx 1 of this
+        # layer; BN stats blobs are skipped (skip_apply_update)
+        assert relerr(net.param(0, diff=True), dsc_ref) < TOL, "bn dscale"
+        assert relerr(net.param(1, diff=True), dbi_ref) < TOL, "bn dbias"
+
+
+def check_relu(mode, n=1000):
+    x = rng.standard_normal((2, 5, 10, 10)).astype(np.float32)
+    body = """layer {
+  name: "c1"
+  type: "Convolution"
+  bottom: "in0"
+  top: "mid"
+  convolution_param { num_output: 5 kernel_size: 1 bias_term: false }
+}
+layer { name: "r" type: "ReLU" bottom: "mid" top: "out" }"""
+    w = np.eye(5, dtype=np.float32).reshape(5, 5, 1, 1).copy()
+    y_ref = orc.relu_fwd(x)
+    dy = rng.standard_normal(x.shape).astype(np.float32)
+    net, y = run_layer(mode, [x.shape], body, [x], params=[w], top_diff=dy)
+    assert relerr(y, y_ref) < TOL
+    assert relerr(net.blob("mid", diff=True), orc.relu_bwd(x, dy)) < TOL
+
+
+def check_lrn(mode, N=2, C=8, H=6, W=6, size=5, alpha=1e-4, beta=0.75):
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w = np.eye(C, dtype=np.float32).reshape(C, C, 1, 1).copy()
+    body = f"""layer {{
+  name: "c1"
+  type: "Convolution"
+  bottom: "in0"
+  top: "mid"
+  convolution_param {{ num_output: {C} kernel_size: 1 bias_term: false }}
+}}
+layer {{
+  name: "lrn"
+  type: "LRN"
+  bottom: "mid"
+  top: "out"
+  lrn_param {{ local_size: {size} alpha: {alpha} beta: {beta} }}
+}}"""
+    y_ref, scale = orc.lrn_fwd(x, size, alpha, beta)
+    dy = rng.standard_normal(x.shape).astype(np.float32)
+    net, y = run_layer(mode, [x.shape], body, [x], params=[w], top_diff=dy)
+    assert relerr(y, y_ref) < TOL
+    dx_ref = orc.lrn_bwd(x, y_ref, dy, scale, size, alpha, beta)
+    assert relerr(net.blob("mid", diff=True), dx_ref) < TOL
+
+
+def check_softmaxloss(mode, N=8, C=10):
+    x = (rng.standard_normal((N, C)) * 3).astype(np.float32)
+    lab = rng.integers(0, C, N).astype(np.float32)
+    body = """layer {
+  name: "ip"
+  type: "InnerProduct"
+  bottom: "in0"
+  top: "fc"
+  inner_product_param { num_output: %d }
+}
+layer {
+  name: "loss"
+  type: "SoftmaxWithLoss"
+  bottom: "fc"
+  bottom: "in1"
+  top: "out"
+}""" % C
+    w = np.eye(C, dtype=np.float32)
+    prob = orc.softmax_fwd(x, N, C, 1)
+    loss_ref = orc.softmaxloss_fwd(prob, lab, N, C, 1)
+    net, y = run_layer(mode, [(N, C), (N,)], body, [x, lab],
+                       params=[w, np.zeros(C, np.float32)])
+    assert abs(y[0] - loss_ref) < 1e-5 * max(1, abs(loss_ref))
+    net.backward()
+    dx_ref = orc.softmaxloss_bwd(prob, lab, N, C, 1)
+    assert relerr(net.blob("fc", diff=True), dx_ref) < TOL
+
+
+def check_eltwise_concat(mode):
+    N, C, H, W = 2, 4, 5, 5
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w1 = np.eye(C, dtype=np.float32).reshape(C, C, 1, 1).copy()
+    w2 = (np.eye(C) * 2).astype(np.float32).reshape(C, C, 1, 1).copy()
+    body = f"""layer {{
+  name: "a"
+  type: "Convolution"
+  bottom: "in0"
+  top: "a"
+  convolution_param {{ num_output: {C} kernel_size: 1 bias_term: false }}
+}}
+layer {{
+  name: "b"
+  type: "Convolution"
+  bottom: "in0"
+  top: "b"
+  convolution_param {{ num_output: {C} kernel_size: 1 bias_term: false }}
+}}
+layer {{
+  name: "sum"
+  type: "Eltwise"
+  bottom: "a"
+  bottom: "b"
+  top: "sum"
+  eltwise_param {{ operation: SUM }}
+}}
+layer {{
+  name: "cat"
+  type: "Concat"
+  bottom: "sum"
+  bottom: "a"
+  top: "out"
+}}"""
+    dy = rng.standard_normal((N, 2 * C, H, W)).astype(np.float32)
+    # learnable order is backward-completion (reverse layers): conv b, conv a
+    net, y = run_layer(mode, [x.shape], body, [x], params=[w2, w1],
+                       top_diff=dy)
+    a, b = x, 2 * x
+    y_ref = np.concatenate([a + b, a], axis=1)
+    assert relerr(y, y_ref) < TOL
+    # backward: d(sum)=dy[:, :C], d(a)=dy[:, :C] (via sum) + dy[:, C:]
+    assert relerr(net.blob("sum", diff=True), dy[:, :C]) < TOL
+    assert relerr(net.blob("a", diff=True), dy[:, :C] + dy[:, C:]) < TOL
+    # in0 diff: conv a contributes 1*d(a), conv b contributes 2*d(sum)
+    din_ref = (dy[:, :C] + dy[:, C:]) + 2 * dy[:, :C]
+    assert relerr(net.blob("in0", diff=True), din_ref) < TOL
+
+
+ALL_CHECKS = {
+    "conv_3x3": lambda m: check_conv(m),
+    "conv_7x7s2": lambda m: check_conv(m, C=3, H=19, W=19, Co=8, k=7, s=2,
+                                       p=3),
+    "conv_1x1": lambda m: check_conv(m, k=1, p=0, bias=False),
+    "conv_1x1s2": lambda m: check_conv(m, k=1, p=0, s=2, bias=False),
+    "conv_group": lambda m: check_conv(m, C=8, Co=16, k=5, p=2, grp=2),
+    "conv_5x5_odd": lambda m: check_conv(m, N=3, C=5, H=11, W=17, Co=7, k=5,
+                                         s=3, p=2),
+    "ip": check_ip,
+    "ip_large": lambda m: check_ip(m, M=130, K=260, Nout=140),
+    "pool_max": lambda m: check_pool(m, "MAX"),
+    "pool_ave": lambda m: check_pool(m, "AVE"),
+    "pool_max_bwd": lambda m: check_pool_bwd(m, "MAX"),
+    "pool_ave_bwd": lambda m: check_pool_bwd(m, "AVE"),
+    "bn": check_bn,
+    "bn_large_s": lambda m: check_bn(m, N=3, C=4, H=17, W=17),
+    "relu": check_relu,
+    "lrn": check_lrn,
+    "softmaxloss": check_softmaxloss,
+    "eltwise_concat": check_eltwise_concat,
+}
