@@ -208,13 +208,15 @@ def main():
 
     import caffe_amd as ca
 
+    # initialize HIP through /opt/rocm's runtime BEFORE torch gets a chance
+    # to load its bundled one (same soname; first load wins process-wide)
+    ca.set_mode("gpu", local_rank)
+
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
         dist.init_process_group("gloo", rank=rank, world_size=world)
-
-    ca.set_mode("gpu", local_rank)
     shapes = {"resnet50": (3, 224, 224), "alexnet": (3, 227, 227),
               "googlenet": (3, 224, 224)}
     ca.set_synthetic_shape(*shapes[args.model], 1000)

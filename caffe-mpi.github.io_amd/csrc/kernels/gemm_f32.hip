@@ -269,10 +269,13 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     g.relu = epi->relu;
   }
   const long tm = (M + BM - 1) / BM, tn = (N + BN - 1) / BN;
-  // split-K when the output grid cannot fill the chip (wgrad shapes)
+  // split-K when the output grid cannot fill the chip (wgrad shapes);
+  // restricted to plain-C calls (no epilogue, beta=0) — the fused-epilogue
+  // GEMMs (conv/IP forward) have huge N and never need it
   int SK = 1;
-  if (tm * tn < 512 && K > 4 * BK) {
-    SK = (int)std::min<long>({512 / (tm * tn) + 1, (K + 4 * BK - 1) / (4 * BK), 64});
+  if (!epi && beta == 0.f && tm * tn < 512 && K > 4 * BK) {
+    SK = (int)std::min<long>(
+        {512 / (tm * tn) + 1, (K + 4 * BK - 1) / (4 * BK), 64});
   }
   PerfScope perf("gemm", s, 2.0 * M * N * K,
                  4.0 * (M * K + N * K + M * N));
