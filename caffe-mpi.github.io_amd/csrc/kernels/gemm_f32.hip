@@ -1,20 +1,21 @@
 // gemm_f32.hip — hand-written fp32 MFMA GEMM for gfx950 (CDNA4).
 //
-// The one kernel family that carries the conv/IP contractions (SURVEY.md
-// §8a a1-a4).  Uses the f32-input MFMA `v_mfma_f32_32x32x2_f32`
-// (exact f32, 157 TF chip peak = the f32 vector peak; no xf32 on gfx950 —
-// cdna_hip_programming.md §3).  Structure:
-//   128×128 block tile, 256 threads = 4 waves in a 2×2 grid, each wave owns
-//   a 64×64 sub-tile as 2×2 MFMA 32×32 accumulators (4 independent
-//   accumulators per wave reach the 64-cyc issue rate with one wave/SIMD).
-//   BK=32 K-steps staged via registers into double-buffered LDS
-//   [BK][128+1] (pad +1 → conflict-free b32 reads/writes), T14-style
-//   issue-early/write-late (cdna_hip_programming.md §6 G15).
-// All four transpose combos; α/β; fused epilogue: per-row or per-col bias,
-// optional NCHW scatter (conv output goes straight to N-major layout, no
-// separate bias/copy pass).  Deterministic split-K for the
-// reduction-over-batch wgrad shapes (few output tiles, huge K): partial
-// slabs + fixed-order reduce — no atomics (SURVEY.md §7 hard part (b)).
+// The kernel family carrying the conv/IP contractions (SURVEY.md §8a
+// a1-a4), built on the f32-input MFMA `v_mfma_f32_32x32x2_f32` (exact f32,
+// 157.3 TF chip peak; no xf32 on gfx950 — cdna_hip_programming.md §3).
+//
+// Structure: 128×128 block tile, 256 threads = 4 waves (2×2), each wave a
+// 64×64 sub-tile as 2×2 MFMA 32×32 accumulators; BK=32 K-steps staged
+// through double-buffered LDS [BK][128+1] (+1 pad → conflict-free b32
+// banking), T14 issue-early/write-late, vectorized float4 staging loads on
+// the aligned in-bounds fast path, bijective XCD-aware block swizzle (T1).
+//
+// Operands can be NCHW *views* (GemmView): conv GEMMs read x / dY straight
+// from NCHW layout — no transpose pass, and 1×1/s1 convolutions run with no
+// im2col/col2im at all.  Fused epilogue: bias (per row/col), optional
+// ReLU, NCHW scatter.  Deterministic split-K (partial slabs + fixed-order
+// reduce, no atomics — SURVEY.md §7 hard part (b)) for the few-tile/huge-K
+// wgrad shapes.
 #include <hip/hip_runtime.h>
 
 #include "../layers.hpp"
@@ -23,10 +24,10 @@ namespace camd {
 namespace gpu {
 
 using f32x16 = __attribute__((ext_vector_type(16))) float;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
 
 constexpr int BM = 128, BN = 128, BK = 32;
-constexpr int LDA_S = BM + 1;  // LDS row stride (floats)
-constexpr int LDB_S = BN + 1;
+constexpr int LDS_S = BM + 1;  // LDS row stride (floats), BM == BN
 
 struct GemmArgs {
   const float* A;
@@ -40,34 +41,65 @@ struct GemmArgs {
   const float* bias;
   int bias_per_col;
   int relu;
+  // operand views (spad==0 => plain)
+  GemmView av, bv;
   // split-K
   float* slab;  // partials [SK][M][N] when SK>1
   int SK;
+  // swizzle
+  long tn, tiles;  // column-tile count, total tiles
 };
 
-// stage op(A) tile rows [m0,m0+BM) x [k0,k0+BK) into regs (16 floats)
+// Read 16 consecutive elements along the contiguous axis.
+//   plain:  base = P + r*ld + q,      valid j while q + j < qmax, r < rmax
+//   view:   n = q / spad, sp = q % spad,
+//           base = P + (n*chan + r)*S + sp, valid j while sp + j < S
+// (q is 16-aligned and spad % 64 == 0, so a chunk never crosses images)
+__device__ __forceinline__ void read16(const float* __restrict__ P, long r,
+                                       long q, long ld, long rmax,
+                                       long qmax, const GemmView& v,
+                                       float (&out)[16]) {
+  const float* p;
+  long nvalid;  // elements valid from j=0
+  if (v.spad) {
+    const long n = q / v.spad;
+    const long sp = q - n * v.spad;
+    p = P + (n * v.chan + r) * v.S + sp;
+    nvalid = (r < rmax && sp < v.S && q < qmax) ? v.S - sp : 0;
+  } else {
+    p = P + r * ld + q;
+    nvalid = (r < rmax && q < qmax) ? qmax - q : 0;
+  }
+  if (nvalid >= 16 && (((uintptr_t)p) & 15) == 0) {
+    const f32x4* p4 = (const f32x4*)p;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const f32x4 t = p4[j];
+      out[4 * j + 0] = t.x;
+      out[4 * j + 1] = t.y;
+      out[4 * j + 2] = t.z;
+      out[4 * j + 3] = t.w;
+    }
+  } else {
+#pragma unroll
+    for (int j = 0; j < 16; ++j) out[j] = j < nvalid ? p[j] : 0.f;
+  }
+}
+
+// ---- staging: global -> regs (load) and regs -> LDS (write)
+// A tile is held in LDS as As[k][m] (k-major); B as Bs[k][n].
 template <bool TRANS>
 __device__ __forceinline__ void stage_a_load(const GemmArgs& g, long m0,
                                              long k0, float (&r)[16]) {
   const int t = threadIdx.x;
   if (!TRANS) {
-    // A[M][K]: thread reads A[m0 + (t&127)][k0 + (t>>7)*16 + j]
-    const long m = m0 + (t & 127);
-    const long kb = k0 + (t >> 7) * 16;
-    const float* p = g.A + m * g.lda + kb;
-    const bool mok = m < g.M;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      r[j] = (mok && kb + j < g.K) ? p[j] : 0.f;
+    // A[M][Kd] — contiguous along Kd (view axis allowed)
+    read16(g.A, m0 + (t & 127), k0 + (t >> 7) * 16, g.lda, g.M, g.K, g.av,
+           r);
   } else {
-    // A stored [K][M]; op(A)(m,k)=A[k][m]: read A[k0 + (t&31)][m0+(t>>5)*16+j]
-    const long k = k0 + (t & 31);
-    const long mb = m0 + (t >> 5) * 16;
-    const float* p = g.A + k * g.lda + mb;
-    const bool kok = k < g.K;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      r[j] = (kok && mb + j < g.M) ? p[j] : 0.f;
+    // A stored [Kd][M] — contiguous along M (no view on this case)
+    read16(g.A, k0 + (t & 31), m0 + (t >> 5) * 16, g.lda, g.K, g.M, g.av,
+           r);
   }
 }
 
@@ -79,12 +111,12 @@ __device__ __forceinline__ void stage_a_write(float* As,
     const int m = t & 127;
     const int kb = (t >> 7) * 16;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) As[(kb + j) * LDA_S + m] = r[j];
+    for (int j = 0; j < 16; ++j) As[(kb + j) * LDS_S + m] = r[j];
   } else {
     const int k = t & 31;
     const int mb = (t >> 5) * 16;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) As[k * LDA_S + mb + j] = r[j];
+    for (int j = 0; j < 16; ++j) As[k * LDS_S + mb + j] = r[j];
   }
 }
 
@@ -93,23 +125,13 @@ __device__ __forceinline__ void stage_b_load(const GemmArgs& g, long n0,
                                              long k0, float (&r)[16]) {
   const int t = threadIdx.x;
   if (!TRANS) {
-    // B[K][N]: read B[k0 + (t&31)][n0 + (t>>5)*16 + j]
-    const long k = k0 + (t & 31);
-    const long nb = n0 + (t >> 5) * 16;
-    const float* p = g.B + k * g.ldb + nb;
-    const bool kok = k < g.K;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      r[j] = (kok && nb + j < g.N) ? p[j] : 0.f;
+    // B[Kd][N] — contiguous along N (dY view in fwd/dgrad)
+    read16(g.B, k0 + (t & 31), n0 + (t >> 5) * 16, g.ldb, g.K, g.N, g.bv,
+           r);
   } else {
-    // B stored [N][K]; op(B)(k,n)=B[n][k]: read B[n0+(t&127)][k0+(t>>7)*16+j]
-    const long n = n0 + (t & 127);
-    const long kb = k0 + (t >> 7) * 16;
-    const float* p = g.B + n * g.ldb + kb;
-    const bool nok = n < g.N;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      r[j] = (nok && kb + j < g.K) ? p[j] : 0.f;
+    // B stored [N][Kd] — contiguous along Kd (x/col view in wgrad)
+    read16(g.B, n0 + (t & 127), k0 + (t >> 7) * 16, g.ldb, g.N, g.K, g.bv,
+           r);
   }
 }
 
@@ -121,22 +143,31 @@ __device__ __forceinline__ void stage_b_write(float* Bs,
     const int k = t & 31;
     const int nb = (t >> 5) * 16;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) Bs[k * LDB_S + nb + j] = r[j];
+    for (int j = 0; j < 16; ++j) Bs[k * LDS_S + nb + j] = r[j];
   } else {
     const int n = t & 127;
     const int kb = (t >> 7) * 16;
 #pragma unroll
-    for (int j = 0; j < 16; ++j) Bs[(kb + j) * LDB_S + n] = r[j];
+    for (int j = 0; j < 16; ++j) Bs[(kb + j) * LDS_S + n] = r[j];
   }
 }
 
 template <bool TA, bool TB, bool SPLITK>
 __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
-  __shared__ float As[2][BK * LDA_S];
-  __shared__ float Bs[2][BK * LDB_S];
+  __shared__ float As[2][BK * LDS_S];
+  __shared__ float Bs[2][BK * LDS_S];
 
-  const long tile_n = blockIdx.x;
-  const long tile_m = blockIdx.y;
+  // bijective XCD-aware swizzle (cdna_hip_programming.md T1): contiguous
+  // tile chunks per XCD so neighbouring tiles share L2-resident panels
+  long flat = blockIdx.x;
+  {
+    const long nwg = g.tiles;
+    const long q = nwg / 8, rr = nwg % 8;
+    const long xcd = flat % 8, idx = flat / 8;
+    flat = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const long tile_m = flat / g.tn;
+  const long tile_n = flat - tile_m * g.tn;
   const long m0 = tile_m * BM, n0 = tile_n * BN;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -144,16 +175,13 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   const int row_in = lane & 31;             // MFMA row/col index
   const int ksel = lane >> 5;               // which of the 2 K elems
 
-  // K range for this block (split-K slice)
+  // K range (split-K slice; starts BK-aligned so chunks stay 16-aligned)
   long k_lo = 0, k_hi = g.K;
   if (SPLITK) {
     const int sk = blockIdx.z;
-    k_lo = g.K * sk / g.SK;
-    k_hi = g.K * (sk + 1) / g.SK;
-    // align slice starts to BK so staging tiles stay aligned
-    k_lo = k_lo / BK * BK;
-    k_hi = (blockIdx.z == g.SK - 1) ? g.K : k_hi / BK * BK;
-    if (k_lo >= k_hi) return;
+    k_lo = g.K * sk / g.SK / BK * BK;
+    k_hi = (sk == g.SK - 1) ? g.K : g.K * (sk + 1) / g.SK / BK * BK;
+    if (k_lo >= k_hi) return;  // cannot happen for SK <= ceil(K/4BK)
   }
   const long ntiles = (k_hi - k_lo + BK - 1) / BK;
 
@@ -177,10 +205,10 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 2) {
       const int krow = kk + ksel;
-      const float a0 = Ab[krow * LDA_S + wr * 64 + row_in];
-      const float a1 = Ab[krow * LDA_S + wr * 64 + 32 + row_in];
-      const float b0 = Bb[krow * LDB_S + wc * 64 + row_in];
-      const float b1 = Bb[krow * LDB_S + wc * 64 + 32 + row_in];
+      const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+      const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+      const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+      const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
       acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
       acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
       acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
@@ -196,8 +224,7 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   }
 
   // ---- epilogue: acc reg r -> (row, col)
-  // C/D map for 32x32 MFMA: col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
-  // (rule 20: no runtime-indexed vector arrays — four explicit calls)
+  // C/D map (32x32 MFMA): col = lane&31, row = (r&3)+8*(r>>2)+4*(lane>>5)
   auto epi_tile = [&](const f32x16& a, int ti, int tj) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -244,10 +271,12 @@ __global__ void k_splitk_reduce(const float* __restrict__ slab, long MN,
 
 void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
           float alpha, const float* A, long lda, const float* B, long ldb,
-          float beta, float* C, long ldc, const GemmEpi* epi) {
+          float beta, float* C, long ldc, const GemmEpi* epi,
+          const GemmView* aview, const GemmView* bview) {
   CHECK_GT_(M, 0);
   CHECK_GT_(N, 0);
   CHECK_GT_(K, 0);
+  CHECK_(!(aview && transA)) << "A view requires no transpose";
   GemmArgs g{};
   g.A = A;
   g.B = B;
@@ -268,21 +297,23 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     g.bias_per_col = epi->bias_per_col;
     g.relu = epi->relu;
   }
+  if (aview) g.av = *aview;
+  if (bview) g.bv = *bview;
   const long tm = (M + BM - 1) / BM, tn = (N + BN - 1) / BN;
+  g.tn = tn;
+  g.tiles = tm * tn;
   // split-K when the output grid cannot fill the chip (wgrad shapes);
-  // restricted to plain-C calls (no epilogue, beta=0) — the fused-epilogue
-  // GEMMs (conv/IP forward) have huge N and never need it
+  // plain-C calls only (no epilogue) — fused-epilogue GEMMs have huge N
   int SK = 1;
-  if (!epi && beta == 0.f && tm * tn < 512 && K > 4 * BK) {
+  if (!epi && beta == 0.f && g.tiles < 512 && K > 4 * BK) {
     SK = (int)std::min<long>(
-        {512 / (tm * tn) + 1, (K + 4 * BK - 1) / (4 * BK), 64});
+        {512 / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 64});
   }
   PerfScope perf("gemm", s, 2.0 * M * N * K,
                  4.0 * (M * K + N * K + M * N));
-  dim3 grid((unsigned)tn, (unsigned)tm, (unsigned)SK);
+  dim3 grid((unsigned)g.tiles, 1, (unsigned)SK);
   dim3 block(256);
   if (SK > 1) {
-    CHECK_(!epi && beta == 0.f) << "split-K path supports plain C only";
     float* slab = (float*)Workspace::get_global().get(
         10, sizeof(float) * (size_t)SK * M * N);
     g.slab = slab;
@@ -317,6 +348,34 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   else
     hipLaunchKernelGGL((k_gemm_f32<true, true, false>), grid, block, 0, s,
                        g);
+}
+
+// db[c] = Σ_n Σ_s dy[n][c][s] — one block per channel, double block-reduce
+__global__ void k_bias_grad(const float* __restrict__ dy, int N, int C,
+                            long S, float* __restrict__ db) {
+  for (int c = blockIdx.x; c < C; c += gridDim.x) {
+    double acc = 0;
+    for (long i = threadIdx.x; i < (long)N * S; i += blockDim.x) {
+      const long n = i / S;
+      const long sp = i - n * S;
+      acc += dy[(n * C + c) * S + sp];
+    }
+    __shared__ double sh[256];
+    sh[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = 128; off > 0; off >>= 1) {
+      if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) db[c] = (float)sh[0];
+    __syncthreads();
+  }
+}
+void bias_grad(hipStream_t s, const float* dy, int N, int C, long S,
+               float* db) {
+  PerfScope perf("reduce", s, 0, 4.0 * N * C * S);
+  hipLaunchKernelGGL(k_bias_grad, dim3(std::min(C, 2048)), dim3(256), 0, s,
+                     dy, N, C, S, db);
 }
 
 }  // namespace gpu

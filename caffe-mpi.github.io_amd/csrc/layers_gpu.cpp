@@ -12,6 +12,7 @@
 namespace camd {
 
 using gpu::GemmEpi;
+using gpu::GemmView;
 
 void DataLayer::Forward_gpu(const std::vector<Blob*>&,
                             const std::vector<Blob*>& top) {
@@ -26,6 +27,16 @@ void DataLayer::Forward_gpu(const std::vector<Blob*>&,
 }
 
 // ------------------------------------------------------------------ Conv
+// MI355X-first conv: whole-batch GEMMs with NCHW-*view* operands — dY and
+// (for 1x1/s1) x are read straight out of NCHW inside the GEMM staging, so
+// there is no transpose pass anywhere and 1x1/s1 convolutions run with no
+// im2col/col2im at all.  Non-1x1 convs keep an explicit batched col buffer
+// [K][N*Spad] (implicit-GEMM fusion is the next step).
+static bool conv_is_1x1(const ConvolutionLayer& l) {
+  return l.kh_ == 1 && l.kw_ == 1 && l.sh_ == 1 && l.sw_ == 1 && !l.ph_ &&
+         !l.pw_ && l.group_ == 1;
+}
+
 void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
                                    const std::vector<Blob*>& top) {
   Engine& E = Engine::get();
@@ -36,33 +47,27 @@ void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   const float* w = blobs_[0]->gpu_data();
   float* y = top[0]->mutable_gpu_data();
 
-  const float* col;
-  if (kh_ == 1 && kw_ == 1 && sh_ == 1 && sw_ == 1 && !ph_ && !pw_ &&
-      S_ == Spad_ && group_ == 1) {
-    // 1x1/s1 with aligned S: x[N][C][S] viewed as [C][N*S]? NOT directly —
-    // x is N-major.  Still need the transpose into [C][N*Spad].
-    float* xt = (float*)ws.get(0, sizeof(float) * (size_t)C_ * NS);
-    gpu::nchw_to_cpad(E.stream, x, N_, C_, S_, Spad_, xt);
-    col = xt;
-  } else {
-    float* colb =
-        (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
-    gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_,
-                        sh_, sw_, dh_, dw_, OH_, OW_, Spad_, colb);
-    col = colb;
-  }
   GemmEpi epi;
   epi.spad = Spad_;
   epi.S = S_;
   epi.n_stride = (long)Cout_ * S_;
+  epi.bias = bias_ ? blobs_[1]->gpu_data() : nullptr;
+  if (conv_is_1x1(*this)) {
+    GemmView xv{Spad_, S_, C_};  // x[N][C][S] as [C][N*Spad]
+    gpu::gemm(E.stream, false, false, Cout_, NS, K, 1.f, w, K, x, 0, 0.f, y,
+              S_, &epi, nullptr, &xv);
+    return;
+  }
+  float* colb =
+      (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
+  gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_, sh_,
+                      sw_, dh_, dw_, OH_, OW_, Spad_, colb);
   for (int g = 0; g < group_; ++g) {
     epi.bias = bias_ ? blobs_[1]->gpu_data() + (long)g * (Cout_ / group_)
                      : nullptr;
     gpu::gemm(E.stream, false, false, Cout_ / group_, NS, K, 1.f,
-              w + (long)g * (Cout_ / group_) * K, K, col + (long)g * K * NS,
-              NS, 0.f,
-              // base pre-offset by the group's channel block; the epilogue
-              // scatters C[n*n_stride + row*S + s]
+              w + (long)g * (Cout_ / group_) * K, K,
+              colb + (long)g * K * NS, NS, 0.f,
               y + (long)g * (Cout_ / group_) * S_, S_, &epi);
   }
 }
@@ -77,57 +82,55 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
   const float* x = bottom[0]->gpu_data();
   const float* w = blobs_[0]->gpu_data();
   const float* dy = top[0]->gpu_diff();
+  GemmView dyv{Spad_, S_, Cout_};  // dY[N][Cout][S] as [Cout][N*Spad]
 
-  // dY into [Cout][N*Spad]
-  float* dyp = (float*)ws.get(2, sizeof(float) * (size_t)Cout_ * NS);
-  gpu::nchw_to_cpad(E.stream, dy, N_, Cout_, S_, Spad_, dyp);
-
-  // bias grad: db = dY · 1 — use GEMM-free reduction via gemm with K=NS on
-  // a ones vector is wasteful; a small dedicated reduce would do, but the
-  // wgrad GEMM already streams dyp; reuse gemm with N=1 costs little.
-  const bool is1x1 = kh_ == 1 && kw_ == 1 && sh_ == 1 && sw_ == 1 && !ph_ &&
-                     !pw_ && S_ == Spad_ && group_ == 1;
-  const float* col;
-  if (is1x1) {
-    float* xt = (float*)ws.get(0, sizeof(float) * (size_t)C_ * NS);
-    gpu::nchw_to_cpad(E.stream, x, N_, C_, S_, Spad_, xt);
-    col = xt;
-  } else {
-    float* colb =
-        (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
-    gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_,
-                        sh_, sw_, dh_, dw_, OH_, OW_, Spad_, colb);
-    col = colb;
-  }
-  // wgrad: dW[Cout][K] = dYp · colᵀ  (K-dim = N*Spad; pads are zero in col)
-  for (int g = 0; g < group_; ++g)
-    gpu::gemm(E.stream, false, true, Cout_ / group_, K, NS, 1.f,
-              dyp + (long)g * (Cout_ / group_) * NS, NS,
-              col + (long)g * K * NS, NS, 0.f,
-              blobs_[0]->mutable_gpu_diff() + (long)g * (Cout_ / group_) * K,
-              K, nullptr);
   if (bias_)
-    gpu::rowsum(E.stream, dyp, Cout_, NS, blobs_[1]->mutable_gpu_diff());
-  if (prop_down[0]) {
-    float* dx = bottom[0]->mutable_gpu_diff();
-    if (is1x1) {
-      // dxt[C][N*S] = Wᵀ·dYp, then transpose back — avoid extra buffer by
-      // computing into dcol then scattering with col2im (k=1 case works)
-      float* dcol = (float*)ws.get(1, sizeof(float) * (size_t)K * NS);
-      gpu::gemm(E.stream, true, false, K, NS, Cout_, 1.f, w, K, dyp, NS,
-                0.f, dcol, NS, nullptr);
-      gpu::col2im_batched(E.stream, dcol, N_, C_, H_, W_, 1, 1, 0, 0, 1, 1,
-                          1, 1, OH_, OW_, Spad_, dx);
-    } else {
-      float* dcol = (float*)ws.get(1, sizeof(float) * (size_t)K * group_ * NS);
-      for (int g = 0; g < group_; ++g)
-        gpu::gemm(E.stream, true, false, K, NS, Cout_ / group_, 1.f,
-                  w + (long)g * (Cout_ / group_) * K, K,
-                  dyp + (long)g * (Cout_ / group_) * NS, NS, 0.f,
-                  dcol + (long)g * K * NS, NS, nullptr);
-      gpu::col2im_batched(E.stream, dcol, N_, C_, H_, W_, kh_, kw_, ph_,
-                          pw_, sh_, sw_, dh_, dw_, OH_, OW_, Spad_, dx);
+    gpu::bias_grad(E.stream, dy, N_, Cout_, S_,
+                   blobs_[1]->mutable_gpu_diff());
+
+  if (conv_is_1x1(*this)) {
+    GemmView xv{Spad_, S_, C_};
+    // wgrad: dW[Cout][C] = dY-view · x-viewᵀ (Kd = N*Spad, split-K capable)
+    gpu::gemm(E.stream, false, true, Cout_, K, NS, 1.f, dy, 0, x, 0, 0.f,
+              blobs_[0]->mutable_gpu_diff(), K, nullptr, &dyv, &xv);
+    if (prop_down[0]) {
+      // dgrad: dx = Wᵀ · dY-view, scattered straight into NCHW
+      GemmEpi epi;
+      epi.spad = Spad_;
+      epi.S = S_;
+      epi.n_stride = (long)C_ * S_;
+      gpu::gemm(E.stream, true, false, K, NS, Cout_, 1.f, w, K, dy, 0, 0.f,
+                bottom[0]->mutable_gpu_diff(), S_, &epi, nullptr, &dyv);
     }
+    return;
+  }
+
+  float* colb =
+      (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
+  gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_, sh_,
+                      sw_, dh_, dw_, OH_, OW_, Spad_, colb);
+  for (int g = 0; g < group_; ++g) {
+    // wgrad: dW = dY-view · colᵀ; the group's dY channels start at
+    // g*(Cout/group) — fold the channel offset into the view base pointer
+    GemmView dyvg{Spad_, S_, Cout_};
+    gpu::gemm(E.stream, false, true, Cout_ / group_, K, NS, 1.f,
+              dy + (long)g * (Cout_ / group_) * S_, 0,
+              colb + (long)g * K * NS, NS, 0.f,
+              blobs_[0]->mutable_gpu_diff() +
+                  (long)g * (Cout_ / group_) * K,
+              K, nullptr, &dyvg);
+  }
+  if (prop_down[0]) {
+    float* dcol =
+        (float*)ws.get(1, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
+    for (int g = 0; g < group_; ++g)
+      gpu::gemm(E.stream, true, false, K, NS, Cout_ / group_, 1.f,
+                w + (long)g * (Cout_ / group_) * K, K,
+                dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+                dcol + (long)g * K * NS, NS, nullptr, nullptr, &dyv);
+    gpu::col2im_batched(E.stream, dcol, N_, C_, H_, W_, kh_, kw_, ph_, pw_,
+                        sh_, sw_, dh_, dw_, OH_, OW_, Spad_,
+                        bottom[0]->mutable_gpu_diff());
   }
 }
 

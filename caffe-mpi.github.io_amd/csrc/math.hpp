@@ -48,9 +48,24 @@ struct GemmEpi {
   bool relu = false;
 };
 
+// NCHW-view operand: the GEMM axis that is contiguous in the stored tensor
+// is an image axis q = n*spad + sp (sp < S valid, rest padding); element
+// (r, q) lives at base[(n*chan + r)*S + sp].  Lets conv GEMMs read x / dY
+// straight out of NCHW with no transpose/materialization pass.
+struct GemmView {
+  long spad = 0;  // 0 = plain operand
+  long S = 0;
+  long chan = 0;
+};
+
 void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
           float alpha, const float* A, long lda, const float* B, long ldb,
-          float beta, float* C, long ldc, const GemmEpi* epi = nullptr);
+          float beta, float* C, long ldc, const GemmEpi* epi = nullptr,
+          const GemmView* aview = nullptr, const GemmView* bview = nullptr);
+
+// db[c] = Σ_n Σ_s dy[n][c][s] (deterministic)
+void bias_grad(hipStream_t s, const float* dy, int N, int C, long S,
+               float* db);
 
 // col[K][Nimg*Spad] from x[Nimg][C][H][W]; pad columns zero-filled.
 void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
