@@ -26,30 +26,30 @@ static inline int nblocks(long n, int per_thread = 1) {
        i += (long)gridDim.x * blockDim.x)
 
 // ------------------------------------------------------------ im2col
-__global__ void k_im2col_b(const float* __restrict__ x, int Nimg, int C,
-                           int H, int W, int kh, int kw, int ph, int pw,
-                           int sh, int sw, int dh, int dw, int OH, int OW,
-                           long Spad, float* __restrict__ col) {
-  const long S = (long)OH * OW;
-  const long cols = (long)Nimg * Spad;
-  const long total = (long)C * kh * kw * cols;
-  GRID_STRIDE(idx, total) {
-    const long kck = idx / cols;        // row of col = (c, i, j)
-    const long cidx = idx - kck * cols; // column = n*Spad + sp
-    const int n = (int)(cidx / Spad);
-    const long sp = cidx - (long)n * Spad;
+// 3D grid: z = image, y = col row (c,ki,kj), x-threads sweep the output
+// pixels — all inner indexing 32-bit, writes fully coalesced along sp.
+__global__ void k_im2col_b(const float* __restrict__ x, int C, int H, int W,
+                           int kh, int kw, int ph, int pw, int sh, int sw,
+                           int dh, int dw, int OH, int OW, long Spad,
+                           long cols, float* __restrict__ col) {
+  const int n = blockIdx.z;
+  const int row = blockIdx.y;  // c*kh*kw + ki*kw + kj
+  const int c = row / (kh * kw);
+  const int ki = (row / kw) % kh;
+  const int kj = row % kw;
+  const int S = OH * OW;
+  const float* xp = x + ((long)n * C + c) * H * W;
+  float* cp = col + (long)row * cols + (long)n * Spad;
+  for (int sp = blockIdx.x * blockDim.x + threadIdx.x; sp < (int)Spad;
+       sp += gridDim.x * blockDim.x) {
     float v = 0.f;
     if (sp < S) {
-      const int c = (int)(kck / (kh * kw));
-      const int ki = (int)(kck / kw) % kh;
-      const int kj = (int)(kck % kw);
-      const int oh = (int)(sp / OW), ow = (int)(sp % OW);
+      const int oh = sp / OW, ow = sp - oh * OW;
       const int h = oh * sh - ph + ki * dh;
       const int w = ow * sw - pw + kj * dw;
-      if (h >= 0 && h < H && w >= 0 && w < W)
-        v = x[(((long)n * C + c) * H + h) * W + w];
+      if (h >= 0 && h < H && w >= 0 && w < W) v = xp[h * W + w];
     }
-    col[idx] = v;
+    cp[sp] = v;
   }
 }
 
@@ -59,23 +59,26 @@ void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
   const long total = (long)C * kh * kw * Nimg * Spad;
   PerfScope perf("im2col", s, 0,
                  8.0 * total);  // ~1 read + 1 write per element
-  hipLaunchKernelGGL(k_im2col_b, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
-                     x, Nimg, C, H, W, kh, kw, ph, pw, sh, sw, dh, dw, OH,
-                     OW, Spad, col);
+  const int bx = (int)std::min<long>((Spad + TPB - 1) / TPB, 16);
+  dim3 grid(bx, C * kh * kw, Nimg);
+  hipLaunchKernelGGL(k_im2col_b, grid, dim3(TPB), 0, s, x, C, H, W, kh, kw,
+                     ph, pw, sh, sw, dh, dw, OH, OW, Spad,
+                     (long)Nimg * Spad, col);
 }
 
 // gather col2im (reference im2col.cu:256-295 pattern — no atomics)
-__global__ void k_col2im_b(const float* __restrict__ col, int Nimg, int C,
-                           int H, int W, int kh, int kw, int ph, int pw,
-                           int sh, int sw, int dh, int dw, int OH, int OW,
-                           long Spad, float* __restrict__ dx) {
-  const long total = (long)Nimg * C * H * W;
-  const long cols = (long)Nimg * Spad;
-  GRID_STRIDE(idx, total) {
-    const int w = (int)(idx % W);
-    const int h = (int)((idx / W) % H);
-    const int c = (int)((idx / ((long)W * H)) % C);
-    const int n = (int)(idx / ((long)W * H * C));
+// 3D grid: z = image, y = channel, x-threads sweep H*W (32-bit indices)
+__global__ void k_col2im_b(const float* __restrict__ col, int C, int H,
+                           int W, int kh, int kw, int ph, int pw, int sh,
+                           int sw, int dh, int dw, int OH, int OW,
+                           long Spad, long cols, float* __restrict__ dx) {
+  const int n = blockIdx.z;
+  const int c = blockIdx.y;
+  const float* cp = col + (long)n * Spad;
+  float* dp = dx + ((long)n * C + c) * H * W;
+  for (int hw = blockIdx.x * blockDim.x + threadIdx.x; hw < H * W;
+       hw += gridDim.x * blockDim.x) {
+    const int h = hw / W, w = hw - h * W;
     float acc = 0.f;
     for (int i = 0; i < kh; ++i) {
       int hk = h + ph - i * dh;
@@ -87,11 +90,11 @@ __global__ void k_col2im_b(const float* __restrict__ col, int Nimg, int C,
         if (wk < 0 || wk % sw) continue;
         wk /= sw;
         if (wk >= OW) continue;
-        const long row = ((long)c * kh + i) * kw + j;
-        acc += col[row * cols + (long)n * Spad + (long)hk * OW + wk];
+        const int row = (c * kh + i) * kw + j;
+        acc += cp[(long)row * cols + hk * OW + wk];
       }
     }
-    dx[idx] = acc;
+    dp[hw] = acc;
   }
 }
 
@@ -100,9 +103,11 @@ void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
                     int dh, int dw, int OH, int OW, long Spad, float* dx) {
   const long total = (long)Nimg * C * H * W;
   PerfScope perf("col2im", s, 0, 8.0 * total * kh * kw / (sh * sw));
-  hipLaunchKernelGGL(k_col2im_b, dim3(nblocks(total, 2)), dim3(TPB), 0, s,
-                     dcol, Nimg, C, H, W, kh, kw, ph, pw, sh, sw, dh, dw,
-                     OH, OW, Spad, dx);
+  const int bx = (int)std::min<long>(((long)H * W + TPB - 1) / TPB, 16);
+  dim3 grid(bx, C, Nimg);
+  hipLaunchKernelGGL(k_col2im_b, grid, dim3(TPB), 0, s, dcol, C, H, W, kh,
+                     kw, ph, pw, sh, sw, dh, dw, OH, OW, Spad,
+                     (long)Nimg * Spad, dx);
 }
 
 // y[N][C][S] -> out[C][N*Spad] (zero pad)

@@ -309,7 +309,9 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     SK = (int)std::min<long>(
         {512 / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 64});
   }
-  PerfScope perf("gemm", s, 2.0 * M * N * K,
+  const char* pcls = !transA ? (!transB ? "gemm_nn" : "gemm_nt")
+                            : (!transB ? "gemm_tn" : "gemm_tt");
+  PerfScope perf(pcls, s, 2.0 * M * N * K,
                  4.0 * (M * K + N * K + M * N));
   dim3 grid((unsigned)g.tiles, 1, (unsigned)SK);
   dim3 block(256);
