@@ -268,7 +268,12 @@ def main():
         ca.device_synchronize()
         ca.set_perf_timing(False)
         perf = ca.perf_snapshot()
-        g = perf.get("gemm", {"flops": 0, "ns": 1, "launches": 0})
+        g = {"flops": 0.0, "ns": 0.0, "launches": 0}
+        for k, v in perf.items():
+            if k.startswith("gemm"):
+                g["flops"] += v["flops"]
+                g["ns"] += v["ns"]
+                g["launches"] += v["launches"]
         achieved = g["flops"] / max(g["ns"], 1) * 1e9  # FLOP/s
         roofline = {
             "bound": "mfma",
@@ -283,6 +288,10 @@ def main():
         tot_ns = sum(v["ns"] for v in perf.values())
         if tot_ns > 0:
             roofline["kernel_time_frac"] = g["ns"] / tot_ns
+        roofline["classes"] = {
+            k: {"ms": v["ns"] / 1e6,
+                "tf": v["flops"] / max(v["ns"], 1) / 1e3}
+            for k, v in sorted(perf.items(), key=lambda kv: -kv[1]["ns"])}
 
         cpu_baseline = None
         if world == 1 and not args.no_cpu_baseline:
