@@ -163,6 +163,8 @@ class ConvolutionLayer : public Layer {
   bool bias_ = true;
   int N_ = 0, C_ = 0, H_ = 0, W_ = 0, OH_ = 0, OW_ = 0;
   long S_ = 0, Spad_ = 0;  // spatial count and 64-padded count
+  int col_slot_ = -1;      // per-layer cached col buffer (fwd fills,
+                           // bwd reuses — no im2col recompute)
 };
 
 class InnerProductLayer : public Layer {

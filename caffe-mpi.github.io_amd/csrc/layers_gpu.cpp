@@ -58,8 +58,12 @@ void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
               S_, &epi, nullptr, &xv);
     return;
   }
+  if (col_slot_ < 0) {
+    static int next_slot = 100;  // per-conv-layer slots start at 100
+    col_slot_ = next_slot++;
+  }
   float* colb =
-      (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
+      (float*)ws.get(col_slot_, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
   gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_, sh_,
                       sw_, dh_, dw_, OH_, OW_, Spad_, colb);
   for (int g = 0; g < group_; ++g) {
@@ -105,10 +109,11 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
     return;
   }
 
+  // col buffer cached from this iteration's forward (col_slot_ assigned
+  // there); workspace grow-only so the pointer is stable
+  CHECK_GE_(col_slot_, 0) << "conv backward before forward";
   float* colb =
-      (float*)ws.get(0, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
-  gpu::im2col_batched(E.stream, x, N_, C_, H_, W_, kh_, kw_, ph_, pw_, sh_,
-                      sw_, dh_, dw_, OH_, OW_, Spad_, colb);
+      (float*)ws.get(col_slot_, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
   for (int g = 0; g < group_; ++g) {
     // wgrad: dW = dY-view · colᵀ; the group's dY channels start at
     // g*(Cout/group) — fold the channel offset into the view base pointer
@@ -210,7 +215,7 @@ void BatchNormLayer::Forward_gpu(const std::vector<Blob*>& bottom,
                      eps_, y);
     return;
   }
-  const int nb = gpu::bn_blocks_per_channel(N, S);
+  const int nb = gpu::bn_blocks_per_channel(N, C_);
   partials_.Reshape({(int)(C_ * nb * 4)});  // double2 = 4 floats
   void* parts = partials_.mutable_gpu_data();
   gpu::bn_fwd_stats(E.stream, x, N, C_, S, nb, parts);
@@ -233,7 +238,7 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
   const long S = bottom[0]->count() / ((long)N * C_);
   const float* x = bottom[0]->gpu_data();
   const float* dy = top[0]->gpu_diff();
-  const int nb = gpu::bn_blocks_per_channel(N, S);
+  const int nb = gpu::bn_blocks_per_channel(N, C_);
   partials_.Reshape({(int)(C_ * nb * 4)});
   void* parts = partials_.mutable_gpu_data();
   gpu::bn_bwd_stats(E.stream, x, dy, mean_.gpu_data(), inv_std_.gpu_data(),

@@ -99,7 +99,7 @@ void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
 // Fused BatchNorm (replaces the reference's ~10-launch GEMV chain,
 // batch_norm_layer.hpp:96-120, with 3 kernels fwd / 3 bwd).
 // partials: double2[nb*C] workspace. mean/var/inv_std: float[C] device.
-int bn_blocks_per_channel(int N, long S);
+int bn_blocks_per_channel(int N, int C);
 void bn_fwd_stats(hipStream_t s, const float* x, int N, int C, long S,
                   int nb, void* partials);
 void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
