@@ -329,14 +329,15 @@ __global__ void k_bn_fwd_stats(const float* __restrict__ x, int N, int C,
   const int slice = blockIdx.x / C;
   const int n0 = (int)((long)N * slice / nb);
   const int n1 = (int)((long)N * (slice + 1) / nb);
+  const int Si = (int)S;
+  const int span = (n1 - n0) * Si;
   double s1 = 0, s2 = 0;
-  for (int n = n0; n < n1; ++n) {
-    const float* xp = x + ((long)n * C + c) * S;
-    for (long sp = threadIdx.x; sp < S; sp += blockDim.x) {
-      const double v = xp[sp];
-      s1 += v;
-      s2 += v * v;
-    }
+  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+    const int n = n0 + i / Si;
+    const int sp = i - (n - n0) * Si;
+    const double v = x[((long)n * C + c) * S + sp];
+    s1 += v;
+    s2 += v * v;
   }
   __shared__ double sh1[TPB], sh2[TPB];
   sh1[threadIdx.x] = s1;
@@ -389,14 +390,14 @@ __global__ void k_bn_fwd_norm(const float* __restrict__ x,
                               const float* __restrict__ mean,
                               const float* __restrict__ inv_std,
                               const float* __restrict__ scale,
-                              const float* __restrict__ bias, int sb, int N,
-                              int C, long S, float* __restrict__ y) {
-  const int c = blockIdx.y;
-  const long base = ((long)blockIdx.z * gridDim.y + c) * S;
+                              const float* __restrict__ bias, int sb, int C,
+                              int S, float* __restrict__ y) {
+  const int c = blockIdx.x % C;  // blockIdx.x = n*C + c? no: row = n*C+c
+  const long base = (long)blockIdx.x * S;
   const float mu = mean[c], inv = inv_std[c];
   const float sc = sb ? scale[c] : 1.f, bi = sb ? bias[c] : 0.f;
-  for (long sp = blockIdx.x * (long)blockDim.x + threadIdx.x; sp < S;
-       sp += (long)gridDim.x * blockDim.x)
+  for (int sp = blockIdx.y * blockDim.x + threadIdx.x; sp < S;
+       sp += gridDim.y * blockDim.x)
     y[base + sp] = (x[base + sp] - mu) * inv * sc + bi;
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
@@ -404,9 +405,10 @@ void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  int sb, int N, int C, long S, float* y) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 8.0 * total);
-  const int bx = (int)std::min<long>((S + TPB - 1) / TPB, 16);
-  hipLaunchKernelGGL(k_bn_fwd_norm, dim3(bx, C, N), dim3(TPB), 0, s, x,
-                     mean, inv_std, scale, bias, sb, N, C, S, y);
+  const int bsz = S < 128 ? 64 : TPB;
+  const int by = (int)std::min<long>((S + bsz - 1) / bsz, 16);
+  hipLaunchKernelGGL(k_bn_fwd_norm, dim3(N * C, by), dim3(bsz), 0, s, x,
+                     mean, inv_std, scale, bias, sb, C, (int)S, y);
 }
 
 __global__ void k_bn_moving_avg(const float* __restrict__ mean,
@@ -464,15 +466,16 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
   const int slice = blockIdx.x / C;
   const int n0 = (int)((long)N * slice / nb);
   const int n1 = (int)((long)N * (slice + 1) / nb);
+  const int Si = (int)S;
+  const int span = (n1 - n0) * Si;
   const float m = mean[c], inv = inv_std[c];
   double s_dy = 0, s_dyxn = 0;
-  for (int n = n0; n < n1; ++n) {
-    const long off0 = ((long)n * C + c) * S;
-    for (long sp = threadIdx.x; sp < S; sp += blockDim.x) {
-      const double d = dy[off0 + sp];
-      s_dy += d;
-      s_dyxn += d * (double)((x[off0 + sp] - m) * inv);
-    }
+  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+    const int n = n0 + i / Si;
+    const long off = ((long)n * C + c) * S + (i - (n - n0) * Si);
+    const double d = dy[off];
+    s_dy += d;
+    s_dyxn += d * (double)((x[off] - m) * inv);
   }
   __shared__ double sh1[TPB], sh2[TPB];
   sh1[threadIdx.x] = s_dy;
@@ -531,15 +534,15 @@ __global__ void k_bn_bwd_apply(const float* __restrict__ x,
                                const float* __restrict__ inv_std,
                                const float* __restrict__ scale, int sb,
                                const float* __restrict__ m_dy,
-                               const float* __restrict__ m_dyxn, int N,
-                               int C, long S, float* __restrict__ dx) {
-  const int c = blockIdx.y;
-  const long base = ((long)blockIdx.z * gridDim.y + c) * S;
+                               const float* __restrict__ m_dyxn, int C,
+                               int S, float* __restrict__ dx) {
+  const int c = blockIdx.x % C;
+  const long base = (long)blockIdx.x * S;
   const float mu = mean[c], inv = inv_std[c];
   const float sc = sb ? scale[c] : 1.f;
   const float mdy = m_dy[c], mdyxn = m_dyxn[c];
-  for (long sp = blockIdx.x * (long)blockDim.x + threadIdx.x; sp < S;
-       sp += (long)gridDim.x * blockDim.x) {
+  for (int sp = blockIdx.y * blockDim.x + threadIdx.x; sp < S;
+       sp += gridDim.y * blockDim.x) {
     const float xn = (x[base + sp] - mu) * inv;
     dx[base + sp] = (dy[base + sp] * sc - mdy - mdyxn * xn) * inv;
   }
@@ -550,9 +553,10 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* m_dyxn, int N, int C, long S, float* dx) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 12.0 * total);
-  const int bx = (int)std::min<long>((S + TPB - 1) / TPB, 16);
-  hipLaunchKernelGGL(k_bn_bwd_apply, dim3(bx, C, N), dim3(TPB), 0, s, x,
-                     dy, mean, inv_std, scale, sb, m_dy, m_dyxn, N, C, S,
+  const int bsz = S < 128 ? 64 : TPB;
+  const int by = (int)std::min<long>((S + bsz - 1) / bsz, 16);
+  hipLaunchKernelGGL(k_bn_bwd_apply, dim3(N * C, by), dim3(bsz), 0, s, x,
+                     dy, mean, inv_std, scale, sb, m_dy, m_dyxn, C, (int)S,
                      dx);
 }
 
