@@ -895,6 +895,14 @@ void SplitLayer::Backward_gpu(const std::vector<Blob*>& top,
   Engine& E = Engine::get();
   const long n = bottom[0]->count();
   float* dx = bottom[0]->mutable_gpu_diff();
+  if (top.size() == 1) {
+    bottom[0]->ShareDiff(*top[0]);
+    return;
+  }
+  if (top.size() == 2) {  // single fused pass for the common fan-out of 2
+    gpu::add3(E.stream, n, top[0]->gpu_diff(), top[1]->gpu_diff(), dx);
+    return;
+  }
   gpu::copy(E.stream, n, top[0]->gpu_diff(), dx);
   for (size_t i = 1; i < top.size(); ++i)
     gpu::acc(E.stream, n, top[i]->gpu_diff(), dx);

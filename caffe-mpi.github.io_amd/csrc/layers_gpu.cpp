@@ -297,14 +297,21 @@ void EltwiseLayer::Backward_gpu(const std::vector<Blob*>& top,
                                 const std::vector<Blob*>& bottom) {
   Engine& E = Engine::get();
   const long n = top[0]->count();
+  bool all_one = true;
+  for (float c : coeffs_) all_one = all_one && c == 1.f;
+  if (all_one) {
+    // dx_i == dy exactly: alias the diffs instead of copying (zero kernels;
+    // safe because each bottom's diff is consumed before anything rewrites
+    // the shared buffer — backward runs top-down on one stream)
+    for (size_t i = 0; i < bottom.size(); ++i)
+      if (prop_down[i]) bottom[i]->ShareDiff(*top[0]);
+    return;
+  }
   const float* dy = top[0]->gpu_diff();
   for (size_t i = 0; i < bottom.size(); ++i) {
     if (!prop_down[i]) continue;
     float* dx = bottom[i]->mutable_gpu_diff();
-    if (coeffs_[i] == 1.f)
-      gpu::copy(E.stream, n, dy, dx);
-    else
-      gpu::axpby(E.stream, n, coeffs_[i], dy, 0.f, dx);
+    gpu::axpby(E.stream, n, coeffs_[i], dy, 0.f, dx);
   }
 }
 
