@@ -23,10 +23,19 @@ WORKER = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                       "_dist_worker.py")
 
 
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
 def run_dist(nproc, args):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
-    env["MASTER_PORT"] = "29517"
+    env["MASTER_PORT"] = str(_free_port())
     procs = []
     for rank in range(nproc):
         env_r = dict(env, RANK=str(rank), WORLD_SIZE=str(nproc),
