@@ -226,28 +226,34 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   // ---- epilogue: acc reg r -> (row, col)
   // C/D map (32x32 MFMA): col = lane&31, row = (r&3)+8*(r>>2)+4*(lane>>5)
   auto epi_tile = [&](const f32x16& a, int ti, int tj) {
+    // col is constant across the 16 regs — hoist the scatter division
+    const long col = n0 + wc * 64 + tj * 32 + row_in;
+    if (col >= g.N) return;
+    long col_base = 0;  // n*n_stride + sp for scatter, col for plain
+    bool col_ok = true;
+    if (!SPLITK && g.spad > 0) {
+      const long n = col / g.spad;
+      const long sp = col - n * g.spad;
+      col_ok = sp < g.S;
+      col_base = n * g.n_stride + sp;
+    }
+    if (!col_ok) return;
+    const float cbias =
+        (!SPLITK && g.bias && g.bias_per_col) ? g.bias[col] : 0.f;
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const long row =
           m0 + wr * 64 + ti * 32 + ((r & 3) + 8 * (r >> 2) + 4 * ksel);
-      const long col = n0 + wc * 64 + tj * 32 + row_in;
-      if (row >= g.M || col >= g.N) continue;
+      if (row >= g.M) continue;
       float v = g.alpha * a[r];
       if (SPLITK) {
         g.slab[((long)blockIdx.z * g.M + row) * g.N + col] = v;
         continue;
       }
-      if (g.bias) v += g.bias[g.bias_per_col ? col : row];
+      if (g.bias) v += g.bias_per_col ? cbias : g.bias[row];
       if (g.relu) v = fmaxf(v, 0.f);
-      long off;
-      if (g.spad > 0) {
-        const long n = col / g.spad;
-        const long sp = col - n * g.spad;
-        if (sp >= g.S) continue;  // padding column
-        off = n * g.n_stride + row * g.S + sp;
-      } else {
-        off = row * g.ldc + col;
-      }
+      const long off = g.spad > 0 ? col_base + row * g.S
+                                  : row * g.ldc + col;
       if (g.beta != 0.f) v += g.beta * g.C[off];
       g.C[off] = v;
     }

@@ -218,7 +218,9 @@ void ConvolutionLayer::Reshape(const std::vector<Blob*>& bottom,
   CHECK_GT_(OH_, 0);
   CHECK_GT_(OW_, 0);
   S_ = (long)OH_ * OW_;
-  Spad_ = (S_ + 63) / 64 * 64;
+  // pad S to 16 (the GEMM staging chunk width): view chunks stay inside one
+  // image and padding waste is <= 15 columns per image
+  Spad_ = (S_ + 15) / 16 * 16;
   top[0]->Reshape({N_, Cout_, OH_, OW_});
 }
 
