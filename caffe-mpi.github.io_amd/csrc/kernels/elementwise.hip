@@ -391,24 +391,28 @@ __global__ void k_bn_fwd_norm(const float* __restrict__ x,
                               const float* __restrict__ inv_std,
                               const float* __restrict__ scale,
                               const float* __restrict__ bias, int sb, int C,
-                              int S, float* __restrict__ y) {
-  const int c = blockIdx.x % C;  // blockIdx.x = n*C + c? no: row = n*C+c
+                              int S, int frelu, float* __restrict__ y) {
+  const int c = blockIdx.x % C;
   const long base = (long)blockIdx.x * S;
   const float mu = mean[c], inv = inv_std[c];
   const float sc = sb ? scale[c] : 1.f, bi = sb ? bias[c] : 0.f;
   for (int sp = blockIdx.y * blockDim.x + threadIdx.x; sp < S;
-       sp += gridDim.y * blockDim.x)
-    y[base + sp] = (x[base + sp] - mu) * inv * sc + bi;
+       sp += gridDim.y * blockDim.x) {
+    float v = (x[base + sp] - mu) * inv * sc + bi;
+    if (frelu) v = fmaxf(v, 0.f);
+    y[base + sp] = v;
+  }
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
-                 int sb, int N, int C, long S, float* y) {
+                 int sb, int N, int C, long S, float* y, int fuse_relu) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 8.0 * total);
   const int bsz = S < 128 ? 64 : TPB;
   const int by = (int)std::min<long>((S + bsz - 1) / bsz, 16);
   hipLaunchKernelGGL(k_bn_fwd_norm, dim3(N * C, by), dim3(bsz), 0, s, x,
-                     mean, inv_std, scale, bias, sb, C, (int)S, y);
+                     mean, inv_std, scale, bias, sb, C, (int)S, fuse_relu,
+                     y);
 }
 
 __global__ void k_bn_moving_avg(const float* __restrict__ mean,
@@ -819,14 +823,23 @@ void set_const(hipStream_t s, long n, float v, float* y) {
 }
 
 __global__ void k_add3(long n4, const f4* __restrict__ a,
-                       const f4* __restrict__ b, f4* __restrict__ y) {
-  VEC_GRID(i, n4) y[i] = a[i] + b[i];
+                       const f4* __restrict__ b, int frelu,
+                       f4* __restrict__ y) {
+  VEC_GRID(i, n4) {
+    f4 v = a[i] + b[i];
+    if (frelu) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
+    }
+    y[i] = v;
+  }
 }
-void add3(hipStream_t s, long n, const float* a, const float* b, float* y) {
+void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
+          int fuse_relu) {
   PerfScope perf("eltwise", s, 0, 12.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_add3, dim3(nblocks(n4, 4)), dim3(TPB), 0, s, n4,
-                     (const f4*)a, (const f4*)b, (f4*)y);
+                     (const f4*)a, (const f4*)b, fuse_relu, (f4*)y);
 }
 
 __global__ void k_acc(long n4, const f4* __restrict__ x,

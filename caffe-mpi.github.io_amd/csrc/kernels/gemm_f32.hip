@@ -214,7 +214,9 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
       acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
       acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
     }
-    __syncthreads();
+    // single barrier per K-tile: writing buf[cur^1] here is safe — its
+    // last readers finished before the previous iteration's barrier; the
+    // barrier below publishes these writes for the next iteration's reads
     if (t + 1 < ntiles) {
       stage_a_write<TA>(As[cur ^ 1], ra);
       stage_b_write<TB>(Bs[cur ^ 1], rb);

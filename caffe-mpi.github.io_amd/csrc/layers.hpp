@@ -226,6 +226,7 @@ class BatchNormLayer : public Layer {
 
   float maf_ = 0.999f, eps_ = 1e-5f;
   bool scale_bias_ = false, use_global_ = false;
+  bool fuse_relu_ = false;  // GPU graph fusion: BN+ReLU forward in one pass
   int C_ = 0;
   long iter_ = 0;
   Blob mean_, var_, inv_std_, m_dy_, m_dyxn_, partials_;
@@ -234,6 +235,7 @@ class BatchNormLayer : public Layer {
 class ReLULayer : public Layer {
  public:
   using Layer::Layer;
+  bool fused_away_ = false;  // producer applies the ReLU in its epilogue
   void Reshape(const std::vector<Blob*>& b,
                const std::vector<Blob*>& t) override {
     if (b[0] != t[0]) t[0]->ReshapeLike(*b[0]);
@@ -267,6 +269,7 @@ class EltwiseLayer : public Layer {
                     const std::vector<Blob*>&) override;
   std::vector<float> coeffs_;
   std::string op_ = "SUM";
+  bool fuse_relu_ = false;  // GPU graph fusion: SUM+ReLU forward in one pass
 };
 
 class LRNLayer : public Layer {

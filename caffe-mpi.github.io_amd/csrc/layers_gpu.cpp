@@ -223,7 +223,7 @@ void BatchNormLayer::Forward_gpu(const std::vector<Blob*>& bottom,
                        mean_.mutable_gpu_data(), var_.mutable_gpu_data(),
                        inv_std_.mutable_gpu_data());
   gpu::bn_fwd_norm(E.stream, x, mean_.gpu_data(), inv_std_.gpu_data(), sc,
-                   bi, scale_bias_, N, C_, S, y);
+                   bi, scale_bias_, N, C_, S, y, fuse_relu_);
   gpu::bn_moving_avg(E.stream, mean_.gpu_data(), var_.gpu_data(), C_, maf_,
                      iter_ <= 1 ? 1 : 0, blobs_[0]->mutable_gpu_data(),
                      blobs_[1]->mutable_gpu_data());
@@ -260,6 +260,7 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
 // ------------------------------------------------------------------ ReLU
 void ReLULayer::Forward_gpu(const std::vector<Blob*>& bottom,
                             const std::vector<Blob*>& top) {
+  if (fused_away_) return;  // producer already applied the activation
   auto rp = param_->sub("relu_param");
   const float slope = rp ? (float)rp->num("negative_slope", 0) : 0.f;
   gpu::relu_fwd(Engine::get().stream, bottom[0]->gpu_data(),
@@ -284,7 +285,8 @@ void EltwiseLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   const long n = top[0]->count();
   float* y = top[0]->mutable_gpu_data();
   if (bottom.size() == 2 && coeffs_[0] == 1.f && coeffs_[1] == 1.f) {
-    gpu::add3(E.stream, n, bottom[0]->gpu_data(), bottom[1]->gpu_data(), y);
+    gpu::add3(E.stream, n, bottom[0]->gpu_data(), bottom[1]->gpu_data(), y,
+              fuse_relu_);
     return;
   }
   gpu::set_const(E.stream, n, 0.f, y);

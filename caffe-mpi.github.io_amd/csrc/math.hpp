@@ -107,7 +107,8 @@ void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
                      float* inv_std);
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
-                 int scale_bias, int N, int C, long S, float* y);
+                 int scale_bias, int N, int C, long S, float* y,
+                 int fuse_relu = 0);
 void bn_moving_avg(hipStream_t s, const float* mean, const float* var, int C,
                    float maf, int copy_only, float* gmean, float* gvar);
 void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,
@@ -149,7 +150,8 @@ void colsum(hipStream_t s, const float* A, long M, long N, float* out);
 void axpby(hipStream_t s, long n, float a, const float* x, float b, float* y);
 void copy(hipStream_t s, long n, const float* x, float* y);
 void set_const(hipStream_t s, long n, float v, float* y);
-void add3(hipStream_t s, long n, const float* a, const float* b, float* y);
+void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
+          int fuse_relu = 0);
 // y[n0..] accumulate: y += x
 void acc(hipStream_t s, long n, const float* x, float* y);
 

@@ -188,7 +188,36 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
       params_.push_back(p);
     }
   }
-  if (Engine::get().mode == Mode::GPU) setup_arena();
+  // GPU-mode forward fusion: an in-place slope-0 ReLU directly after a
+  // BatchNorm or Eltwise(SUM) is applied in the producer's epilogue; the
+  // ReLU layer keeps its backward (it reads its own output, which the
+  // fused producer already wrote post-activation)
+  if (Engine::get().mode == Mode::GPU) {
+    for (size_t i = 1; i < layers_.size(); ++i) {
+      auto* relu = dynamic_cast<ReLULayer*>(layers_[i].get());
+      if (!relu) continue;
+      if (bottoms_[i].size() != 1 || tops_[i].size() != 1) continue;
+      if (bottoms_[i][0] != tops_[i][0]) continue;  // must be in-place
+      auto rp = layers_[i]->param()->sub("relu_param");
+      if (rp && rp->num("negative_slope", 0) != 0) continue;
+      // producer = previous layer producing this blob
+      if (tops_[i - 1].size() != 1 || tops_[i - 1][0] != bottoms_[i][0])
+        continue;
+      if (auto* bn = dynamic_cast<BatchNormLayer*>(layers_[i - 1].get())) {
+        bn->fuse_relu_ = true;
+        relu->fused_away_ = true;
+      } else if (auto* el =
+                     dynamic_cast<EltwiseLayer*>(layers_[i - 1].get())) {
+        bool ones = el->op_ == "SUM";
+        for (float c : el->coeffs_) ones = ones && c == 1.f;
+        if (ones && bottoms_[i - 1].size() == 2) {
+          el->fuse_relu_ = true;
+          relu->fused_away_ = true;
+        }
+      }
+    }
+    setup_arena();
+  }
 }
 
 void Net::setup_arena() {
