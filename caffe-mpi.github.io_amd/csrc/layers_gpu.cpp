@@ -27,14 +27,20 @@ void DataLayer::Forward_gpu(const std::vector<Blob*>&,
 }
 
 // ------------------------------------------------------------------ Conv
-// MI355X-first conv: whole-batch GEMMs with NCHW-*view* operands — dY and
-// (for 1x1/s1) x are read straight out of NCHW inside the GEMM staging, so
-// there is no transpose pass anywhere and 1x1/s1 convolutions run with no
-// im2col/col2im at all.  Non-1x1 convs keep an explicit batched col buffer
-// [K][N*Spad] (implicit-GEMM fusion is the next step).
+// MI355X-first conv, two paths:
+//  - IMPLICIT GEMM (stride 1, dilation 1 — every conv that needs
+//    backward-data in the four models): the GEMM staging reads x / dY
+//    straight from NCHW through im2col/NCHW views; no col buffer, no
+//    im2col/col2im kernels; backward-data runs as a forward convolution
+//    over dY with flipped/transposed weights (Wt), scattered into dx.
+//  - explicit batched col buffer [K][N*Spad] for strided convs (cached per
+//    layer: forward fills, backward reuses).
+static bool conv_s1d1(const ConvolutionLayer& l) {
+  return l.sh_ == 1 && l.sw_ == 1 && l.dh_ == 1 && l.dw_ == 1;
+}
 static bool conv_is_1x1(const ConvolutionLayer& l) {
-  return l.kh_ == 1 && l.kw_ == 1 && l.sh_ == 1 && l.sw_ == 1 && !l.ph_ &&
-         !l.pw_ && l.group_ == 1;
+  return l.kh_ == 1 && l.kw_ == 1 && conv_s1d1(l) && !l.ph_ && !l.pw_ &&
+         l.group_ == 1;
 }
 
 void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
@@ -53,13 +59,27 @@ void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   epi.n_stride = (long)Cout_ * S_;
   epi.bias = bias_ ? blobs_[1]->gpu_data() : nullptr;
   if (conv_is_1x1(*this)) {
-    GemmView xv{Spad_, S_, C_};  // x[N][C][S] as [C][N*Spad]
+    GemmView xv{Spad_, S_, C_};  // plain NCHW view
     gpu::gemm(E.stream, false, false, Cout_, NS, K, 1.f, w, K, x, 0, 0.f, y,
               S_, &epi, nullptr, &xv);
     return;
   }
+  if (conv_s1d1(*this)) {  // implicit im2col view
+    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_};
+    for (int g = 0; g < group_; ++g) {
+      epi.bias = bias_ ? blobs_[1]->gpu_data() + (long)g * (Cout_ / group_)
+                       : nullptr;
+      gpu::gemm(E.stream, false, false, Cout_ / group_, NS, K, 1.f,
+                w + (long)g * (Cout_ / group_) * K, K,
+                x + (long)g * (C_ / group_) * H_ * W_, 0, 0.f,
+                y + (long)g * (Cout_ / group_) * S_, S_, &epi, nullptr,
+                &xv);
+    }
+    return;
+  }
+  // strided: explicit col (cached for backward)
   if (col_slot_ < 0) {
-    static int next_slot = 100;  // per-conv-layer slots start at 100
+    static int next_slot = 100;
     col_slot_ = next_slot++;
   }
   float* colb =
@@ -94,11 +114,9 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
 
   if (conv_is_1x1(*this)) {
     GemmView xv{Spad_, S_, C_};
-    // wgrad: dW[Cout][C] = dY-view · x-viewᵀ (Kd = N*Spad, split-K capable)
     gpu::gemm(E.stream, false, true, Cout_, K, NS, 1.f, dy, 0, x, 0, 0.f,
               blobs_[0]->mutable_gpu_diff(), K, nullptr, &dyv, &xv);
     if (prop_down[0]) {
-      // dgrad: dx = Wᵀ · dY-view, scattered straight into NCHW
       GemmEpi epi;
       epi.spad = Spad_;
       epi.S = S_;
@@ -109,22 +127,57 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
     return;
   }
 
-  // col buffer cached from this iteration's forward (col_slot_ assigned
-  // there); workspace grow-only so the pointer is stable
+  if (conv_s1d1(*this)) {
+    // wgrad: dW = dY-view · (implicit col of x)ᵀ
+    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_};
+    for (int g = 0; g < group_; ++g)
+      gpu::gemm(E.stream, false, true, Cout_ / group_, K, NS, 1.f,
+                dy + (long)g * (Cout_ / group_) * S_, 0,
+                x + (long)g * (C_ / group_) * H_ * W_, 0, 0.f,
+                blobs_[0]->mutable_gpu_diff() +
+                    (long)g * (Cout_ / group_) * K,
+                K, nullptr, &dyv, &xv);
+    if (prop_down[0]) {
+      // dgrad = forward conv of dY with flipped/transposed weights:
+      // dx[ci] = Σ_{co,ki,kj} dy[co][h-ki+ (kh-1-ph) ...] · Wt
+      float* wt =
+          (float*)ws.get(11, sizeof(float) * blobs_[0]->count());
+      gpu::weight_flip_grouped(E.stream, w, Cout_, C_ / group_, kh_, kw_,
+                               group_, wt);
+      const long S_in = (long)H_ * W_;
+      const long spad_in = (S_in + 15) / 16 * 16;
+      const long NSin = (long)N_ * spad_in;
+      const int Kd = Cout_ / group_ * kh_ * kw_;
+      GemmEpi epi;
+      epi.spad = spad_in;
+      epi.S = S_in;
+      epi.n_stride = (long)C_ * S_in;
+      // dY viewed with the transposed-conv geometry: input dims (OH,OW),
+      // pad (k-1-p), output dims (H,W)
+      GemmView dyc{spad_in, S_in, Cout_, kh_, kw_, kh_ - 1 - ph_,
+                   kw_ - 1 - pw_, OH_, OW_, W_};
+      for (int g = 0; g < group_; ++g)
+        gpu::gemm(E.stream, false, false, C_ / group_, NSin, Kd, 1.f,
+                  wt + (long)g * (C_ / group_) * Kd, Kd,
+                  dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+                  bottom[0]->mutable_gpu_diff() +
+                      (long)g * (C_ / group_) * S_in,
+                  S_in, &epi, nullptr, &dyc);
+    }
+    return;
+  }
+
+  // strided: explicit col cached from this iteration's forward
   CHECK_GE_(col_slot_, 0) << "conv backward before forward";
   float* colb =
       (float*)ws.get(col_slot_, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
-  for (int g = 0; g < group_; ++g) {
-    // wgrad: dW = dY-view · colᵀ; the group's dY channels start at
-    // g*(Cout/group) — fold the channel offset into the view base pointer
-    GemmView dyvg{Spad_, S_, Cout_};
+  for (int g = 0; g < group_; ++g)
     gpu::gemm(E.stream, false, true, Cout_ / group_, K, NS, 1.f,
               dy + (long)g * (Cout_ / group_) * S_, 0,
               colb + (long)g * K * NS, NS, 0.f,
               blobs_[0]->mutable_gpu_diff() +
                   (long)g * (Cout_ / group_) * K,
-              K, nullptr, &dyvg);
-  }
+              K, nullptr, &dyv);
   if (prop_down[0]) {
     float* dcol =
         (float*)ws.get(1, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);

@@ -52,11 +52,28 @@ struct GemmEpi {
 // is an image axis q = n*spad + sp (sp < S valid, rest padding); element
 // (r, q) lives at base[(n*chan + r)*S + sp].  Lets conv GEMMs read x / dY
 // straight out of NCHW with no transpose/materialization pass.
+//
+// With kh > 0 the view becomes an IMPLICIT-IM2COL view (the north_star's
+// implicit-GEMM conv): r indexes a col row (c, ki, kj), sp an output pixel
+// (oh, ow); the element is x[n][c][oh*sh-ph+ki][ow*sw-pw+kj] (0 outside) —
+// the col matrix is never materialized.  stride/dilation 1 only (every
+// ResNet/GoogLeNet/AlexNet conv that needs backward-data is s1/d1; strided
+// convs use the explicit col path).
 struct GemmView {
   long spad = 0;  // 0 = plain operand
-  long S = 0;
-  long chan = 0;
+  long S = 0;     // valid pixels per image (OH*OW)
+  long chan = 0;  // channels of the viewed tensor (address stride term)
+  int kh = 0, kw = 0, ph = 0, pw = 0;  // kh>0 => implicit im2col
+  int H = 0, W = 0, OW = 0;            // input dims / output row width
 };
+
+// Wt[ci][co*kh*kw + ki*kw + kj] = W[co][ci][kh-1-ki][kw-1-kj] — the
+// flipped/transposed weights that turn backward-data into a forward
+// convolution over dY (s1 only)
+void weight_flip(hipStream_t s, const float* w, int Cout, int Cin, int kh,
+                 int kw, float* wt);
+void weight_flip_grouped(hipStream_t s, const float* w, int Cout, int Cin_g,
+                         int kh, int kw, int groups, float* wt);
 
 void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
           float alpha, const float* A, long lda, const float* B, long ldb,
