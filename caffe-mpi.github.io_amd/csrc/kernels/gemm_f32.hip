@@ -79,6 +79,26 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
     int w = ow - v.pw + kj;
     const float* xp = P + (n * v.chan + c) * (long)v.H * v.W;
     const long smax = v.S - sp;  // elements left in this image
+    // fast path: chunk stays in one output row, fully interior
+    if (smax >= 16 && ow + 16 <= v.OW && h >= 0 && h < v.H && w >= 0 &&
+        w + 15 < v.W) {
+      const float* p = xp + h * v.W + w;
+      if ((((uintptr_t)p) & 15) == 0) {
+        const f32x4* p4 = (const f32x4*)p;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const f32x4 t = p4[j];
+          out[4 * j + 0] = t.x;
+          out[4 * j + 1] = t.y;
+          out[4 * j + 2] = t.z;
+          out[4 * j + 3] = t.w;
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) out[j] = p[j];
+      }
+      return;
+    }
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       const bool ok = j < smax && h >= 0 && h < v.H && w >= 0 && w < v.W;
@@ -431,32 +451,48 @@ void weight_flip_grouped(hipStream_t s, const float* w, int Cout, int Cin_g,
                      dim3(256), 0, s, w, Cout, Cin_g, kh, kw, groups, wt);
 }
 
-// db[c] = Σ_n Σ_s dy[n][c][s] — one block per channel, double block-reduce
-__global__ void k_bias_grad(const float* __restrict__ dy, int N, int C,
-                            long S, float* __restrict__ db) {
-  for (int c = blockIdx.x; c < C; c += gridDim.x) {
+// db[c] = Σ_n Σ_s dy[n][c][s] — image-sliced partials + fixed-order sum
+__global__ void k_bias_grad_part(const float* __restrict__ dy, int N, int C,
+                                 int S, int nb, double* __restrict__ part) {
+  const int c = blockIdx.x % C;
+  const int slice = blockIdx.x / C;
+  const int n0 = (int)((long)N * slice / nb);
+  const int n1 = (int)((long)N * (slice + 1) / nb);
+  const int span = (n1 - n0) * S;
+  double acc = 0;
+  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+    const int n = n0 + i / S;
+    const int sp = i - (n - n0) * S;
+    acc += dy[((long)n * C + c) * S + sp];
+  }
+  __shared__ double sh[256];
+  sh[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) part[(long)c * nb + slice] = sh[0];
+}
+__global__ void k_bias_grad_fin(const double* __restrict__ part, int nb,
+                                int C, float* __restrict__ db) {
+  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += gridDim.x * blockDim.x) {
     double acc = 0;
-    for (long i = threadIdx.x; i < (long)N * S; i += blockDim.x) {
-      const long n = i / S;
-      const long sp = i - n * S;
-      acc += dy[(n * C + c) * S + sp];
-    }
-    __shared__ double sh[256];
-    sh[threadIdx.x] = acc;
-    __syncthreads();
-    for (int off = 128; off > 0; off >>= 1) {
-      if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
-      __syncthreads();
-    }
-    if (threadIdx.x == 0) db[c] = (float)sh[0];
-    __syncthreads();
+    for (int b = 0; b < nb; ++b) acc += part[(long)c * nb + b];
+    db[c] = (float)acc;
   }
 }
 void bias_grad(hipStream_t s, const float* dy, int N, int C, long S,
                float* db) {
   PerfScope perf("reduce", s, 0, 4.0 * N * C * S);
-  hipLaunchKernelGGL(k_bias_grad, dim3(std::min(C, 2048)), dim3(256), 0, s,
-                     dy, N, C, S, db);
+  const int nb = std::max(1, std::min(N, 2048 / std::max(1, C)));
+  double* part = (double*)Workspace::get_global().get(
+      12, sizeof(double) * (size_t)C * nb);
+  hipLaunchKernelGGL(k_bias_grad_part, dim3(C * nb), dim3(256), 0, s, dy, N,
+                     C, (int)S, nb, part);
+  hipLaunchKernelGGL(k_bias_grad_fin, dim3(1), dim3(256), 0, s, part, nb, C,
+                     db);
 }
 
 }  // namespace gpu
