@@ -110,6 +110,10 @@ def main():
         for i in range(net.num_params()):
             vals.extend(net.param(i)[:8].tolist())
         print("PARAMS " + " ".join(f"{v:.8e}" for v in vals))
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+        dist.destroy_process_group()
 
 
 if __name__ == "__main__":
