@@ -52,6 +52,10 @@ _lib.caffe_net_blob_shape.argtypes = [_vp, ctypes.c_char_p,
                                       ctypes.c_int,
                                       ctypes.POINTER(ctypes.c_int)]
 _lib.caffe_net_num_params.argtypes = [_vp]
+_lib.caffe_solver_snapshot.argtypes = [_vp]
+_lib.caffe_solver_restore.argtypes = [_vp, ctypes.c_char_p]
+_lib.caffe_net_save_weights.argtypes = [_vp, ctypes.c_char_p]
+_lib.caffe_net_load_weights.argtypes = [_vp, ctypes.c_char_p]
 _lib.caffe_solver_create.restype = ctypes.c_void_p
 _lib.caffe_solver_create_from_text.restype = ctypes.c_void_p
 _lib.caffe_solver_net.restype = ctypes.c_void_p

@@ -117,6 +117,34 @@ caffe_net_t caffe_solver_net(caffe_solver_t s) {
   return (caffe_net_t)&S(s)->net();
 }
 
+int caffe_solver_snapshot(caffe_solver_t s) {
+  API_TRY
+  S(s)->Snapshot();
+  return 0;
+  API_CATCH
+}
+
+int caffe_solver_restore(caffe_solver_t s, const char* state_path) {
+  API_TRY
+  S(s)->Restore(state_path);
+  return 0;
+  API_CATCH
+}
+
+int caffe_net_save_weights(caffe_net_t n, const char* path) {
+  API_TRY
+  N(n)->SaveWeights(path);
+  return 0;
+  API_CATCH
+}
+
+int caffe_net_load_weights(caffe_net_t n, const char* path) {
+  API_TRY
+  N(n)->LoadWeights(path);
+  return 0;
+  API_CATCH
+}
+
 int caffe_comm_unique_id(uint8_t out[128]) {
   API_TRY
   rccl_unique_id(out);

@@ -134,6 +134,10 @@ class Engine {
   // synthetic-data configuration (no LMDB datasets in this environment)
   bool synthetic = true;
   int syn_classes = 1000;
+  // current training iteration (set by Solver::Step) — synthetic data and
+  // dropout masks key off this so snapshot/restore resumes the exact
+  // stream (the reference's LMDB cursor position analog)
+  uint64_t data_iter = 0;
 
   bool gpu_inited = false;
   hipStream_t stream = nullptr;       // compute stream

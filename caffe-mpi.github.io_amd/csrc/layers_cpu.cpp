@@ -163,7 +163,7 @@ void DataLayer::Forward_cpu(const std::vector<Blob*>&,
                             const std::vector<Blob*>& top) {
   Engine& E = Engine::get();
   const uint64_t key = h_splitmix64(E.seed ^ ((uint64_t)E.rank << 40) ^
-                                  (iter_ << 8));
+                                  (E.data_iter << 8));
   float* d = top[0]->mutable_cpu_data();
   const long n = top[0]->count();
   for (long i = 0; i < n; ++i)
@@ -800,7 +800,8 @@ void DropoutLayer::Forward_cpu(const std::vector<Blob*>& bottom,
   uint8_t* mask = (uint8_t*)mask_.mutable_cpu_data();
   Engine& E = Engine::get();
   const uint64_t key =
-      h_splitmix64(E.seed ^ 0xD0D0ull ^ ((uint64_t)E.rank << 40) ^ iter_);
+      h_splitmix64(E.seed ^ 0xD0D0ull ^ ((uint64_t)E.rank << 40) ^
+                   E.data_iter);
   for (long i = 0; i < n; ++i) {
     mask[i] = h_u01(h_splitmix64(key ^ (uint64_t)i)) >= ratio_;
     y[i] = x[i] * mask[i] * scale_;

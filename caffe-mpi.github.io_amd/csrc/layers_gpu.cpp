@@ -18,7 +18,7 @@ void DataLayer::Forward_gpu(const std::vector<Blob*>&,
                             const std::vector<Blob*>& top) {
   Engine& E = Engine::get();
   const uint64_t key =
-      h_splitmix64(E.seed ^ ((uint64_t)E.rank << 40) ^ (iter_ << 8));
+      h_splitmix64(E.seed ^ ((uint64_t)E.rank << 40) ^ (E.data_iter << 8));
   gpu::fill_uniform(E.stream, top[0]->count(), key, 0, -1.f, 1.f,
                     top[0]->mutable_gpu_data());
   gpu::fill_labels(E.stream, top[1]->count(), key, 0, E.syn_classes,
@@ -401,7 +401,8 @@ void DropoutLayer::Forward_gpu(const std::vector<Blob*>& bottom,
     return;
   }
   const uint64_t key =
-      h_splitmix64(E.seed ^ 0xD0D0ull ^ ((uint64_t)E.rank << 40) ^ iter_);
+      h_splitmix64(E.seed ^ 0xD0D0ull ^ ((uint64_t)E.rank << 40) ^
+                   E.data_iter);
   gpu::dropout_fwd(E.stream, bottom[0]->gpu_data(), n, key, 0, ratio_,
                    scale_, top[0]->mutable_gpu_data(),
                    (uint8_t*)mask_.mutable_gpu_data());

@@ -1,5 +1,6 @@
 #include "net.hpp"
 
+#include <chrono>
 #include <set>
 
 namespace camd {
@@ -286,6 +287,18 @@ float Net::loss() {
   return total;
 }
 
+std::vector<std::pair<std::string, float>> Net::scores() {
+  Engine::get().sync();
+  std::vector<std::pair<std::string, float>> out;
+  for (size_t i = 0; i < layers_.size(); ++i)
+    for (size_t t = 0; t < tops_[i].size(); ++t) {
+      const bool acc = layers_[i]->type() == "Accuracy";
+      if (layers_[i]->loss((int)t) != 0.f || acc)
+        out.emplace_back(layers_[i]->name(), tops_[i][t]->cpu_data()[0]);
+    }
+  return out;
+}
+
 std::vector<std::string> Net::blob_names() const {
   std::vector<std::string> out;
   for (auto& kv : blob_map_) out.push_back(kv.first);
@@ -303,6 +316,117 @@ void Net::ShareTrainedLayersWith(Net& other) {
     for (size_t i = 0; i < dst.size() && i < src.size(); ++i)
       dst[i]->ShareData(*src[i]);
   }
+}
+
+}  // namespace camd
+
+// ------------------------------------------------- snapshot interop
+// .caffemodel = binary NetParameter carrying each parametered layer's blobs
+// (reference Net::ToProto / CopyTrainedLayersFrom, net.cpp:1055-1248).
+#include <fstream>
+
+#include "proto_wire.hpp"
+
+namespace camd {
+
+void Net::SaveWeights(const std::string& path) {
+  Engine::get().sync();
+  wire::Writer net_w;
+  net_w.str(1, name_);
+  for (size_t i = 0; i < layers_.size(); ++i) {
+    auto& lb = layers_[i]->blobs();
+    if (lb.empty()) continue;
+    wire::Writer lw;
+    lw.str(1, layers_[i]->name());
+    lw.str(2, layers_[i]->type());
+    for (auto& b : lb) {
+      wire::Writer bw;
+      bw.packed_floats(5, b->cpu_data(), b->count());
+      wire::Writer sw;
+      std::vector<int64_t> dims(b->shape().begin(), b->shape().end());
+      sw.packed_i64(1, dims);
+      bw.submsg(7, sw.out);
+      lw.submsg(7, bw.out);
+    }
+    net_w.submsg(100, lw.out);
+  }
+  std::ofstream f(path, std::ios::binary);
+  CHECK_(f.good()) << "cannot write " << path;
+  f.write(net_w.out.data(), (long)net_w.out.size());
+}
+
+void Net::LoadWeights(const std::string& path) {
+  std::ifstream f(path, std::ios::binary);
+  CHECK_(f.good()) << "cannot read " << path;
+  std::string buf((std::istreambuf_iterator<char>(f)),
+                  std::istreambuf_iterator<char>());
+  std::map<std::string, Layer*> by_name;
+  for (auto& l : layers_) by_name[l->name()] = l.get();
+  wire::Reader r(buf.data(), buf.size());
+  wire::Field fld;
+  int loaded = 0;
+  while (r.next(&fld)) {
+    if (fld.num != 100 || fld.wt != 2) continue;  // LayerParameter
+    wire::Reader lr(fld.data, fld.len);
+    wire::Field lf;
+    std::string lname;
+    std::vector<wire::BlobData> blobs;
+    while (lr.next(&lf)) {
+      if (lf.num == 1 && lf.wt == 2)
+        lname.assign(lf.data, lf.len);
+      else if (lf.num == 7 && lf.wt == 2)
+        blobs.push_back(wire::parse_blob(lf.data, lf.len));
+    }
+    auto it = by_name.find(lname);
+    if (it == by_name.end()) continue;  // reference skips unknown layers
+    auto& lb = it->second->blobs();
+    for (size_t j = 0; j < blobs.size() && j < lb.size(); ++j) {
+      CHECK_EQ_((long)blobs[j].data.size(), lb[j]->count())
+          << "blob size mismatch in " << lname << " blob " << j;
+      memcpy(lb[j]->mutable_cpu_data(), blobs[j].data.data(),
+             blobs[j].data.size() * sizeof(float));
+      ++loaded;
+    }
+  }
+  CHECK_GT_(loaded, 0) << "no matching layer blobs in " << path;
+}
+
+void Net::time_layers(int iters) {
+  Engine& E = Engine::get();
+  const size_t L = layers_.size();
+  std::vector<double> fwd_ms(L, 0), bwd_ms(L, 0);
+  auto tick = [&]() {
+    E.sync();
+    return std::chrono::steady_clock::now();
+  };
+  Forward();  // warmup + shapes
+  Backward(nullptr);
+  for (int it = 0; it < iters; ++it) {
+    for (size_t i = 0; i < L; ++i) {
+      auto t0 = tick();
+      layers_[i]->Forward(bottoms_[i], tops_[i]);
+      auto t1 = tick();
+      fwd_ms[i] += std::chrono::duration<double, std::milli>(t1 - t0).count();
+    }
+    for (size_t i = L; i-- > 0;) {
+      if (!layer_need_bwd_[i]) continue;
+      auto t0 = tick();
+      layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+      auto t1 = tick();
+      bwd_ms[i] += std::chrono::duration<double, std::milli>(t1 - t0).count();
+    }
+  }
+  double ftot = 0, btot = 0;
+  fprintf(stderr, "%-28s %12s %12s\n", "layer", "forward(ms)",
+          "backward(ms)");
+  for (size_t i = 0; i < L; ++i) {
+    fprintf(stderr, "%-28s %12.3f %12.3f\n", layers_[i]->name().c_str(),
+            fwd_ms[i] / iters, bwd_ms[i] / iters);
+    ftot += fwd_ms[i] / iters;
+    btot += bwd_ms[i] / iters;
+  }
+  fprintf(stderr, "%-28s %12.3f %12.3f  (total %.3f ms/iter)\n", "TOTAL",
+          ftot, btot, ftot + btot);
 }
 
 }  // namespace camd

@@ -30,6 +30,8 @@ class Net {
   void Forward();
   void Backward(ReduceHook* hook = nullptr);
   float loss();  // syncs; Σ loss_weight · loss-top
+  // test-score outputs: every loss-weighted top + every Accuracy top
+  std::vector<std::pair<std::string, float>> scores();
 
   struct LParam {
     Blob* blob;
@@ -57,6 +59,14 @@ class Net {
 
   // copy weights from another net (test-net sharing / snapshot restore)
   void ShareTrainedLayersWith(Net& other);
+
+  // .caffemodel binaryproto interop (reference Net weight load/save,
+  // net.cpp:1055-1248; format: proto_wire.hpp)
+  void SaveWeights(const std::string& path);
+  void LoadWeights(const std::string& path);
+
+  // `caffe time`-style per-layer forward/backward timing report
+  void time_layers(int iters);
 
  private:
   void init(const PMsgPtr& msg, int batch_override);

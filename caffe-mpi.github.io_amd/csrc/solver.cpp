@@ -1,6 +1,9 @@
 #include "solver.hpp"
 
 #include <cmath>
+#include <fstream>
+
+#include "proto_wire.hpp"
 
 namespace camd {
 
@@ -13,6 +16,7 @@ Solver::Solver(const PMsgPtr& sp, int batch_override) : param_(sp) {
   } else {
     CAMD_FATAL << "solver has no net";
   }
+  net_msg_ = net_msg;
   // solver-pinned seed (models use random_seed; `caffe time` pins 1371)
   Engine& E = Engine::get();
   if (sp->has("random_seed")) {
@@ -175,7 +179,17 @@ void Reducer::iteration_end(hipEvent_t backward_done) {
 void Solver::Step(int iters) {
   Engine& E = Engine::get();
   const long display = param_->inum("display", 0);
+  const long test_interval = param_->inum("test_interval", 0);
+  const long test_iter = param_->inum("test_iter", 1);
+  const long snap_interval = param_->inum("snapshot", 0);
+  // this fork always runs a 1-iter test at iter 0 regardless of
+  // test_initialization (reference solver.cpp:243-248)
+  if (iter_ == 0 && test_interval > 0 && test_net()) TestAll(1);
   for (int i = 0; i < iters; ++i) {
+    if (test_interval > 0 && iter_ > 0 && iter_ % test_interval == 0 &&
+        test_net())
+      TestAll(test_iter);
+    E.data_iter = (uint64_t)iter_;
     cur_lr_ = GetLearningRate();
     cur_mom_ = GetMomentum();
     grad_scale_ = comm_ && comm_->world() > 1
@@ -185,6 +199,7 @@ void Solver::Step(int iters) {
     net_->Forward();
     net_->Backward(&reducer_);
     ++iter_;
+    if (snap_interval > 0 && iter_ % snap_interval == 0) Snapshot();
     if (display > 0 && iter_ % display == 0) {
       const float l = net_->loss();
       fprintf(stderr, "[caffe_amd] Iteration %ld, loss = %g, lr = %g\n",
@@ -197,6 +212,109 @@ void Solver::Step(int iters) {
 std::shared_ptr<Solver> create_solver_from_file(const std::string& path,
                                                 int batch_override) {
   return std::make_shared<Solver>(parse_prototxt_file(path), batch_override);
+}
+
+Net* Solver::test_net() {
+  if (test_net_) return test_net_.get();
+  // only build when the net defines TEST-phase layers
+  bool has_test = false;
+  for (auto& lm : net_msg_->subs("layer")) {
+    for (auto& inc : lm->subs("include"))
+      if (inc->str("phase") == "TEST") has_test = true;
+  }
+  if (!has_test) return nullptr;
+  test_net_.reset(new Net(net_msg_, Phase::TEST));
+  test_net_->ShareTrainedLayersWith(*net_);
+  return test_net_.get();
+}
+
+void Solver::TestAll(long iters) {
+  Net* tn = test_net();
+  if (!tn) return;
+  // average every loss-weighted / Accuracy top over the test iterations
+  // (reference Solver::Test, solver.cpp:439-540; cross-rank SharedScores
+  // aggregation is multi-node machinery — single-node here)
+  std::map<std::string, double> scores;
+  for (long it = 0; it < iters; ++it) {
+    tn->Forward();
+    for (auto& kv : tn->scores()) scores[kv.first] += kv.second;
+  }
+  for (auto& kv : scores)
+    fprintf(stderr, "[caffe_amd] Test net output: %s = %g\n",
+            kv.first.c_str(), kv.second / iters);
+}
+
+void Solver::Snapshot() {
+  const std::string prefix = param_->str("snapshot_prefix", "snapshot");
+  const std::string model =
+      prefix + "_iter_" + std::to_string(iter_) + ".caffemodel";
+  const std::string state =
+      prefix + "_iter_" + std::to_string(iter_) + ".solverstate";
+  net_->SaveWeights(model);
+  // SolverState { iter=1, learned_net=2, history=3 (BlobProto),
+  // current_step=4 } — caffe.proto:303-308
+  Engine& E = Engine::get();
+  wire::Writer sw;
+  sw.vint(1, iter_);
+  sw.str(2, model);
+  std::vector<float> host;
+  for (auto& p : net_->learnable_params()) {
+    wire::Writer bw;
+    if (E.mode == Mode::GPU) {
+      host.resize(p.count);
+      HIP_CHECK(hipMemcpy(host.data(), history_ + p.offset,
+                          sizeof(float) * p.count, hipMemcpyDeviceToHost));
+      bw.packed_floats(5, host.data(), p.count);
+    } else {
+      bw.packed_floats(5, host_history_.data() + p.offset, p.count);
+    }
+    wire::Writer shw;
+    std::vector<int64_t> dims(p.blob->shape().begin(),
+                              p.blob->shape().end());
+    shw.packed_i64(1, dims);
+    bw.submsg(7, shw.out);
+    sw.submsg(3, bw.out);
+  }
+  sw.vint(4, current_step_);
+  std::ofstream f(state, std::ios::binary);
+  CHECK_(f.good()) << "cannot write " << state;
+  f.write(sw.out.data(), (long)sw.out.size());
+  fprintf(stderr, "[caffe_amd] Snapshotting to %s\n", model.c_str());
+}
+
+void Solver::Restore(const std::string& path) {
+  std::ifstream f(path, std::ios::binary);
+  CHECK_(f.good()) << "cannot read " << path;
+  std::string buf((std::istreambuf_iterator<char>(f)),
+                  std::istreambuf_iterator<char>());
+  Engine& E = Engine::get();
+  wire::Reader r(buf.data(), buf.size());
+  wire::Field fld;
+  std::string learned;
+  size_t hidx = 0;
+  auto& params = net_->learnable_params();
+  while (r.next(&fld)) {
+    if (fld.num == 1 && fld.wt == 0) iter_ = (long)fld.vint;
+    else if (fld.num == 2 && fld.wt == 2) learned.assign(fld.data, fld.len);
+    else if (fld.num == 4 && fld.wt == 0) current_step_ = (int)fld.vint;
+    else if (fld.num == 3 && fld.wt == 2) {
+      auto b = wire::parse_blob(fld.data, fld.len);
+      CHECK_LT_(hidx, params.size());
+      CHECK_EQ_((long)b.data.size(), params[hidx].count);
+      if (E.mode == Mode::GPU) {
+        HIP_CHECK(hipMemcpy(history_ + params[hidx].offset, b.data.data(),
+                            sizeof(float) * b.data.size(),
+                            hipMemcpyHostToDevice));
+      } else {
+        memcpy(host_history_.data() + params[hidx].offset, b.data.data(),
+               sizeof(float) * b.data.size());
+      }
+      ++hidx;
+    }
+  }
+  if (!learned.empty()) net_->LoadWeights(learned);
+  fprintf(stderr, "[caffe_amd] Restored iter %ld from %s\n", iter_,
+          path.c_str());
 }
 
 }  // namespace camd

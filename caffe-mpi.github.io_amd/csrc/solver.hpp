@@ -59,6 +59,13 @@ class Solver {
   float GetMomentum() const { return (float)param_->num("momentum", 0.0); }
 
   Net& net() { return *net_; }
+  Net* test_net();  // lazily built TEST-phase net sharing train weights
+  void TestAll(long iters);  // solver.cpp:439-540 semantics
+  void Snapshot();           // .caffemodel + .solverstate (solver.cpp:542)
+  void Restore(const std::string& state_path);
+  void LoadWeights(const std::string& model_path) {
+    net_->LoadWeights(model_path);
+  }
   long iter() const { return iter_; }
   float last_loss() { return net_->loss(); }
   const PMsgPtr& param() const { return param_; }
@@ -76,7 +83,9 @@ class Solver {
 
  private:
   PMsgPtr param_;
+  PMsgPtr net_msg_;
   std::unique_ptr<Net> net_;
+  std::unique_ptr<Net> test_net_;
   std::unique_ptr<Comm> comm_;
   Reducer reducer_{this};
   long iter_ = 0;

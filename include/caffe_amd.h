@@ -53,6 +53,14 @@ long caffe_solver_iter(caffe_solver_t s);
 float caffe_solver_loss(caffe_solver_t s);
 caffe_net_t caffe_solver_net(caffe_solver_t s);
 
+/* Solver::Snapshot / Restore (solver.cpp:542-604; .caffemodel binaryproto
+ * weights + .solverstate with SGD history, sgd_solver.cpp:262-353) */
+int caffe_solver_snapshot(caffe_solver_t s);
+int caffe_solver_restore(caffe_solver_t s, const char* state_path);
+/* Net weight interop (net.cpp:1055-1248) */
+int caffe_net_save_weights(caffe_net_t n, const char* path);
+int caffe_net_load_weights(caffe_net_t n, const char* path);
+
 /* one-process-per-GPU collective bootstrap (replaces Clusters::Init +
  * MPI_Bcast of ncclUniqueId, clusters.cpp:8 / parallel.cpp:42-45):
  * rank 0 generates the 128-byte id, the launcher distributes it, every
