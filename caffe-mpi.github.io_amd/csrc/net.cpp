@@ -96,7 +96,12 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
   name_ = msg->str("name");
   std::vector<PMsgPtr> lmsgs;
   for (auto& lm : msg->subs("layer"))
-    if (layer_in_phase(lm, phase_)) lmsgs.push_back(lm);
+    if (layer_in_phase(lm, phase_)) {
+      // shallow-clone: insert_splits rewrites bottom names in place, and
+      // the source tree is shared with other nets (train/test pair)
+      auto copy = std::make_shared<PMsg>(*lm);
+      lmsgs.push_back(copy);
+    }
   CHECK_(!lmsgs.empty()) << "net has no layers for this phase";
   insert_splits(lmsgs);
 
