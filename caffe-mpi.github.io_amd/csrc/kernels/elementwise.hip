@@ -386,21 +386,40 @@ void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
                      inv_std);
 }
 
-__global__ void k_bn_fwd_norm(const float* __restrict__ x,
+// flat float4 walk; per 4-pack ONE division recovers the channel (packs
+// never straddle a channel boundary check: split handled elementwise)
+__global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                               const float* __restrict__ mean,
                               const float* __restrict__ inv_std,
                               const float* __restrict__ scale,
                               const float* __restrict__ bias, int sb, int C,
-                              int S, int frelu, float* __restrict__ y) {
-  const int c = blockIdx.x % C;
-  const long base = (long)blockIdx.x * S;
-  const float mu = mean[c], inv = inv_std[c];
-  const float sc = sb ? scale[c] : 1.f, bi = sb ? bias[c] : 0.f;
-  for (int sp = blockIdx.y * blockDim.x + threadIdx.x; sp < S;
-       sp += gridDim.y * blockDim.x) {
-    float v = (x[base + sp] - mu) * inv * sc + bi;
-    if (frelu) v = fmaxf(v, 0.f);
-    y[base + sp] = v;
+                              int S, int frelu, long n4,
+                              f4* __restrict__ y) {
+  VEC_GRID(i, n4) {
+    const long e0 = i * 4;
+    const int row = (int)(e0 / S);  // n*C + c
+    const int c0 = row % C;
+    const int rem = (int)(e0 - (long)row * S);
+    f4 v = x[i];
+    if (rem + 4 <= S) {
+      const float mu = mean[c0], inv = inv_std[c0];
+      const float sc = sb ? scale[c0] : 1.f, bi = sb ? bias[c0] : 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float t = (v[j] - mu) * inv * sc + bi;
+        v[j] = frelu ? fmaxf(t, 0.f) : t;
+      }
+    } else {  // pack crosses a channel boundary
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long e = e0 + j;
+        const int cc = (int)((e / S) % C);
+        float t = (v[j] - mean[cc]) * inv_std[cc] * (sb ? scale[cc] : 1.f) +
+                  (sb ? bias[cc] : 0.f);
+        v[j] = frelu ? fmaxf(t, 0.f) : t;
+      }
+    }
+    y[i] = v;
   }
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
@@ -408,11 +427,12 @@ void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  int sb, int N, int C, long S, float* y, int fuse_relu) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 8.0 * total);
-  const int bsz = S < 128 ? 64 : TPB;
-  const int by = (int)std::min<long>((S + bsz - 1) / bsz, 16);
-  hipLaunchKernelGGL(k_bn_fwd_norm, dim3(N * C, by), dim3(bsz), 0, s, x,
-                     mean, inv_std, scale, bias, sb, C, (int)S, fuse_relu,
-                     y);
+  // blobs are 16-float padded: the final partial pack reads/writes pad
+  // space with wrapped (but in-bounds) channel indices — harmless
+  const long n4 = (total + 3) / 4;
+  hipLaunchKernelGGL(k_bn_fwd_norm, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
+                     (const f4*)x, mean, inv_std, scale, bias, sb, C,
+                     (int)S, fuse_relu, n4, (f4*)y);
 }
 
 __global__ void k_bn_moving_avg(const float* __restrict__ mean,
@@ -532,23 +552,41 @@ void bn_bwd_finalize(hipStream_t s, const void* partials, int nb, int C,
                      dbias, m_dy, m_dyxn);
 }
 
-__global__ void k_bn_bwd_apply(const float* __restrict__ x,
-                               const float* __restrict__ dy,
+__global__ void k_bn_bwd_apply(const f4* __restrict__ x,
+                               const f4* __restrict__ dy,
                                const float* __restrict__ mean,
                                const float* __restrict__ inv_std,
                                const float* __restrict__ scale, int sb,
                                const float* __restrict__ m_dy,
                                const float* __restrict__ m_dyxn, int C,
-                               int S, float* __restrict__ dx) {
-  const int c = blockIdx.x % C;
-  const long base = (long)blockIdx.x * S;
-  const float mu = mean[c], inv = inv_std[c];
-  const float sc = sb ? scale[c] : 1.f;
-  const float mdy = m_dy[c], mdyxn = m_dyxn[c];
-  for (int sp = blockIdx.y * blockDim.x + threadIdx.x; sp < S;
-       sp += gridDim.y * blockDim.x) {
-    const float xn = (x[base + sp] - mu) * inv;
-    dx[base + sp] = (dy[base + sp] * sc - mdy - mdyxn * xn) * inv;
+                               int S, long n4, f4* __restrict__ dx) {
+  VEC_GRID(i, n4) {
+    const long e0 = i * 4;
+    const int row = (int)(e0 / S);
+    const int rem = (int)(e0 - (long)row * S);
+    const f4 xv = x[i];
+    f4 d = dy[i];
+    if (rem + 4 <= S) {
+      const int c = row % C;
+      const float mu = mean[c], inv = inv_std[c];
+      const float sc = sb ? scale[c] : 1.f;
+      const float mdy = m_dy[c], mdyxn = m_dyxn[c];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float xn = (xv[j] - mu) * inv;
+        d[j] = (d[j] * sc - mdy - mdyxn * xn) * inv;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int cc = (int)(((e0 + j) / S) % C);
+        const float xn = (xv[j] - mean[cc]) * inv_std[cc];
+        d[j] = (d[j] * (sb ? scale[cc] : 1.f) - m_dy[cc] -
+                m_dyxn[cc] * xn) *
+               inv_std[cc];
+      }
+    }
+    dx[i] = d;
   }
 }
 void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
@@ -557,11 +595,10 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* m_dyxn, int N, int C, long S, float* dx) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 12.0 * total);
-  const int bsz = S < 128 ? 64 : TPB;
-  const int by = (int)std::min<long>((S + bsz - 1) / bsz, 16);
-  hipLaunchKernelGGL(k_bn_bwd_apply, dim3(N * C, by), dim3(bsz), 0, s, x,
-                     dy, mean, inv_std, scale, sb, m_dy, m_dyxn, C, (int)S,
-                     dx);
+  const long n4 = (total + 3) / 4;
+  hipLaunchKernelGGL(k_bn_bwd_apply, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
+                     (const f4*)x, (const f4*)dy, mean, inv_std, scale, sb,
+                     m_dy, m_dyxn, C, (int)S, n4, (f4*)dx);
 }
 
 // ------------------------------------------------------------ LRN
