@@ -208,8 +208,13 @@ __device__ __forceinline__ void stage_b_write(float* Bs,
 
 template <bool TA, bool TB, bool SPLITK>
 __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
-  __shared__ float As[2][BK * LDS_S];
-  __shared__ float Bs[2][BK * LDS_S];
+  // one LDS arena carved into As[2] | Bs[2]; the tail-wave reduction below
+  // reuses it after the K-loop
+  __shared__ float smem[4 * BK * LDS_S];
+  auto As = [&](int buf) -> float* { return smem + buf * (BK * LDS_S); };
+  auto Bs = [&](int buf) -> float* {
+    return smem + (2 + buf) * (BK * LDS_S);
+  };
 
   // bijective XCD-aware swizzle (cdna_hip_programming.md T1): contiguous
   // tile chunks per XCD so neighbouring tiles share L2-resident panels
@@ -225,7 +230,17 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   const long m0 = tile_m * BM, n0 = tile_n * BN;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wr = wave >> 1, wc = wave & 1;  // 2x2 wave grid
+  // wave decomposition: 2x2 over the 128x128 tile normally; when M or N
+  // fits in 64 the spare waves take K-ranges instead (intra-block split-K,
+  // combined deterministically through LDS) so small-Cout layers keep all
+  // 4 waves busy
+  const int mwaves = g.M > 64 ? 2 : 1;
+  const int nwaves = g.N > 64 ? 2 : 1;
+  const int mn = mwaves * nwaves;
+  const int kwaves = 4 / mn;
+  const int wr = wave % mwaves;
+  const int wc = (wave / mwaves) % nwaves;
+  const int wk = wave / mn;
   const int row_in = lane & 31;             // MFMA row/col index
   const int ksel = lane >> 5;               // which of the 2 K elems
 
@@ -244,39 +259,74 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
 
   stage_a_load<TA>(g, m0, k_lo, ra);
   stage_b_load<TB>(g, n0, k_lo, rb);
-  stage_a_write<TA>(As[0], ra);
-  stage_b_write<TB>(Bs[0], rb);
+  stage_a_write<TA>(As(0), ra);
+  stage_b_write<TB>(Bs(0), rb);
   __syncthreads();
 
+  // this wave's K-tile range (whole range when kwaves == 1)
+  const long tk0 = ntiles * wk / kwaves;
+  const long tk1 = ntiles * (wk + 1) / kwaves;
   int cur = 0;
   for (long t = 0; t < ntiles; ++t) {
     if (t + 1 < ntiles) {  // issue next tile's global loads early (T14)
       stage_a_load<TA>(g, m0, k_lo + (t + 1) * BK, ra);
       stage_b_load<TB>(g, n0, k_lo + (t + 1) * BK, rb);
     }
-    const float* Ab = As[cur];
-    const float* Bb = Bs[cur];
+    if (t >= tk0 && t < tk1) {
+      const float* Ab = As(cur);
+      const float* Bb = Bs(cur);
 #pragma unroll
-    for (int kk = 0; kk < BK; kk += 2) {
-      const int krow = kk + ksel;
-      const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
-      const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
-      const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
-      const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
-      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-      acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+      for (int kk = 0; kk < BK; kk += 2) {
+        const int krow = kk + ksel;
+        const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+        const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+        const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+        const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
+        acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+        acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+        acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+        acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+      }
     }
     // single barrier per K-tile: writing buf[cur^1] here is safe — its
     // last readers finished before the previous iteration's barrier; the
     // barrier below publishes these writes for the next iteration's reads
     if (t + 1 < ntiles) {
-      stage_a_write<TA>(As[cur ^ 1], ra);
-      stage_b_write<TB>(Bs[cur ^ 1], rb);
+      stage_a_write<TA>(As(cur ^ 1), ra);
+      stage_b_write<TB>(Bs(cur ^ 1), rb);
     }
     __syncthreads();
     cur ^= 1;
+  }
+
+  if (kwaves > 1) {
+    // combine the K-range partials: spare waves park their accumulators in
+    // the (now idle) staging LDS; the wk==0 wave adds them in ascending wk
+    // order — deterministic
+    __syncthreads();  // all reads of the staging buffers are done
+    if (wk > 0) {
+      float* slot = smem + (long)(wave - mn) * 4096 + lane * 64;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        slot[r] = acc00[r];
+        slot[16 + r] = acc01[r];
+        slot[32 + r] = acc10[r];
+        slot[48 + r] = acc11[r];
+      }
+    }
+    __syncthreads();
+    if (wk > 0) return;  // primaries finish the epilogue
+    for (int k2 = 1; k2 < kwaves; ++k2) {
+      const int src_wave = k2 * mn + (wc * mwaves + wr);
+      const float* slot = smem + (long)(src_wave - mn) * 4096 + lane * 64;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        acc00[r] += slot[r];
+        acc01[r] += slot[16 + r];
+        acc10[r] += slot[32 + r];
+        acc11[r] += slot[48 + r];
+      }
+    }
   }
 
   // ---- epilogue: acc reg r -> (row, col)
