@@ -226,9 +226,15 @@ def main():
     gen = os.path.join(REPO, "models", "generated",
                        f"{args.model}_solver.prototxt")
     if not os.path.exists(gen):
-        subprocess.check_call([sys.executable,
-                               os.path.join(REPO, "models",
-                                            "gen_models.py")])
+        if rank == 0:
+            subprocess.check_call([sys.executable,
+                                   os.path.join(REPO, "models",
+                                                "gen_models.py")])
+        else:  # avoid the generation race: wait for rank 0
+            for _ in range(120):
+                if os.path.exists(gen):
+                    break
+                time.sleep(0.5)
     solver = ca.Solver(path=gen, batch_override=args.batch)
 
     if world > 1:

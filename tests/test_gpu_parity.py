@@ -24,3 +24,40 @@ def test_gemm_shapes_gpu(shape):
     from layer_checks import check_ip
     M, N, K = shape
     check_ip("gpu", M=M, K=K, Nout=N)
+
+
+def test_resnet50_fullnet_gpu_vs_cpu():
+    """Whole-graph integration parity: one training step of ResNet-50 at
+    batch 2 in GPU mode vs CPU mode (identical seeds => identical synthetic
+    data, fillers and update), compared at the fp32 tolerance on the loss,
+    a mid-net activation and every param after the update."""
+    import os
+    import subprocess
+    import sys
+    import caffe_amd as ca
+    from engine_util import REPO, relerr
+
+    gen = os.path.join(REPO, "models", "generated",
+                       "resnet50_solver.prototxt")
+    if not os.path.exists(gen):
+        subprocess.check_call([sys.executable,
+                               os.path.join(REPO, "models",
+                                            "gen_models.py")])
+
+    results = {}
+    for mode in ("cpu", "gpu"):
+        ca.set_mode(mode)
+        ca.set_synthetic_shape(3, 224, 224, 1000)
+        ca.set_random_seed(99)
+        s = ca.Solver(path=gen, batch_override=2)
+        s.step(1)
+        blob = s.net.blob("pool1")
+        loss = s.loss()
+        params = [s.net.param(i) for i in range(0, s.net.num_params(), 16)]
+        results[mode] = (loss, blob, params)
+    lc, bc, pc = results["cpu"]
+    lg, bg, pg = results["gpu"]
+    assert abs(lc - lg) < 1e-3 * max(1.0, abs(lc)), (lc, lg)
+    assert relerr(bg, bc) < 5e-4, relerr(bg, bc)
+    for a, b in zip(pc, pg):
+        assert relerr(b, a) < 1e-3
