@@ -485,6 +485,7 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
                                const float* __restrict__ mean,
                                const float* __restrict__ inv_std, int N,
                                int C, long S, int nb,
+                               const float* __restrict__ relu_y,
                                double2* __restrict__ out) {
   const int c = blockIdx.x % C;
   const int slice = blockIdx.x / C;
@@ -497,7 +498,8 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
   for (int i = threadIdx.x; i < span; i += blockDim.x) {
     const int n = n0 + i / Si;
     const long off = ((long)n * C + c) * S + (i - (n - n0) * Si);
-    const double d = dy[off];
+    double d = dy[off];
+    if (relu_y && relu_y[off] <= 0.f) d = 0.0;  // fused ReLU backward
     s_dy += d;
     s_dyxn += d * (double)((x[off] - m) * inv);
   }
@@ -516,10 +518,11 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
 }
 void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std, int N, int C,
-                  long S, int nb, void* partials) {
+                  long S, int nb, const float* relu_y, void* partials) {
   PerfScope perf("bn", s, 0, 8.0 * N * C * S);
   hipLaunchKernelGGL(k_bn_bwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, dy,
-                     mean, inv_std, N, C, S, nb, (double2*)partials);
+                     mean, inv_std, N, C, S, nb, relu_y,
+                     (double2*)partials);
 }
 
 __global__ void k_bn_bwd_finalize(const double2* __restrict__ partials,
@@ -559,13 +562,20 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                                const float* __restrict__ scale, int sb,
                                const float* __restrict__ m_dy,
                                const float* __restrict__ m_dyxn, int C,
-                               int S, long n4, f4* __restrict__ dx) {
+                               int S, long n4, const f4* __restrict__ relu_y,
+                               f4* __restrict__ dx) {
   VEC_GRID(i, n4) {
     const long e0 = i * 4;
     const int row = (int)(e0 / S);
     const int rem = (int)(e0 - (long)row * S);
     const f4 xv = x[i];
     f4 d = dy[i];
+    if (relu_y) {  // fused ReLU backward: mask by the post-activation sign
+      const f4 yv = relu_y[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (yv[j] <= 0.f) d[j] = 0.f;
+    }
     if (rem + 4 <= S) {
       const int c = row % C;
       const float mu = mean[c], inv = inv_std[c];
@@ -592,13 +602,15 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
 void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std,
                   const float* scale, int sb, const float* m_dy,
-                  const float* m_dyxn, int N, int C, long S, float* dx) {
+                  const float* m_dyxn, int N, int C, long S,
+                  const float* relu_y, float* dx) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 12.0 * total);
   const long n4 = (total + 3) / 4;
   hipLaunchKernelGGL(k_bn_bwd_apply, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, (const f4*)dy, mean, inv_std, scale, sb,
-                     m_dy, m_dyxn, C, (int)S, n4, (f4*)dx);
+                     m_dy, m_dyxn, C, (int)S, n4, (const f4*)relu_y,
+                     (f4*)dx);
 }
 
 // ------------------------------------------------------------ LRN

@@ -236,6 +236,7 @@ class ReLULayer : public Layer {
  public:
   using Layer::Layer;
   bool fused_away_ = false;  // producer applies the ReLU in its epilogue
+  bool bwd_fused_ = false;   // producer also absorbs the backward mask
   void Reshape(const std::vector<Blob*>& b,
                const std::vector<Blob*>& t) override {
     if (b[0] != t[0]) t[0]->ReshapeLike(*b[0]);

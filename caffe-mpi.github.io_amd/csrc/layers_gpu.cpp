@@ -294,8 +294,11 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
   const int nb = gpu::bn_blocks_per_channel(N, C_);
   partials_.Reshape({(int)(C_ * nb * 4)});
   void* parts = partials_.mutable_gpu_data();
+  // fused in-place ReLU backward: the following ReLU wrote y in place
+  // (top data == post-activation); mask dy by y>0 inside the BN kernels
+  const float* relu_y = fuse_relu_ ? top[0]->gpu_data() : nullptr;
   gpu::bn_bwd_stats(E.stream, x, dy, mean_.gpu_data(), inv_std_.gpu_data(),
-                    N, C_, S, nb, parts);
+                    N, C_, S, nb, relu_y, parts);
   gpu::bn_bwd_finalize(
       E.stream, parts, nb, C_, (long)N * S,
       scale_bias_ ? blobs_[3]->gpu_data() : nullptr, scale_bias_,
@@ -307,7 +310,7 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
                       inv_std_.gpu_data(),
                       scale_bias_ ? blobs_[3]->gpu_data() : nullptr,
                       scale_bias_, m_dy_.gpu_data(), m_dyxn_.gpu_data(), N,
-                      C_, S, bottom[0]->mutable_gpu_diff());
+                      C_, S, relu_y, bottom[0]->mutable_gpu_diff());
 }
 
 // ------------------------------------------------------------------ ReLU
@@ -324,6 +327,7 @@ void ReLULayer::Backward_gpu(const std::vector<Blob*>& top,
                              const std::vector<bool>& prop_down,
                              const std::vector<Blob*>& bottom) {
   if (!prop_down[0]) return;
+  if (fused_away_ && bwd_fused_) return;  // absorbed into BN backward
   auto rp = param_->sub("relu_param");
   const float slope = rp ? (float)rp->num("negative_slope", 0) : 0.f;
   gpu::relu_bwd(Engine::get().stream, bottom[0]->gpu_data(),

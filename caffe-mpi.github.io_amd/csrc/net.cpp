@@ -212,6 +212,7 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
       if (auto* bn = dynamic_cast<BatchNormLayer*>(layers_[i - 1].get())) {
         bn->fuse_relu_ = true;
         relu->fused_away_ = true;
+        relu->bwd_fused_ = true;  // BN backward masks dy by the activation
       } else if (auto* el =
                      dynamic_cast<EltwiseLayer*>(layers_[i - 1].get())) {
         bool ones = el->op_ == "SUM";
