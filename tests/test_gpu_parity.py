@@ -69,4 +69,4 @@ def test_resnet50_fullnet_gpu_vs_cpu():
         # check: any graph/fusion bug shows as an O(1) relative error.
         na = float(np.linalg.norm(a))
         nd = float(np.linalg.norm(b - a))
-        assert nd < 0.1 * na + 1e-4, (nd, na)
+        assert nd < 0.1 * na + 1e-3, (nd, na)  # atol: zero-init bias params are pure update noise
