@@ -485,7 +485,8 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
                                const float* __restrict__ mean,
                                const float* __restrict__ inv_std, int N,
                                int C, long S, int nb,
-                               const float* __restrict__ relu_y,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ bias, int frelu,
                                double2* __restrict__ out) {
   const int c = blockIdx.x % C;
   const int slice = blockIdx.x / C;
@@ -494,14 +495,19 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
   const int Si = (int)S;
   const int span = (n1 - n0) * Si;
   const float m = mean[c], inv = inv_std[c];
+  // fused ReLU backward: recompute the forward activation's sign exactly
+  // (same fp32 op sequence as k_bn_fwd_norm) — no extra memory stream
+  const float sc = scale ? scale[c] : 1.f;
+  const float bi = bias ? bias[c] : 0.f;
   double s_dy = 0, s_dyxn = 0;
   for (int i = threadIdx.x; i < span; i += blockDim.x) {
     const int n = n0 + i / Si;
     const long off = ((long)n * C + c) * S + (i - (n - n0) * Si);
+    const float xn = (x[off] - m) * inv;
     double d = dy[off];
-    if (relu_y && relu_y[off] <= 0.f) d = 0.0;  // fused ReLU backward
+    if (frelu && xn * sc + bi <= 0.f) d = 0.0;
     s_dy += d;
-    s_dyxn += d * (double)((x[off] - m) * inv);
+    s_dyxn += d * (double)xn;
   }
   __shared__ double sh1[TPB], sh2[TPB];
   sh1[threadIdx.x] = s_dy;
@@ -518,10 +524,11 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
 }
 void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std, int N, int C,
-                  long S, int nb, const float* relu_y, void* partials) {
+                  long S, int nb, const float* scale, const float* bias,
+                  int frelu, void* partials) {
   PerfScope perf("bn", s, 0, 8.0 * N * C * S);
   hipLaunchKernelGGL(k_bn_bwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, dy,
-                     mean, inv_std, N, C, S, nb, relu_y,
+                     mean, inv_std, N, C, S, nb, scale, bias, frelu,
                      (double2*)partials);
 }
 
@@ -561,39 +568,38 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                                const float* __restrict__ inv_std,
                                const float* __restrict__ scale, int sb,
                                const float* __restrict__ m_dy,
-                               const float* __restrict__ m_dyxn, int C,
-                               int S, long n4, const f4* __restrict__ relu_y,
-                               f4* __restrict__ dx) {
+                               const float* __restrict__ m_dyxn,
+                               const float* __restrict__ bias, int frelu,
+                               int C, int S, long n4, f4* __restrict__ dx) {
   VEC_GRID(i, n4) {
     const long e0 = i * 4;
     const int row = (int)(e0 / S);
     const int rem = (int)(e0 - (long)row * S);
     const f4 xv = x[i];
     f4 d = dy[i];
-    if (relu_y) {  // fused ReLU backward: mask by the post-activation sign
-      const f4 yv = relu_y[i];
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        if (yv[j] <= 0.f) d[j] = 0.f;
-    }
     if (rem + 4 <= S) {
       const int c = row % C;
       const float mu = mean[c], inv = inv_std[c];
       const float sc = sb ? scale[c] : 1.f;
+      const float bi = (frelu && bias) ? bias[c] : 0.f;
       const float mdy = m_dy[c], mdyxn = m_dyxn[c];
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const float xn = (xv[j] - mu) * inv;
-        d[j] = (d[j] * sc - mdy - mdyxn * xn) * inv;
+        float dj = d[j];
+        if (frelu && xn * sc + bi <= 0.f) dj = 0.f;
+        d[j] = (dj * sc - mdy - mdyxn * xn) * inv;
       }
     } else {
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int cc = (int)(((e0 + j) / S) % C);
         const float xn = (xv[j] - mean[cc]) * inv_std[cc];
-        d[j] = (d[j] * (sb ? scale[cc] : 1.f) - m_dy[cc] -
-                m_dyxn[cc] * xn) *
-               inv_std[cc];
+        const float scc = sb ? scale[cc] : 1.f;
+        float dj = d[j];
+        if (frelu && xn * scc + ((frelu && bias) ? bias[cc] : 0.f) <= 0.f)
+          dj = 0.f;
+        d[j] = (dj * scc - m_dy[cc] - m_dyxn[cc] * xn) * inv_std[cc];
       }
     }
     dx[i] = d;
@@ -603,14 +609,13 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std,
                   const float* scale, int sb, const float* m_dy,
                   const float* m_dyxn, int N, int C, long S,
-                  const float* relu_y, float* dx) {
+                  const float* bias, int frelu, float* dx) {
   const long total = (long)N * C * S;
   PerfScope perf("bn", s, 0, 12.0 * total);
   const long n4 = (total + 3) / 4;
   hipLaunchKernelGGL(k_bn_bwd_apply, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, (const f4*)dy, mean, inv_std, scale, sb,
-                     m_dy, m_dyxn, C, (int)S, n4, (const f4*)relu_y,
-                     (f4*)dx);
+                     m_dy, m_dyxn, bias, frelu, C, (int)S, n4, (f4*)dx);
 }
 
 // ------------------------------------------------------------ LRN

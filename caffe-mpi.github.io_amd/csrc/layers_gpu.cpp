@@ -294,11 +294,13 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
   const int nb = gpu::bn_blocks_per_channel(N, C_);
   partials_.Reshape({(int)(C_ * nb * 4)});
   void* parts = partials_.mutable_gpu_data();
-  // fused in-place ReLU backward: the following ReLU wrote y in place
-  // (top data == post-activation); mask dy by y>0 inside the BN kernels
-  const float* relu_y = fuse_relu_ ? top[0]->gpu_data() : nullptr;
+  // fused in-place ReLU backward: the activation's sign is recomputed
+  // from xn*scale+bias inside the BN kernels (bit-identical to the
+  // forward's op sequence) — no extra memory stream
+  const float* sc_p = scale_bias_ ? blobs_[3]->gpu_data() : nullptr;
+  const float* bi_p = scale_bias_ ? blobs_[4]->gpu_data() : nullptr;
   gpu::bn_bwd_stats(E.stream, x, dy, mean_.gpu_data(), inv_std_.gpu_data(),
-                    N, C_, S, nb, relu_y, parts);
+                    N, C_, S, nb, sc_p, bi_p, fuse_relu_ ? 1 : 0, parts);
   gpu::bn_bwd_finalize(
       E.stream, parts, nb, C_, (long)N * S,
       scale_bias_ ? blobs_[3]->gpu_data() : nullptr, scale_bias_,
@@ -310,7 +312,8 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
                       inv_std_.gpu_data(),
                       scale_bias_ ? blobs_[3]->gpu_data() : nullptr,
                       scale_bias_, m_dy_.gpu_data(), m_dyxn_.gpu_data(), N,
-                      C_, S, relu_y, bottom[0]->mutable_gpu_diff());
+                      C_, S, bi_p, fuse_relu_ ? 1 : 0,
+                      bottom[0]->mutable_gpu_diff());
 }
 
 // ------------------------------------------------------------------ ReLU

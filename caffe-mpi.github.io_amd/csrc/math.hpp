@@ -133,7 +133,8 @@ void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,
                  int scale_bias, int N, int C, long S, float eps, float* y);
 void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std, int N, int C,
-                  long S, int nb, const float* relu_y, void* partials);
+                  long S, int nb, const float* scale, const float* bias,
+                  int fuse_relu, void* partials);
 // writes dscale/dbias (if scale_bias) and the per-channel m_dy/m_dyxn terms
 void bn_bwd_finalize(hipStream_t s, const void* partials, int nb, int C,
                      long NS, const float* scale, int scale_bias,
@@ -143,7 +144,7 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std,
                   const float* scale, int scale_bias, const float* m_dy,
                   const float* m_dyxn, int N, int C, long S,
-                  const float* relu_y, float* dx);
+                  const float* bias, int fuse_relu, float* dx);
 
 void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
              int size, float alpha, float beta, float k, float* scale,
