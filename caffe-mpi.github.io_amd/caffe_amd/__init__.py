@@ -122,7 +122,7 @@ def device_synchronize():
 
 
 def perf_snapshot():
-    cap, rows = 64, 64
+    cap, rows = 64, 256
     names = ctypes.create_string_buffer(cap * rows)
     launches = (ctypes.c_long * rows)()
     flops = (ctypes.c_double * rows)()

@@ -423,7 +423,12 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   }
   const char* pcls = !transA ? (!transB ? "gemm_nn" : "gemm_nt")
                             : (!transB ? "gemm_tn" : "gemm_tt");
-  PerfScope perf(pcls, s, 2.0 * M * N * K,
+  std::string pname = pcls;
+  static const bool by_shape = getenv("CAFFE_GEMM_BY_SHAPE") != nullptr;
+  if (by_shape)
+    pname += "_" + std::to_string(M) + "x" + std::to_string(N) + "x" +
+             std::to_string(K);
+  PerfScope perf(pname, s, 2.0 * M * N * K,
                  4.0 * (M * K + N * K + M * N));
   dim3 grid((unsigned)g.tiles, 1, (unsigned)SK);
   dim3 block(256);
