@@ -419,7 +419,7 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   int SK = 1;
   if (!epi && beta == 0.f && g.tiles < 1024 && K > 4 * BK) {
     SK = (int)std::min<long>(
-        {1024 / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 64});
+        {1024 / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 256});
   }
   const char* pcls = !transA ? (!transB ? "gemm_nn" : "gemm_nt")
                             : (!transB ? "gemm_tn" : "gemm_tt");
