@@ -263,40 +263,103 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   stage_b_write<TB>(Bs(0), rb);
   __syncthreads();
 
-  // this wave's K-tile range (whole range when kwaves == 1)
-  const long tk0 = ntiles * wk / kwaves;
-  const long tk1 = ntiles * (wk + 1) / kwaves;
   int cur = 0;
-  for (long t = 0; t < ntiles; ++t) {
-    if (t + 1 < ntiles) {  // issue next tile's global loads early (T14)
-      stage_a_load<TA>(g, m0, k_lo + (t + 1) * BK, ra);
-      stage_b_load<TB>(g, n0, k_lo + (t + 1) * BK, rb);
-    }
-    if (t >= tk0 && t < tk1) {
-      const float* Ab = As(cur);
-      const float* Bb = Bs(cur);
-#pragma unroll
-      for (int kk = 0; kk < BK; kk += 2) {
-        const int krow = kk + ksel;
-        const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
-        const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
-        const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
-        const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
-        acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
-        acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
-        acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
-        acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+  if (kwaves == 1) {
+    for (long t = 0; t < ntiles; ++t) {
+      if (t + 1 < ntiles) {  // issue next tile's global loads early (T14)
+        stage_a_load<TA>(g, m0, k_lo + (t + 1) * BK, ra);
+        stage_b_load<TB>(g, n0, k_lo + (t + 1) * BK, rb);
       }
+      {
+        const float* Ab = As(cur);
+        const float* Bb = Bs(cur);
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += 2) {
+          const int krow = kk + ksel;
+          const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+          const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+          const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+          const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
+          acc00 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+          acc01 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+          acc10 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+          acc11 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+      }
+      // single barrier per K-tile: writing buf[cur^1] here is safe — its
+      // last readers finished before the previous iteration's barrier; the
+      // barrier below publishes these writes for the next iteration's reads
+      if (t + 1 < ntiles) {
+        stage_a_write<TA>(As(cur ^ 1), ra);
+        stage_b_write<TB>(Bs(cur ^ 1), rb);
+      }
+      __syncthreads();
+      cur ^= 1;
     }
-    // single barrier per K-tile: writing buf[cur^1] here is safe — its
-    // last readers finished before the previous iteration's barrier; the
-    // barrier below publishes these writes for the next iteration's reads
-    if (t + 1 < ntiles) {
-      stage_a_write<TA>(As(cur ^ 1), ra);
-      stage_b_write<TB>(Bs(cur ^ 1), rb);
+  } else {
+    // kwaves path: BOTH LDS buffers are live per iteration — the wave group
+    // with (wk & 1) == p computes buffer p, so all 4 waves issue MFMAs
+    // concurrently even when the output tile only needs 1 or 2 of them.
+    // For kwaves == 4 the two waves sharing a buffer split the K-step in
+    // half; the partials are summed in the LDS reduction below.
+    const int grp = wk & 1;          // which buffer this wave computes
+    const int khalf = wk >> 1;       // kwaves==4: kk half within the step
+    const int kk0 = (kwaves == 4) ? khalf * (BK / 2) : 0;
+    const int kk1 = (kwaves == 4) ? kk0 + BK / 2 : BK;
+    // prologue already staged tile 0 into buf0; stage tile 1 into buf1
+    if (1 < ntiles) {
+      stage_a_load<TA>(g, m0, k_lo + 1 * BK, ra);
+      stage_b_load<TB>(g, n0, k_lo + 1 * BK, rb);
+      stage_a_write<TA>(As(1), ra);
+      stage_b_write<TB>(Bs(1), rb);
     }
     __syncthreads();
-    cur ^= 1;
+    for (long p = 0; p < ntiles; p += 2) {
+      const long t_next0 = p + 2, t_next1 = p + 3;
+      if (t_next0 < ntiles) {
+        stage_a_load<TA>(g, m0, k_lo + t_next0 * BK, ra);
+        stage_b_load<TB>(g, n0, k_lo + t_next0 * BK, rb);
+      }
+      float ra2[16], rb2[16];
+      if (t_next1 < ntiles) {
+        stage_a_load<TA>(g, m0, k_lo + t_next1 * BK, ra2);
+        stage_b_load<TB>(g, n0, k_lo + t_next1 * BK, rb2);
+      }
+      const long t_mine = p + grp;
+      if (t_mine < ntiles) {
+        const float* Ab = As(grp);
+        const float* Bb = Bs(grp);
+        for (int kk = kk0; kk < kk1; kk += 2) {
+          const int krow = kk + ksel;
+          const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+          const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+          const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+          const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
+          acc00 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+          acc01 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+          acc10 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+          acc11 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+      }
+      __syncthreads();  // everyone done computing before rewriting buffers
+      if (t_next0 < ntiles) {
+        stage_a_write<TA>(As(0), ra);
+        stage_b_write<TB>(Bs(0), rb);
+      }
+      if (t_next1 < ntiles) {
+        stage_a_write<TA>(As(1), ra2);
+        stage_b_write<TB>(Bs(1), rb2);
+      }
+      __syncthreads();
+    }
   }
 
   if (kwaves > 1) {
