@@ -42,7 +42,11 @@ static bool conv_s1d1(const ConvolutionLayer& l) {
 // every chunk crosses an output row (slow masked path) — the explicit col
 // buffer wins there (measured: stage-4 3x3 at OW=14 ran at 68 vs ~90 TF)
 static bool conv_implicit(const ConvolutionLayer& l) {
-  return conv_s1d1(l) && (l.kh_ == 1 || l.OW_ >= 24);
+  static const int min_ow = [] {
+    const char* e = getenv("CAFFE_IMPLICIT_MIN_OW");
+    return e ? atoi(e) : 24;
+  }();
+  return conv_s1d1(l) && (l.kh_ == 1 || l.OW_ >= min_ow);
 }
 static bool conv_is_1x1(const ConvolutionLayer& l) {
   return l.kh_ == 1 && l.kw_ == 1 && conv_s1d1(l) && !l.ph_ && !l.pw_ &&
