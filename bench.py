@@ -264,15 +264,16 @@ def main():
 
     value = world * args.batch * args.steps / elapsed
 
+    # roofline leg: 3 instrumented steps, HIP-event timed per kernel class
+    # on the launching stream.  ALL ranks step (the collectives are inside
+    # the step — a rank-0-only step would deadlock the communicator).
+    ca.perf_reset()
+    ca.set_perf_timing(True)
+    solver.step(3)
+    ca.device_synchronize()
+    ca.set_perf_timing(False)
     result = None
     if rank == 0:
-        # roofline leg: 3 instrumented steps, HIP-event timed per kernel
-        # class on the launching stream
-        ca.perf_reset()
-        ca.set_perf_timing(True)
-        solver.step(3)
-        ca.device_synchronize()
-        ca.set_perf_timing(False)
         perf = ca.perf_snapshot()
         g = {"flops": 0.0, "ns": 0.0, "launches": 0}
         for k, v in perf.items():
