@@ -292,6 +292,19 @@ ALL_CHECKS = {
     "conv_group": lambda m: check_conv(m, C=8, Co=16, k=5, p=2, grp=2),
     "conv_5x5_odd": lambda m: check_conv(m, N=3, C=5, H=11, W=17, Co=7, k=5,
                                          s=3, p=2),
+    # implicit-im2col eligible shapes (s1/d1, OW >= 24 on the GPU path):
+    # interior fast path + row-crossing boundary chunks + image edges
+    "conv_3x3_implicit": lambda m: check_conv(m, N=2, C=4, H=32, W=32, Co=8,
+                                              k=3, s=1, p=1),
+    "conv_3x3_implicit_odd": lambda m: check_conv(m, N=3, C=3, H=29, W=37,
+                                                  Co=6, k=3, s=1, p=1),
+    "conv_5x5_implicit_group": lambda m: check_conv(m, N=2, C=8, H=27, W=27,
+                                                    Co=16, k=5, s=1, p=2,
+                                                    grp=2),
+    "conv_7x7_implicit": lambda m: check_conv(m, N=2, C=3, H=30, W=30, Co=8,
+                                              k=7, s=1, p=3),
+    "conv_11x11s4": lambda m: check_conv(m, N=2, C=3, H=47, W=47, Co=8,
+                                         k=11, s=4, p=0),  # AlexNet conv1
     "ip": check_ip,
     "ip_large": lambda m: check_ip(m, M=130, K=260, Nout=140),
     "pool_max": lambda m: check_pool(m, "MAX"),
