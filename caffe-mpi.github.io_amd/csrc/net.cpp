@@ -329,6 +329,7 @@ void Net::ShareTrainedLayersWith(Net& other) {
 // ------------------------------------------------- snapshot interop
 // .caffemodel = binary NetParameter carrying each parametered layer's blobs
 // (reference Net::ToProto / CopyTrainedLayersFrom, net.cpp:1055-1248).
+#include <filesystem>
 #include <fstream>
 
 #include "proto_wire.hpp"
@@ -356,6 +357,8 @@ void Net::SaveWeights(const std::string& path) {
     }
     net_w.submsg(100, lw.out);
   }
+  const auto dir = std::filesystem::path(path).parent_path();
+  if (!dir.empty()) std::filesystem::create_directories(dir);
   std::ofstream f(path, std::ios::binary);
   CHECK_(f.good()) << "cannot write " << path;
   f.write(net_w.out.data(), (long)net_w.out.size());

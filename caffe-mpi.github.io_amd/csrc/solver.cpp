@@ -1,6 +1,7 @@
 #include "solver.hpp"
 
 #include <cmath>
+#include <filesystem>
 #include <fstream>
 
 #include "proto_wire.hpp"
@@ -276,6 +277,10 @@ void Solver::Snapshot() {
     sw.submsg(3, bw.out);
   }
   sw.vint(4, current_step_);
+  {
+    const auto dir = std::filesystem::path(state).parent_path();
+    if (!dir.empty()) std::filesystem::create_directories(dir);
+  }
   std::ofstream f(state, std::ios::binary);
   CHECK_(f.good()) << "cannot write " << state;
   f.write(sw.out.data(), (long)sw.out.size());
