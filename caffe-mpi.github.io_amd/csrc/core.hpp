@@ -175,15 +175,17 @@ class Engine {
   std::mutex perf_mu_;
 };
 
-// RAII perf timer for a kernel launch region on a stream.
+// RAII perf timer for a kernel launch region on a stream.  The string
+// constructor resolves the class each call; hot launchers should pass a
+// cached PerfClass* (PERF_CLASS macro) — the registry lookup + string
+// construction otherwise runs ~2000x per training step on the host path.
 struct PerfScope {
   PerfClass* pc = nullptr;
   hipStream_t s = nullptr;
   hipEvent_t a = nullptr, b = nullptr;
-  PerfScope(const std::string& name, hipStream_t stream, double flops,
-            double bytes) {
+  PerfScope(PerfClass* cls, hipStream_t stream, double flops, double bytes) {
     Engine& E = Engine::get();
-    pc = &E.perf(name);
+    pc = cls;
     pc->add(flops, bytes);
     if (E.perf_timing && E.mode == Mode::GPU) {
       s = stream;
@@ -192,6 +194,9 @@ struct PerfScope {
       HIP_CHECK(hipEventRecord(a, s));
     }
   }
+  PerfScope(const std::string& name, hipStream_t stream, double flops,
+            double bytes)
+      : PerfScope(&Engine::get().perf(name), stream, flops, bytes) {}
   ~PerfScope() {
     if (a) {
       HIP_CHECK(hipEventRecord(b, s));
@@ -199,6 +204,12 @@ struct PerfScope {
     }
   }
 };
+// per-call-site cached perf class (static local: resolved once)
+#define PERF_CLASS(name)                                        \
+  ([]() -> ::camd::PerfClass* {                                 \
+    static ::camd::PerfClass* pc_ = &::camd::Engine::get().perf(name); \
+    return pc_;                                                 \
+  }())
 
 // ---------------------------------------------------------- SyncedMemory
 // Lazy host/device mirror with the reference's head-state machine

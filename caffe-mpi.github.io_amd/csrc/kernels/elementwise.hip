@@ -57,7 +57,7 @@ void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
                     int W, int kh, int kw, int ph, int pw, int sh, int sw,
                     int dh, int dw, int OH, int OW, long Spad, float* col) {
   const long total = (long)C * kh * kw * Nimg * Spad;
-  PerfScope perf("im2col", s, 0,
+  PerfScope perf(PERF_CLASS("im2col"), s, 0,
                  8.0 * total);  // ~1 read + 1 write per element
   const int bx = (int)std::min<long>((Spad + TPB - 1) / TPB, 16);
   dim3 grid(bx, C * kh * kw, Nimg);
@@ -102,7 +102,7 @@ void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
                     int W, int kh, int kw, int ph, int pw, int sh, int sw,
                     int dh, int dw, int OH, int OW, long Spad, float* dx) {
   const long total = (long)Nimg * C * H * W;
-  PerfScope perf("col2im", s, 0, 8.0 * total * kh * kw / (sh * sw));
+  PerfScope perf(PERF_CLASS("col2im"), s, 0, 8.0 * total * kh * kw / (sh * sw));
   const int bx = (int)std::min<long>(((long)H * W + TPB - 1) / TPB, 16);
   dim3 grid(bx, C, Nimg);
   hipLaunchKernelGGL(k_col2im_b, grid, dim3(TPB), 0, s, dcol, C, H, W, kh,
@@ -126,7 +126,7 @@ __global__ void k_nchw_to_cpad(const float* __restrict__ y, int Nimg, int C,
 void nchw_to_cpad(hipStream_t s, const float* y, int Nimg, int C, long S,
                   long Spad, float* out) {
   const long total = (long)C * Nimg * Spad;
-  PerfScope perf("transpose", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("transpose"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_nchw_to_cpad, dim3(nblocks(total, 4)), dim3(TPB), 0,
                      s, y, Nimg, C, S, Spad, out);
 }
@@ -149,7 +149,7 @@ __global__ void k_relu_fwd(const f4* __restrict__ x, long n4, float slope,
   }
 }
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y) {
-  PerfScope perf("relu", s, 0, 8.0 * n);
+  PerfScope perf(PERF_CLASS("relu"), s, 0, 8.0 * n);
   const long n4 = (n + 3) / 4;  // blobs are 64B-padded
   hipLaunchKernelGGL(k_relu_fwd, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, n4, slope, (f4*)y);
@@ -168,7 +168,7 @@ __global__ void k_relu_bwd(const f4* __restrict__ x,
 }
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
               float slope, float* dx) {
-  PerfScope perf("relu", s, 0, 12.0 * n);
+  PerfScope perf(PERF_CLASS("relu"), s, 0, 12.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_relu_bwd, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, (const f4*)dy, n4, slope, (f4*)dx);
@@ -207,7 +207,7 @@ void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
                   int kh, int kw, int ph, int pw, int sh, int sw, int OH,
                   int OW, float* y, int* mask) {
   const long total = (long)N * C * OH * OW;
-  PerfScope perf("pool", s, 0, 4.0 * total * (kh * kw + 2));
+  PerfScope perf(PERF_CLASS("pool"), s, 0, 4.0 * total * (kh * kw + 2));
   hipLaunchKernelGGL(k_pool_max_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y,
                      mask);
@@ -243,7 +243,7 @@ void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
                   int C, int H, int W, int kh, int kw, int ph, int pw,
                   int sh, int sw, int OH, int OW, float* dx) {
   const long total = (long)N * C * H * W;
-  PerfScope perf("pool", s, 0, 12.0 * total);
+  PerfScope perf(PERF_CLASS("pool"), s, 0, 12.0 * total);
   hipLaunchKernelGGL(k_pool_max_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, dy, mask, N, C, H, W, kh, kw, ph, pw, sh, sw, OH,
                      OW, dx);
@@ -276,7 +276,7 @@ void pool_ave_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
                   int kh, int kw, int ph, int pw, int sh, int sw, int OH,
                   int OW, float* y) {
   const long total = (long)N * C * OH * OW;
-  PerfScope perf("pool", s, 0, 4.0 * total * (kh * kw + 1));
+  PerfScope perf(PERF_CLASS("pool"), s, 0, 4.0 * total * (kh * kw + 1));
   hipLaunchKernelGGL(k_pool_ave_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y);
 }
@@ -310,7 +310,7 @@ void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
                   int kh, int kw, int ph, int pw, int sh, int sw, int OH,
                   int OW, float* dx) {
   const long total = (long)N * C * H * W;
-  PerfScope perf("pool", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("pool"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_pool_ave_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, dy, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
 }
@@ -355,7 +355,7 @@ __global__ void k_bn_fwd_stats(const float* __restrict__ x, int N, int C,
 
 void bn_fwd_stats(hipStream_t s, const float* x, int N, int C, long S,
                   int nb, void* partials) {
-  PerfScope perf("bn", s, 0, 4.0 * N * C * S);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, 4.0 * N * C * S);
   hipLaunchKernelGGL(k_bn_fwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, N, C,
                      S, nb, (double2*)partials);
 }
@@ -426,7 +426,7 @@ void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
                  int sb, int N, int C, long S, float* y, int fuse_relu) {
   const long total = (long)N * C * S;
-  PerfScope perf("bn", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, 8.0 * total);
   // blobs are 16-float padded: the final partial pack reads/writes pad
   // space with wrapped (but in-bounds) channel indices — harmless
   const long n4 = (total + 3) / 4;
@@ -475,7 +475,7 @@ void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,
                  const float* gvar, const float* scale, const float* bias,
                  int sb, int N, int C, long S, float eps, float* y) {
   const long total = (long)N * C * S;
-  PerfScope perf("bn", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_bn_fwd_test, dim3(nblocks(total, 8)), dim3(TPB), 0,
                      s, x, gmean, gvar, scale, bias, sb, N, C, S, eps, y);
 }
@@ -526,7 +526,7 @@ void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std, int N, int C,
                   long S, int nb, const float* scale, const float* bias,
                   int frelu, void* partials) {
-  PerfScope perf("bn", s, 0, 8.0 * N * C * S);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, 8.0 * N * C * S);
   hipLaunchKernelGGL(k_bn_bwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, dy,
                      mean, inv_std, N, C, S, nb, scale, bias, frelu,
                      (double2*)partials);
@@ -611,7 +611,7 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* m_dyxn, int N, int C, long S,
                   const float* bias, int frelu, float* dx) {
   const long total = (long)N * C * S;
-  PerfScope perf("bn", s, 0, 12.0 * total);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, 12.0 * total);
   const long n4 = (total + 3) / 4;
   hipLaunchKernelGGL(k_bn_bwd_apply, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, (const f4*)dy, mean, inv_std, scale, sb,
@@ -660,7 +660,7 @@ void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
              int size, float alpha, float beta, float k, float* scale,
              float* y) {
   const long S = (long)H * W;
-  PerfScope perf("lrn", s, 0, 12.0 * N * C * S);
+  PerfScope perf(PERF_CLASS("lrn"), s, 0, 12.0 * N * C * S);
   hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, N,
                      C, S, size, alpha / size, beta, k, scale, y);
 }
@@ -705,7 +705,7 @@ void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
              const float* scale, int N, int C, int H, int W, int size,
              float alpha, float beta, float* dx) {
   const long S = (long)H * W;
-  PerfScope perf("lrn", s, 0, 20.0 * N * C * S);
+  PerfScope perf(PERF_CLASS("lrn"), s, 0, 20.0 * N * C * S);
   hipLaunchKernelGGL(k_lrn_bwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, y,
                      dy, scale, N, C, S, size, 2.f * alpha * beta / size,
                      beta, dx);
@@ -744,7 +744,7 @@ __global__ void k_softmax_fwd(const float* __restrict__ x, int outer, int C,
 void softmax_fwd(hipStream_t s, const float* x, int outer, int C, int inner,
                  float* prob) {
   const long rows = (long)outer * inner;
-  PerfScope perf("softmax", s, 0, 12.0 * rows * C);
+  PerfScope perf(PERF_CLASS("softmax"), s, 0, 12.0 * rows * C);
   const int blocks = (int)std::min<long>((rows + 3) / 4, 2048);
   hipLaunchKernelGGL(k_softmax_fwd, dim3(blocks), dim3(TPB), 0, s, x, outer,
                      C, inner, prob);
@@ -779,7 +779,7 @@ __global__ void k_sm_loss(const float* __restrict__ prob,
 }
 void softmaxloss_fwd(hipStream_t s, const float* prob, const float* label,
                      int outer, int C, int inner, float* loss_out) {
-  PerfScope perf("softmax", s, 0, 8.0 * outer * inner);
+  PerfScope perf(PERF_CLASS("softmax"), s, 0, 8.0 * outer * inner);
   hipLaunchKernelGGL(k_sm_loss, dim3(1), dim3(TPB), 0, s, prob, label,
                      outer, C, inner, (float)((long)outer * inner),
                      loss_out);
@@ -802,7 +802,7 @@ __global__ void k_sm_loss_bwd(const float* __restrict__ prob,
 void softmaxloss_bwd(hipStream_t s, const float* prob, const float* label,
                      int outer, int C, int inner, float scale, float* dx) {
   const long total = (long)outer * C * inner;
-  PerfScope perf("softmax", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("softmax"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_sm_loss_bwd, dim3(nblocks(total, 4)), dim3(TPB), 0,
                      s, prob, label, outer, C, inner, scale, dx);
 }
@@ -826,7 +826,7 @@ __global__ void k_rowsum(const float* __restrict__ A, long M, long N,
   }
 }
 void rowsum(hipStream_t s, const float* A, long M, long N, float* out) {
-  PerfScope perf("reduce", s, 0, 4.0 * M * N);
+  PerfScope perf(PERF_CLASS("reduce"), s, 0, 4.0 * M * N);
   hipLaunchKernelGGL(k_rowsum, dim3((int)std::min<long>(M, 2048)),
                      dim3(TPB), 0, s, A, M, N, out);
 }
@@ -840,7 +840,7 @@ __global__ void k_colsum(const float* __restrict__ A, long M, long N,
   }
 }
 void colsum(hipStream_t s, const float* A, long M, long N, float* out) {
-  PerfScope perf("reduce", s, 0, 4.0 * M * N);
+  PerfScope perf(PERF_CLASS("reduce"), s, 0, 4.0 * M * N);
   hipLaunchKernelGGL(k_colsum, dim3(nblocks(N)), dim3(TPB), 0, s, A, M, N,
                      out);
 }
@@ -852,7 +852,7 @@ __global__ void k_axpby(long n, float a, const float* __restrict__ x,
 }
 void axpby(hipStream_t s, long n, float a, const float* x, float b,
            float* y) {
-  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
   hipLaunchKernelGGL(k_axpby, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, a, x,
                      b, y);
 }
@@ -863,7 +863,7 @@ __global__ void k_copy(long n4, const f4* __restrict__ x,
 }
 void copy(hipStream_t s, long n, const float* x, float* y) {
   if (x == y) return;
-  PerfScope perf("eltwise", s, 0, 8.0 * n);
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 8.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_copy, dim3(nblocks(n4, 4)), dim3(TPB), 0, s, n4,
                      (const f4*)x, (f4*)y);
@@ -890,7 +890,7 @@ __global__ void k_add3(long n4, const f4* __restrict__ a,
 }
 void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
           int fuse_relu) {
-  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_add3, dim3(nblocks(n4, 4)), dim3(TPB), 0, s, n4,
                      (const f4*)a, (const f4*)b, fuse_relu, (f4*)y);
@@ -901,7 +901,7 @@ __global__ void k_acc(long n4, const f4* __restrict__ x,
   VEC_GRID(i, n4) y[i] += x[i];
 }
 void acc(hipStream_t s, long n, const float* x, float* y) {
-  PerfScope perf("eltwise", s, 0, 12.0 * n);
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_acc, dim3(nblocks(n4, 4)), dim3(TPB), 0, s, n4,
                      (const f4*)x, (f4*)y);
@@ -920,7 +920,7 @@ __global__ void k_concat_fwd(const float* __restrict__ x, int N, int Cs,
 void concat_fwd(hipStream_t s, const float* x, int N, int Cs, long S, int Cd,
                 int c_off, float* y) {
   const long total = (long)N * Cs * S;
-  PerfScope perf("concat", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("concat"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_concat_fwd, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
                      x, N, Cs, S, Cd, c_off, y);
 }
@@ -938,7 +938,7 @@ __global__ void k_concat_bwd(const float* __restrict__ dy, int N, int Cs,
 void concat_bwd(hipStream_t s, const float* dy, int N, int Cs, long S,
                 int Cd, int c_off, float* dx) {
   const long total = (long)N * Cs * S;
-  PerfScope perf("concat", s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("concat"), s, 0, 8.0 * total);
   hipLaunchKernelGGL(k_concat_bwd, dim3(nblocks(total, 4)), dim3(TPB), 0, s,
                      dy, N, Cs, S, Cd, c_off, dx);
 }
@@ -967,7 +967,7 @@ __global__ void k_dropout_fwd(const float* __restrict__ x, long n,
 void dropout_fwd(hipStream_t s, const float* x, long n, uint64_t seed,
                  uint64_t counter, float threshold, float scale, float* y,
                  uint8_t* mask) {
-  PerfScope perf("dropout", s, 0, 9.0 * n);
+  PerfScope perf(PERF_CLASS("dropout"), s, 0, 9.0 * n);
   (void)counter;  // caller passes the pre-mixed key via `seed`
   hipLaunchKernelGGL(k_dropout_fwd, dim3(nblocks(n, 4)), dim3(TPB), 0, s, x,
                      n, seed, threshold, scale, y, mask);
@@ -980,7 +980,7 @@ __global__ void k_dropout_bwd(const float* __restrict__ dy,
 }
 void dropout_bwd(hipStream_t s, const float* dy, const uint8_t* mask, long n,
                  float scale, float* dx) {
-  PerfScope perf("dropout", s, 0, 9.0 * n);
+  PerfScope perf(PERF_CLASS("dropout"), s, 0, 9.0 * n);
   hipLaunchKernelGGL(k_dropout_bwd, dim3(nblocks(n, 4)), dim3(TPB), 0, s,
                      dy, mask, n, scale, dx);
 }
@@ -1007,7 +1007,7 @@ void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
   // arena offsets are 16-float aligned and blob memory is 64B-padded, so
   // the float4 body may run over the pad (pad floats update to garbage in
   // h; g pad stays 0; w pad unused) — correct for all live elements
-  PerfScope perf("sgd", s, 0, 20.0 * n);
+  PerfScope perf(PERF_CLASS("sgd"), s, 0, 20.0 * n);
   const long n4 = (n + 3) / 4;
   hipLaunchKernelGGL(k_sgd, dim3(nblocks(n4, 4)), dim3(TPB), 0, s, n4,
                      (f4*)g, (f4*)w, (f4*)h, mom, lr, decay, gscale);
@@ -1021,7 +1021,7 @@ __global__ void k_fill_uniform(long n, uint64_t key, float lo, float hi,
 }
 void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
                   float lo, float hi, float* y) {
-  PerfScope perf("data", s, 0, 4.0 * n);
+  PerfScope perf(PERF_CLASS("data"), s, 0, 4.0 * n);
   (void)counter;
   hipLaunchKernelGGL(k_fill_uniform, dim3(nblocks(n, 4)), dim3(TPB), 0, s,
                      n, seed, lo, hi, y);

@@ -486,12 +486,20 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   }
   const char* pcls = !transA ? (!transB ? "gemm_nn" : "gemm_nt")
                             : (!transB ? "gemm_tn" : "gemm_tt");
-  std::string pname = pcls;
   static const bool by_shape = getenv("CAFFE_GEMM_BY_SHAPE") != nullptr;
-  if (by_shape)
-    pname += "_" + std::to_string(M) + "x" + std::to_string(N) + "x" +
-             std::to_string(K);
-  PerfScope perf(pname, s, 2.0 * M * N * K,
+  PerfClass* pc;
+  if (by_shape) {
+    pc = &Engine::get().perf(std::string(pcls) + "_" + std::to_string(M) +
+                             "x" + std::to_string(N) + "x" +
+                             std::to_string(K));
+  } else {
+    static PerfClass* nn = &Engine::get().perf("gemm_nn");
+    static PerfClass* nt = &Engine::get().perf("gemm_nt");
+    static PerfClass* tn = &Engine::get().perf("gemm_tn");
+    static PerfClass* tt = &Engine::get().perf("gemm_tt");
+    pc = !transA ? (!transB ? nn : nt) : (!transB ? tn : tt);
+  }
+  PerfScope perf(pc, s, 2.0 * M * N * K,
                  4.0 * (M * K + N * K + M * N));
   dim3 grid((unsigned)g.tiles, 1, (unsigned)SK);
   dim3 block(256);
@@ -603,7 +611,7 @@ __global__ void k_bias_grad_fin(const double* __restrict__ part, int nb,
 }
 void bias_grad(hipStream_t s, const float* dy, int N, int C, long S,
                float* db) {
-  PerfScope perf("reduce", s, 0, 4.0 * N * C * S);
+  PerfScope perf(PERF_CLASS("reduce"), s, 0, 4.0 * N * C * S);
   const int nb = std::max(1, std::min(N, 2048 / std::max(1, C)));
   double* part = (double*)Workspace::get_global().get(
       12, sizeof(double) * (size_t)C * nb);
