@@ -99,6 +99,23 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
       }
       return;
     }
+    if (v.OW >= 16) {
+      // at most ONE row wrap inside the 16-chunk: all 16 loads become
+      // independent (the sequential walk would chain their addresses)
+      const int jw = v.OW - ow;       // first j on the next output row
+      const int w0 = w;               // input col at j=0
+      const int wreset = -v.pw + kj;  // input col after the wrap
+#pragma unroll
+      for (int j = 0; j < 16; ++j) {
+        const bool wrapped = j >= jw;
+        const int hj = h + (wrapped ? 1 : 0);
+        const int wj = wrapped ? wreset + (j - jw) : w0 + j;
+        const bool ok =
+            j < smax && hj >= 0 && hj < v.H && wj >= 0 && wj < v.W;
+        out[j] = ok ? xp[hj * v.W + wj] : 0.f;
+      }
+      return;
+    }
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       const bool ok = j < smax && h >= 0 && h < v.H && w >= 0 && w < v.W;
