@@ -110,28 +110,6 @@ void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
                      (long)Nimg * Spad, dx);
 }
 
-// y[N][C][S] -> out[C][N*Spad] (zero pad)
-__global__ void k_nchw_to_cpad(const float* __restrict__ y, int Nimg, int C,
-                               long S, long Spad, float* __restrict__ out) {
-  const long total = (long)C * Nimg * Spad;
-  GRID_STRIDE(idx, total) {
-    const long c = idx / ((long)Nimg * Spad);
-    const long rem = idx - c * (long)Nimg * Spad;
-    const int n = (int)(rem / Spad);
-    const long sp = rem - (long)n * Spad;
-    out[idx] = sp < S ? y[((long)n * C + c) * S + sp] : 0.f;
-  }
-}
-
-void nchw_to_cpad(hipStream_t s, const float* y, int Nimg, int C, long S,
-                  long Spad, float* out) {
-  const long total = (long)C * Nimg * Spad;
-  PerfScope perf(PERF_CLASS("transpose"), s, 0, 8.0 * total);
-  hipLaunchKernelGGL(k_nchw_to_cpad, dim3(nblocks(total, 4)), dim3(TPB), 0,
-                     s, y, Nimg, C, S, Spad, out);
-}
-
-
 // ---- float4 elementwise helpers (tensors are 64B-padded, so the float4
 // body covers n/4*4 and a scalar tail handles the rest)
 using f4 = __attribute__((ext_vector_type(4))) float;
@@ -808,29 +786,6 @@ void softmaxloss_bwd(hipStream_t s, const float* prob, const float* label,
 }
 
 // ------------------------------------------------------- row/col sums
-__global__ void k_rowsum(const float* __restrict__ A, long M, long N,
-                         float* __restrict__ out) {
-  for (long m = blockIdx.x; m < M; m += gridDim.x) {
-    const float* a = A + m * N;
-    double acc = 0;
-    for (long n = threadIdx.x; n < N; n += blockDim.x) acc += a[n];
-    __shared__ double sh[TPB];
-    sh[threadIdx.x] = acc;
-    __syncthreads();
-    for (int off = TPB / 2; off > 0; off >>= 1) {
-      if (threadIdx.x < off) sh[threadIdx.x] += sh[threadIdx.x + off];
-      __syncthreads();
-    }
-    if (threadIdx.x == 0) out[m] = (float)sh[0];
-    __syncthreads();
-  }
-}
-void rowsum(hipStream_t s, const float* A, long M, long N, float* out) {
-  PerfScope perf(PERF_CLASS("reduce"), s, 0, 4.0 * M * N);
-  hipLaunchKernelGGL(k_rowsum, dim3((int)std::min<long>(M, 2048)),
-                     dim3(TPB), 0, s, A, M, N, out);
-}
-
 __global__ void k_colsum(const float* __restrict__ A, long M, long N,
                          float* __restrict__ out) {
   GRID_STRIDE(n, N) {

@@ -2,11 +2,11 @@
 // GEMM (kernels/gemm_f32.hip) and the HBM-bound kernels
 // (kernels/elementwise.hip).  Conv strategy (MI355X-first, replaces the
 // reference's per-image GEMM loop conv_layer.cu:14-21 which is
-// launch/sync-bound):
-//   whole-batch col buffer col[K][Nimg*Spad] (Spad = S rounded to 64) →
-//   ONE GEMM per direction per layer; the GEMM epilogue scatters straight
-//   into NCHW and fuses the bias, so no separate bias pass.
-// Workspace slots: 0 = col, 1 = dcol, 2 = dY in [Cout][Nimg*Spad] layout.
+// launch/sync-bound): whole-batch GEMMs, NCHW-view operands (implicit
+// im2col for s1/d1 at OW>=24), fused bias/ReLU/scatter epilogues, explicit
+// cached col buffers only for strided/small-OW convs.
+// Workspace slots: 1 = dcol, 10 = split-K slabs, 11 = flipped weights,
+// 12 = bias-grad partials, 100+ = per-conv-layer cached col buffers.
 #include "layers.hpp"
 
 namespace camd {

@@ -92,9 +92,6 @@ void im2col_batched(hipStream_t s, const float* x, int Nimg, int C, int H,
 void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
                     int W, int kh, int kw, int ph, int pw, int sh, int sw,
                     int dh, int dw, int OH, int OW, long Spad, float* dx);
-// out[C][Nimg*Spad] from y[Nimg][C][S] (pad columns zeroed)
-void nchw_to_cpad(hipStream_t s, const float* y, int Nimg, int C, long S,
-                  long Spad, float* out);
 
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y);
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
@@ -161,9 +158,7 @@ void softmaxloss_fwd(hipStream_t s, const float* prob, const float* label,
 void softmaxloss_bwd(hipStream_t s, const float* prob, const float* label,
                      int outer, int C, int inner, float scale, float* dx);
 
-// out[m] = Σ_n A[m][n] (deterministic block reduce) — conv bias grad
-void rowsum(hipStream_t s, const float* A, long M, long N, float* out);
-// out[n] = Σ_m A[m][n] — IP bias grad
+// out[n] = Σ_m A[m][n] — IP bias grad (deterministic)
 void colsum(hipStream_t s, const float* A, long M, long N, float* out);
 
 void axpby(hipStream_t s, long n, float a, const float* x, float b, float* y);
