@@ -1,5 +1,6 @@
 #include "solver.hpp"
 
+#include <chrono>
 #include <cmath>
 #include <filesystem>
 #include <fstream>
@@ -179,6 +180,7 @@ void Reducer::iteration_end(hipEvent_t backward_done) {
 
 void Solver::Step(int iters) {
   Engine& E = Engine::get();
+  using clock = std::chrono::steady_clock;
   const long display = param_->inum("display", 0);
   const long test_interval = param_->inum("test_interval", 0);
   const long test_iter = param_->inum("test_iter", 1);
@@ -186,6 +188,16 @@ void Solver::Step(int iters) {
   // this fork always runs a 1-iter test at iter 0 regardless of
   // test_initialization (reference solver.cpp:243-248)
   if (iter_ == 0 && test_interval > 0 && test_net()) TestAll(1);
+  // reference-style perf accounting (solver.cpp:299 skips warmup): the
+  // first Step call (warmup/compile) is excluded; later calls are timed at
+  // call granularity with a single sync so the launch-ahead pipeline stays
+  // intact (no per-iteration host synchronization in this engine)
+  const bool timed = skipped_iters_ > 0;
+  std::chrono::time_point<clock> tstep0;
+  if (timed) {
+    E.sync();
+    tstep0 = clock::now();
+  }
   for (int i = 0; i < iters; ++i) {
     if (test_interval > 0 && iter_ > 0 && iter_ % test_interval == 0 &&
         test_net())
@@ -207,12 +219,34 @@ void Solver::Step(int iters) {
               iter_, l, cur_lr_);
     }
   }
-  (void)E;
+  if (timed) {
+    E.sync();
+    perf_seconds_ +=
+        std::chrono::duration<double>(clock::now() - tstep0).count();
+    perf_iters_ += iters;
+  } else {
+    skipped_iters_ += iters;
+  }
 }
 
 std::shared_ptr<Solver> create_solver_from_file(const std::string& path,
                                                 int batch_override) {
   return std::make_shared<Solver>(parse_prototxt_file(path), batch_override);
+}
+
+void Solver::print_perf_report() const {
+  if (perf_iters_ == 0 || perf_seconds_ <= 0) return;
+  // batch size from the train net's data layer (first layer top 0)
+  long batch = 0;
+  for (auto& l : net_->layers())
+    if (auto* d = dynamic_cast<const DataLayer*>(l.get())) {
+      batch = d->batch_;
+      break;
+    }
+  const double ratio = perf_iters_ / perf_seconds_;
+  fprintf(stderr,
+          "Solver performance on device %d: %.4g * %ld = %.5g img/sec\n",
+          Engine::get().device, ratio, batch, ratio * batch);
 }
 
 Net* Solver::test_net() {

@@ -62,6 +62,9 @@ class Solver {
   Net* test_net();  // lazily built TEST-phase net sharing train weights
   void TestAll(long iters);  // solver.cpp:439-540 semantics
   void Snapshot();           // .caffemodel + .solverstate (solver.cpp:542)
+  // reference per-GPU perf report (solver.cpp:619-628): iters/sec x batch,
+  // skipping the first two measured iterations like the reference (:299)
+  void print_perf_report() const;
   void Restore(const std::string& state_path);
   void LoadWeights(const std::string& model_path) {
     net_->LoadWeights(model_path);
@@ -92,6 +95,10 @@ class Solver {
   mutable int current_step_ = 0;
   float* history_ = nullptr;  // device arena, diff-arena layout
   std::vector<float> host_history_;
+  // perf-report bookkeeping
+  double perf_seconds_ = 0.0;
+  long perf_iters_ = 0;
+  long skipped_iters_ = 0;
 
   friend class Reducer;
 };

@@ -95,6 +95,7 @@ int main(int argc, char** argv) {
         todo = atol(flags["iterations"].c_str());
       CHECK_GT_(todo, 0);
       solver.Step((int)todo);
+      solver.print_perf_report();
       if (solver.param()->boolean("snapshot_after_train", true))
         solver.Snapshot();
       fprintf(stderr, "Optimization Done.\n");
