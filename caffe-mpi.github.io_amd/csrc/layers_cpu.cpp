@@ -612,6 +612,10 @@ void BatchNormLayer::Backward_cpu(const std::vector<Blob*>& top,
   const float* inv = inv_std_.cpu_data();
   const float* sc = scale_bias_ ? blobs_[3]->cpu_data() : nullptr;
   float* dx = prop_down[0] ? bottom[0]->mutable_cpu_diff() : nullptr;
+  // hoist the lazy head-state transitions out of the parallel region
+  // (mutable_cpu_diff inside the omp loop raced on first allocation)
+  float* dscale = scale_bias_ ? blobs_[3]->mutable_cpu_diff() : nullptr;
+  float* dbias = scale_bias_ ? blobs_[4]->mutable_cpu_diff() : nullptr;
 #pragma omp parallel for schedule(static)
   for (int c = 0; c < C_; ++c) {
     double s_dy = 0, s_dyxn = 0;
@@ -625,8 +629,8 @@ void BatchNormLayer::Backward_cpu(const std::vector<Blob*>& top,
       }
     }
     if (scale_bias_) {
-      blobs_[3]->mutable_cpu_diff()[c] = (float)s_dyxn;
-      blobs_[4]->mutable_cpu_diff()[c] = (float)s_dy;
+      dscale[c] = (float)s_dyxn;
+      dbias[c] = (float)s_dy;
     }
     if (!dx) continue;
     const float scc = scale_bias_ ? sc[c] : 1.f;
