@@ -165,6 +165,7 @@ class ConvolutionLayer : public Layer {
   long S_ = 0, Spad_ = 0;  // spatial count and 64-padded count
   int col_slot_ = -1;      // per-layer cached col buffer (fwd fills,
                            // bwd reuses — no im2col recompute)
+  bool fuse_relu_ = false;  // in-place ReLU folded into the GEMM epilogue
 };
 
 class InnerProductLayer : public Layer {

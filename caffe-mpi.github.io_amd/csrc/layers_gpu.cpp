@@ -68,6 +68,7 @@ void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   epi.S = S_;
   epi.n_stride = (long)Cout_ * S_;
   epi.bias = bias_ ? blobs_[1]->gpu_data() : nullptr;
+  epi.relu = fuse_relu_;
   if (conv_is_1x1(*this)) {
     GemmView xv{Spad_, S_, C_};  // plain NCHW view
     gpu::gemm(E.stream, false, false, Cout_, NS, K, 1.f, w, K, x, 0, 0.f, y,

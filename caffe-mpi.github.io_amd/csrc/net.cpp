@@ -213,6 +213,10 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
         bn->fuse_relu_ = true;
         relu->fused_away_ = true;
         relu->bwd_fused_ = true;  // BN backward masks dy by the activation
+      } else if (auto* cv =
+                     dynamic_cast<ConvolutionLayer*>(layers_[i - 1].get())) {
+        cv->fuse_relu_ = true;  // GemmEpi.relu (AlexNet/GoogLeNet pattern)
+        relu->fused_away_ = true;
       } else if (auto* el =
                      dynamic_cast<EltwiseLayer*>(layers_[i - 1].get())) {
         bool ones = el->op_ == "SUM";
