@@ -26,6 +26,51 @@ def test_gemm_shapes_gpu(shape):
     check_ip("gpu", M=M, K=K, Nout=N)
 
 
+def _fullnet_check(model, shape, batch=2, blob_probe=None):
+    import os
+    import subprocess
+    import sys
+    import numpy as np
+    import caffe_amd as ca
+    from engine_util import REPO, relerr
+
+    gen = os.path.join(REPO, "models", "generated", f"{model}_solver.prototxt")
+    if not os.path.exists(gen):
+        subprocess.check_call([sys.executable,
+                               os.path.join(REPO, "models",
+                                            "gen_models.py")])
+    results = {}
+    for mode in ("cpu", "gpu"):
+        ca.set_mode(mode)
+        ca.set_synthetic_shape(*shape, 1000)
+        ca.set_random_seed(99)
+        s = ca.Solver(path=gen, batch_override=batch)
+        s.step(1)
+        blob = s.net.blob(blob_probe) if blob_probe else None
+        loss = s.loss()
+        params = [s.net.param(i) for i in range(0, s.net.num_params(), 8)]
+        results[mode] = (loss, blob, params)
+    lc, bc, pc = results["cpu"]
+    lg, bg, pg = results["gpu"]
+    assert abs(lc - lg) < 2e-3 * max(1.0, abs(lc)), (lc, lg)
+    if blob_probe is not None:
+        assert relerr(bg, bc) < 1e-3, relerr(bg, bc)
+    for a, b in zip(pc, pg):
+        na = float(np.linalg.norm(a))
+        nd = float(np.linalg.norm(b - a))
+        if na < 0.05:
+            continue  # zero-init biases carry only update noise
+        assert nd < 0.1 * na, (nd, na)
+
+
+def test_alexnet_fullnet_gpu_vs_cpu():
+    _fullnet_check("alexnet", (3, 227, 227), blob_probe="pool1")
+
+
+def test_googlenet_fullnet_gpu_vs_cpu():
+    _fullnet_check("googlenet", (3, 224, 224), blob_probe="pool1/3x3_s2")
+
+
 def test_resnet50_fullnet_gpu_vs_cpu():
     """Whole-graph integration parity: one training step of ResNet-50 at
     batch 2 in GPU mode vs CPU mode (identical seeds => identical synthetic
