@@ -145,6 +145,12 @@ int caffe_net_load_weights(caffe_net_t n, const char* path) {
   API_CATCH
 }
 
+int caffe_comm_selftest(void) {
+  API_TRY
+  return rccl_selftest();
+  API_CATCH
+}
+
 int caffe_comm_unique_id(uint8_t out[128]) {
   API_TRY
   rccl_unique_id(out);

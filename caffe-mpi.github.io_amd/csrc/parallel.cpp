@@ -76,6 +76,31 @@ std::unique_ptr<Comm> make_rccl_comm(int rank, int world, const void* uid) {
   return std::unique_ptr<Comm>(new RcclComm(rank, world, uid));
 }
 
+// world-1 smoke of the whole RCCL path (comm init + allreduce + bcast):
+// proves the librccl linkage and call sequence on the box without needing
+// multiple GPUs (the driver's 8-GPU run is the real collective test)
+int rccl_selftest() {
+  Engine& E = Engine::get();
+  CHECK_(E.mode == Mode::GPU) << "rccl selftest needs GPU mode";
+  ncclUniqueId id;
+  NCCL_CHECK(ncclGetUniqueId(&id));
+  RcclComm comm(0, 1, &id);
+  const long n = 4096;
+  float* buf = (float*)E.dalloc.alloc(n * sizeof(float));
+  std::vector<float> host(n);
+  for (long i = 0; i < n; ++i) host[i] = (float)(i % 97) * 0.5f;
+  HIP_CHECK(hipMemcpy(buf, host.data(), n * 4, hipMemcpyHostToDevice));
+  comm.allreduce(buf, n, E.stream);
+  comm.bcast(buf, n, 0, E.stream);
+  HIP_CHECK(hipStreamSynchronize(E.stream));
+  std::vector<float> out(n);
+  HIP_CHECK(hipMemcpy(out.data(), buf, n * 4, hipMemcpyDeviceToHost));
+  E.dalloc.release(buf, n * sizeof(float));
+  for (long i = 0; i < n; ++i)
+    CHECK_EQ_(out[i], host[i]) << "rccl world-1 allreduce mismatch at " << i;
+  return 0;
+}
+
 std::unique_ptr<Comm> make_callback_comm(HostAllreduceFn fn, void* ud,
                                          int world) {
   return std::unique_ptr<Comm>(new CallbackComm(fn, ud, world));

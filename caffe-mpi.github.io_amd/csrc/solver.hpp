@@ -25,6 +25,7 @@ struct Comm {
 };
 
 void rccl_unique_id(void* out_bytes128);
+int rccl_selftest();  // world-1 RCCL linkage/call smoke (GPU mode)
 std::unique_ptr<Comm> make_rccl_comm(int rank, int world,
                                      const void* uid_bytes128);
 using HostAllreduceFn = void (*)(float*, long, void*);
