@@ -66,6 +66,8 @@ int caffe_net_load_weights(caffe_net_t n, const char* path);
  * rank 0 generates the 128-byte id, the launcher distributes it, every
  * rank calls comm_init */
 int caffe_comm_unique_id(uint8_t out[128]);
+/* world-1 RCCL linkage/call smoke (GPU mode) */
+int caffe_comm_selftest(void);
 int caffe_comm_init(caffe_solver_t s, int rank, int world,
                     const uint8_t id[128]);
 /* initial weight broadcast from rank 0 (P2PSync::on_start,

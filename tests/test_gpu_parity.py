@@ -119,3 +119,13 @@ def test_resnet50_fullnet_gpu_vs_cpu():
             # lr-scaled gradient noise — no wiring signal there
             continue
         assert nd < 0.1 * na, (nd, na)
+
+
+def test_rccl_world1_selftest():
+    """RCCL comm init + allreduce + bcast on a world-1 communicator —
+    proves the collective call path/linkage on the box (the 8-GPU driver
+    run is the real multi-rank test; world-2 semantics are pinned by the
+    gloo CPU tests)."""
+    import caffe_amd as ca
+    ca.set_mode("gpu")
+    assert ca._lib.caffe_comm_selftest() == 0, ca._lib.caffe_last_error()
