@@ -968,6 +968,45 @@ void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
                      (f4*)g, (f4*)w, (f4*)h, mom, lr, decay, gscale);
 }
 
+// segmented fused SGD: one launch covers a whole bucket's params.
+// segs: sorted arena offsets; per-seg weight base pointer and lr/decay.
+// Elements in inter-param padding update harmlessly (their g is 0).
+__global__ void k_sgd_seg(long lo, long hi, float* __restrict__ g_arena,
+                          float* __restrict__ h_arena,
+                          const long* __restrict__ seg_off,
+                          float* const* __restrict__ w_ptrs,
+                          const float* __restrict__ lrs,
+                          const float* __restrict__ decays, int nseg,
+                          float mom, float gscale) {
+  for (long i = lo + blockIdx.x * (long)blockDim.x + threadIdx.x; i < hi;
+       i += (long)gridDim.x * blockDim.x) {
+    // binary search: largest k with seg_off[k] <= i
+    int a = 0, b = nseg - 1;
+    while (a < b) {
+      const int m = (a + b + 1) >> 1;
+      if (seg_off[m] <= i)
+        a = m;
+      else
+        b = m - 1;
+    }
+    float* w = w_ptrs[a] + (i - seg_off[a]);
+    float gi = g_arena[i] * gscale + decays[a] * *w;
+    gi = h_arena[i] = mom * h_arena[i] + lrs[a] * gi;
+    *w -= gi;
+    g_arena[i] = 0.f;
+  }
+}
+void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
+                          float* h_arena, const long* seg_off,
+                          float* const* w_ptrs, const float* lrs,
+                          const float* decays, int nseg, float mom,
+                          float gscale) {
+  PerfScope perf(PERF_CLASS("sgd"), s, 0, 20.0 * (hi - lo));
+  hipLaunchKernelGGL(k_sgd_seg, dim3(nblocks(hi - lo, 4)), dim3(TPB), 0, s,
+                     lo, hi, g_arena, h_arena, seg_off, w_ptrs, lrs, decays,
+                     nseg, mom, gscale);
+}
+
 // ------------------------------------------------------------ synthetic
 __global__ void k_fill_uniform(long n, uint64_t key, float lo, float hi,
                                float* __restrict__ y) {

@@ -186,6 +186,14 @@ void dropout_bwd(hipStream_t s, const float* dy, const uint8_t* mask, long n,
 void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
                 float mom, float lr, float decay, float gscale);
 
+// segmented fused SGD over a flat arena range (one launch per bucket);
+// lrs/decays already folded with the per-param multipliers by the caller
+void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
+                          float* h_arena, const long* seg_off,
+                          float* const* w_ptrs, const float* lrs,
+                          const float* decays, int nseg, float mom,
+                          float gscale);
+
 void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
                   float lo, float hi, float* y);
 void fill_labels(hipStream_t s, long n, uint64_t seed, uint64_t counter,

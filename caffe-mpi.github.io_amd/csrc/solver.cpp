@@ -116,6 +116,43 @@ void Reducer::param_ready(int param_id, hipEvent_t done) {
   if (span >= solver_->bucket_budget_) flush(done);
 }
 
+void Solver::ensure_seg_table() {
+  if (d_seg_off_) {
+    // refresh lr/decay rows only when the per-iteration coefficients moved
+    if (table_lr_ == cur_lr_ && table_decay_ == weight_decay_) return;
+  }
+  Engine& E = Engine::get();
+  auto& params = net_->learnable_params();
+  const int n = (int)params.size();
+  std::vector<long> offs(n);
+  std::vector<float*> wps(n);
+  std::vector<float> lrs(n), decays(n);
+  for (int i = 0; i < n; ++i) {
+    offs[i] = params[i].offset;
+    wps[i] = params[i].blob->mutable_gpu_data();
+    lrs[i] = cur_lr_ * params[i].lr_mult;
+    decays[i] = weight_decay_ * params[i].decay_mult;
+  }
+  if (!d_seg_off_) {
+    d_seg_off_ = (long*)E.dalloc.alloc(n * sizeof(long));
+    d_w_ptrs_ = (float**)E.dalloc.alloc(n * sizeof(float*));
+    d_lrs_ = (float*)E.dalloc.alloc(n * sizeof(float));
+    d_decays_ = (float*)E.dalloc.alloc(n * sizeof(float));
+    HIP_CHECK(hipMemcpyAsync(d_seg_off_, offs.data(), n * sizeof(long),
+                             hipMemcpyHostToDevice, E.comm_stream));
+    HIP_CHECK(hipMemcpyAsync(d_w_ptrs_, wps.data(), n * sizeof(float*),
+                             hipMemcpyHostToDevice, E.comm_stream));
+  }
+  // pageable async H2D stages through a bounce buffer at call time, so the
+  // local vectors need not outlive the call and no stream sync is needed
+  HIP_CHECK(hipMemcpyAsync(d_lrs_, lrs.data(), n * sizeof(float),
+                           hipMemcpyHostToDevice, E.comm_stream));
+  HIP_CHECK(hipMemcpyAsync(d_decays_, decays.data(), n * sizeof(float),
+                           hipMemcpyHostToDevice, E.comm_stream));
+  table_lr_ = cur_lr_;
+  table_decay_ = weight_decay_;
+}
+
 void Reducer::flush(hipEvent_t ev) {
   if (bucket_end_ == bucket_start_) return;
   Solver& S = *solver_;
@@ -131,14 +168,13 @@ void Reducer::flush(hipEvent_t ev) {
       S.comm_->allreduce(S.net().diff_arena() + first.offset, count,
                          E.comm_stream);
     }
-    for (long k = bucket_start_; k < bucket_end_; ++k) {
-      const auto& p = params[k];
-      gpu::sgd_update(E.comm_stream, p.count,
-                      S.net().diff_arena() + p.offset,
-                      p.blob->mutable_gpu_data(), S.history() + p.offset,
-                      S.cur_mom_, S.cur_lr_ * p.lr_mult,
-                      S.weight_decay_ * p.decay_mult, S.grad_scale_);
-    }
+    // bucket-fused update: one launch over the flat arena range
+    S.ensure_seg_table();
+    gpu::sgd_update_segmented(
+        E.comm_stream, first.offset, first.offset + count,
+        S.net().diff_arena(), S.history(), S.d_seg_off_, S.d_w_ptrs_,
+        S.d_lrs_, S.d_decays_, (int)params.size(), S.cur_mom_,
+        S.grad_scale_);
   } else {
     if (S.comm_ && S.comm_->world() > 1) {
       // CPU buckets: gather the param diffs into one flat range is

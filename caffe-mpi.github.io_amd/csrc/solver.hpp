@@ -96,6 +96,16 @@ class Solver {
   mutable int current_step_ = 0;
   float* history_ = nullptr;  // device arena, diff-arena layout
   std::vector<float> host_history_;
+  // device segment table for the bucket-fused update (GPU mode): arena
+  // offsets, per-param weight base pointers, lr/decay (mults folded in at
+  // flush time by scaling the per-iteration coefficients)
+  long* d_seg_off_ = nullptr;
+  float** d_w_ptrs_ = nullptr;
+  float* d_lrs_ = nullptr;
+  float* d_decays_ = nullptr;
+  std::vector<float> seg_lr_mult_, seg_decay_mult_;
+  float table_lr_ = -1.f, table_decay_ = -1.f;
+  void ensure_seg_table();
   // perf-report bookkeeping
   double perf_seconds_ = 0.0;
   long perf_iters_ = 0;
