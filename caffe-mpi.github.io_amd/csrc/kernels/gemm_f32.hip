@@ -281,54 +281,38 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   __syncthreads();
 
   int cur = 0;
-#define GEMM_COMPUTE_TILE()                                                \
-  do {                                                                     \
-    const float* Ab = As(cur);                                             \
-    const float* Bb = Bs(cur);                                             \
-    _Pragma("unroll") for (int kk = 0; kk < BK; kk += 2) {                 \
-      const int krow = kk + ksel;                                          \
-      const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];                \
-      const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];           \
-      const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];                \
-      const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];           \
-      acc00 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);\
-      acc01 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);\
-      acc10 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);\
-      acc11 = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);\
-    }                                                                      \
-  } while (0)
   if (kwaves == 1) {
-    // 2-deep register ring: tile t+2's loads are issued while tile t
-    // computes and t+1 sits in the second register set — two compute
-    // spans cover the HBM latency (the single-set form covered one).
-    // Two fixed register sets (ra/rb, ra2/rb2), loop unrolled in pairs
-    // (rule 20: no runtime-indexed register arrays).
-    float ra2[16], rb2[16];
-    if (1 < ntiles) {  // tile 1 into set0 (tile 0 already in LDS buf0)
-      stage_a_load<TA>(g, m0, k_lo + 1 * BK, ra);
-      stage_b_load<TB>(g, n0, k_lo + 1 * BK, rb);
-    }
-    for (long t = 0; t < ntiles; t += 2) {
-      if (t + 2 < ntiles) {
-        stage_a_load<TA>(g, m0, k_lo + (t + 2) * BK, ra2);
-        stage_b_load<TB>(g, n0, k_lo + (t + 2) * BK, rb2);
+    for (long t = 0; t < ntiles; ++t) {
+      if (t + 1 < ntiles) {  // issue next tile's global loads early (T14)
+        stage_a_load<TA>(g, m0, k_lo + (t + 1) * BK, ra);
+        stage_b_load<TB>(g, n0, k_lo + (t + 1) * BK, rb);
       }
-      GEMM_COMPUTE_TILE();  // tile t
+      {
+        const float* Ab = As(cur);
+        const float* Bb = Bs(cur);
+#pragma unroll
+        for (int kk = 0; kk < BK; kk += 2) {
+          const int krow = kk + ksel;
+          const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+          const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+          const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+          const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
+          acc00 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+          acc01 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+          acc10 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+          acc11 =
+              __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+        }
+      }
+      // single barrier per K-tile: writing buf[cur^1] here is safe — its
+      // last readers finished before the previous iteration's barrier; the
+      // barrier below publishes these writes for the next iteration's reads
       if (t + 1 < ntiles) {
         stage_a_write<TA>(As(cur ^ 1), ra);
         stage_b_write<TB>(Bs(cur ^ 1), rb);
-      }
-      __syncthreads();
-      cur ^= 1;
-      if (t + 1 >= ntiles) break;
-      if (t + 3 < ntiles) {
-        stage_a_load<TA>(g, m0, k_lo + (t + 3) * BK, ra);
-        stage_b_load<TB>(g, n0, k_lo + (t + 3) * BK, rb);
-      }
-      GEMM_COMPUTE_TILE();  // tile t+1
-      if (t + 2 < ntiles) {
-        stage_a_write<TA>(As(cur ^ 1), ra2);
-        stage_b_write<TB>(Bs(cur ^ 1), rb2);
       }
       __syncthreads();
       cur ^= 1;
