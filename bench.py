@@ -338,6 +338,13 @@ def main():
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
         }
+        # reference-style summary lines (solver.cpp:619-628, parallel.cpp:85)
+        per_gpu = value / world
+        print(f"Solver performance on device 0: "
+              f"{per_gpu / args.batch:.4g} * {args.batch} = "
+              f"{per_gpu:.5g} img/sec", file=sys.stderr)
+        print(f"Overall multi-GPU performance: {value:.5g} img/sec",
+              file=sys.stderr)
         print(json.dumps(result))
     if dist:
         dist.destroy_process_group()
