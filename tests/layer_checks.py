@@ -15,7 +15,7 @@ rng = np.random.default_rng(20250915)
 
 
 def check_conv(mode, N=2, C=8, H=13, W=13, Co=16, k=3, s=1, p=1, grp=1,
-               bias=True):
+               bias=True, d=1):
     x = rng.standard_normal((N, C, H, W)).astype(np.float32)
     w = (rng.standard_normal((Co, C // grp, k, k)) * 0.2).astype(np.float32)
     b = rng.standard_normal(Co).astype(np.float32) if bias else None
@@ -30,17 +30,19 @@ def check_conv(mode, N=2, C=8, H=13, W=13, Co=16, k=3, s=1, p=1, grp=1,
     stride: {s}
     pad: {p}
     group: {grp}
+    dilation: {d}
     bias_term: {"true" if bias else "false"}
   }}
 }}"""
-    y_ref = orc.conv_fwd(x, w, b, pad=(p, p), stride=(s, s), group=grp)
+    y_ref = orc.conv_fwd(x, w, b, pad=(p, p), stride=(s, s), dil=(d, d),
+                         group=grp)
     dy = rng.standard_normal(y_ref.shape).astype(np.float32)
     net, y = run_layer(mode, [(N, C, H, W)], body, [x],
                        params=[w, b] if bias else [w], top_diff=dy)
     assert relerr(y, y_ref) < TOL, f"conv fwd {relerr(y, y_ref)}"
     dx_ref, dw_ref, db_ref = orc.conv_bwd(x, w, dy, pad=(p, p),
-                                          stride=(s, s), group=grp,
-                                          want_db=bias)
+                                          stride=(s, s), dil=(d, d),
+                                          group=grp, want_db=bias)
     assert relerr(net.blob("in0", diff=True), dx_ref) < TOL, "conv dx"
     assert relerr(net.param(0, diff=True),
                   dw_ref.ravel()) < TOL, "conv dw"
@@ -290,6 +292,11 @@ ALL_CHECKS = {
     "conv_1x1": lambda m: check_conv(m, k=1, p=0, bias=False),
     "conv_1x1s2": lambda m: check_conv(m, k=1, p=0, s=2, bias=False),
     "conv_group": lambda m: check_conv(m, C=8, Co=16, k=5, p=2, grp=2),
+    # dilation>1 forces the explicit im2col path (implicit GEMM is s1/d1
+    # only — layers_gpu.cpp conv policy); reference supports it via
+    # conv_param.dilation (caffe.proto:518)
+    "conv_3x3_dilated": lambda m: check_conv(m, N=2, C=4, H=15, W=15, Co=8,
+                                             k=3, s=1, p=2, d=2),
     "conv_5x5_odd": lambda m: check_conv(m, N=3, C=5, H=11, W=17, Co=7, k=5,
                                          s=3, p=2),
     # implicit-im2col eligible shapes (s1/d1, OW >= 24 on the GPU path):
