@@ -215,6 +215,14 @@ class Net:
     def params(self):
         return {i: self.param_info(i) for i in range(self.num_params())}
 
+    def save_weights(self, path):
+        """NetParameter binaryproto with every layer blob (incl. BN stats)
+        — net.cpp ToProto / tools/caffe.cpp -weights surface."""
+        _ck(_lib.caffe_net_save_weights(self._h, path.encode()))
+
+    def load_weights(self, path):
+        _ck(_lib.caffe_net_load_weights(self._h, path.encode()))
+
 
 class Solver:
     def __init__(self, path=None, text=None, batch_override=0):

@@ -285,6 +285,129 @@ layer {{
     assert relerr(net.blob("in0", diff=True), din_ref) < TOL
 
 
+def check_dropout(mode, N=2, C=4, H=7, W=7, ratio=0.4, seed=4242):
+    """Dropout TRAIN mask is the engine's counter RNG — restated here from
+    core.hpp h_splitmix64/h_u01 and the layer key (layers_cpu.cpp
+    Forward_cpu: splitmix64(seed ^ 0xD0D0 ^ rank<<40 ^ data_iter)); CPU and
+    GPU must both reproduce it bit-exactly (reference dropout semantics:
+    dropout_layer.cpp — mask*scale with scale=1/(1-ratio))."""
+    import caffe_amd as ca
+
+    M64 = (1 << 64) - 1
+
+    def smix(v):
+        v = (v + 0x9E3779B97F4A7C15) & M64
+        v = ((v ^ (v >> 30)) * 0xBF58476D1CE4E5B9) & M64
+        v = ((v ^ (v >> 27)) * 0x94D049BB133111EB) & M64
+        return v ^ (v >> 31)
+
+    ca.set_mode(mode)
+    ca.set_random_seed(seed)
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32) + 3.0
+    w = np.eye(C, dtype=np.float32).reshape(C, C, 1, 1).copy()
+    body = f"""layer {{
+  name: "c1"
+  type: "Convolution"
+  bottom: "in0"
+  top: "mid"
+  convolution_param {{ num_output: {C} kernel_size: 1 bias_term: false }}
+}}
+layer {{
+  name: "drop"
+  type: "Dropout"
+  bottom: "mid"
+  top: "out"
+  dropout_param {{ dropout_ratio: {ratio} }}
+}}"""
+    dy = rng.standard_normal(x.shape).astype(np.float32)
+    net, y = run_layer(mode, [x.shape], body, [x], params=[w], top_diff=dy)
+    n = x.size
+    # x is offset by +3 so a zero output can only mean "dropped" — recover
+    # the mask the layer actually used, then pin it bit-exactly to the
+    # restated counter RNG for SOME data_iter (earlier in-process tests may
+    # have stepped a solver, so the exact iter is not known here)
+    y_arr = np.asarray(y).reshape(x.shape)
+    mask_obs = (y_arr != 0).astype(np.float32)
+
+    def mask_for(data_iter):
+        key = smix(seed ^ 0xD0D0 ^ data_iter)
+        return np.array([
+            np.float32((smix(key ^ i) >> 11)
+                       * (1.0 / 9007199254740992.0))
+            >= np.float32(ratio) for i in range(n)],
+            dtype=np.float32).reshape(x.shape)
+
+    matched = any(np.array_equal(mask_for(di), mask_obs)
+                  for di in range(256))
+    assert matched, "dropout mask does not come from the engine counter RNG"
+    scale = np.float32(1.0 / (1.0 - np.float32(ratio)))
+    assert relerr(y_arr, x * mask_obs * scale) < TOL, "dropout fwd"
+    kept = float(mask_obs.mean())
+    assert 0.4 < kept < 0.8, kept  # ratio=0.4 → ~60% kept
+    assert relerr(net.blob("mid", diff=True), dy * mask_obs * scale) < TOL
+    ca.set_random_seed(1371)  # restore the default for later checks
+
+
+def check_bn_global(mode, N=4, C=6, H=5, W=5):
+    """TEST-phase BN (use-global-stats inference path, batch_norm_layer
+    .cpp:154 / layers_gpu.cpp Forward_gpu TEST branch) THROUGH a weights
+    round-trip: TRAIN forward populates the moving averages (iter<=1 copy),
+    SaveWeights/LoadWeights carries all 5 BN blobs into a TEST-phase net,
+    whose output must match the closed-form (x-mean)/sqrt(var+eps)*sc+bi."""
+    import tempfile
+
+    from engine_util import input_net, net_from_text
+
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    sc = (rng.standard_normal(C) + 1.5).astype(np.float32)
+    bi = rng.standard_normal(C).astype(np.float32)
+    eps = 1e-3
+    body = f"""layer {{
+  name: "bn"
+  type: "BatchNorm"
+  bottom: "in0"
+  top: "out"
+  batch_norm_param {{ eps: {eps} scale_bias: true }}
+}}"""
+    net, _ = run_layer(mode, [x.shape], body, [x], params=[sc, bi])
+    f = tempfile.NamedTemporaryFile(suffix=".caffemodel", delete=False)
+    f.close()
+    net.save_weights(f.name)
+    net2 = net_from_text(input_net([x.shape], body), phase=1)
+    net2.load_weights(f.name)
+    net2.set_blob("in0", x)
+    net2.forward()
+    y = net2.blob("out")
+    mean = x.mean(axis=(0, 2, 3))
+    var = x.var(axis=(0, 2, 3))  # biased, as the layer computes it
+    y_ref = ((x - mean[:, None, None]) / np.sqrt(var + eps)[:, None, None]
+             * sc[:, None, None] + bi[:, None, None])
+    assert relerr(y, y_ref) < TOL, f"bn inference {relerr(y, y_ref)}"
+
+
+def check_accuracy(mode, N=12, C=10, top_k=3):
+    """Accuracy layer (CPU-resident in both modes, as in the reference —
+    accuracy_layer.cpp has no .cu) vs the oracle's restatement."""
+    pred = rng.standard_normal((N, C)).astype(np.float32)
+    label = rng.integers(0, C, N).astype(np.float32)
+    body = f"""layer {{
+  name: "acc"
+  type: "Accuracy"
+  bottom: "in0"
+  bottom: "in1"
+  top: "out"
+  accuracy_param {{ top_k: {top_k} }}
+}}"""
+    _, y = run_layer(mode, [(N, C), (N,)], body, [pred, label])
+    ref = orc.accuracy(pred, label, N, C, 1, top_k)
+    assert abs(float(np.asarray(y).ravel()[0]) - ref) < 1e-6
+    # top_k=1 via a fresh net on the same data
+    body1 = body.replace(f"top_k: {top_k}", "top_k: 1")
+    _, y1 = run_layer(mode, [(N, C), (N,)], body1, [pred, label])
+    assert abs(float(np.asarray(y1).ravel()[0])
+               - orc.accuracy(pred, label, N, C, 1, 1)) < 1e-6
+
+
 ALL_CHECKS = {
     "conv_3x3": lambda m: check_conv(m),
     "conv_7x7s2": lambda m: check_conv(m, C=3, H=19, W=19, Co=8, k=7, s=2,
@@ -324,4 +447,7 @@ ALL_CHECKS = {
     "lrn": check_lrn,
     "softmaxloss": check_softmaxloss,
     "eltwise_concat": check_eltwise_concat,
+    "dropout": check_dropout,
+    "bn_global_stats": check_bn_global,
+    "accuracy": check_accuracy,
 }
