@@ -493,6 +493,10 @@ void BatchNormLayer::LayerSetUp(const std::vector<Blob*>& bottom,
   eps_ = std::max(eps_, 1e-5f);  // batch_norm_layer.cpp:25
   scale_bias_ = bp && (bp->boolean("scale_bias", false) ||
                        bp->has("scale_filler") || bp->has("bias_filler"));
+  // parsed for spec completeness but, exactly like the reference, never
+  // consulted: NVCaffe-0.16 assigns use_global_stats_ (batch_norm_layer
+  // .cpp:18) and then dispatches purely on phase (== TEST at :154); we
+  // mirror that so a prototxt setting it behaves identically
   use_global_ = bp && bp->boolean("use_global_stats", false);
   C_ = bottom[0]->channels();
   if (blobs_.empty()) {

@@ -85,6 +85,35 @@ def _ck(rc):
     assert rc == 0
 
 
+def test_cli_binary_train_then_test():
+    """The full `caffe train` → snapshot → `caffe test -weights` chain
+    through the BINARY (tools/caffe.cpp:213/:291 surfaces): train writes
+    lenet_iter_4.caffemodel, test loads it on the TEST phase net and
+    prints averaged scores (accuracy + loss tops)."""
+    with tempfile.TemporaryDirectory() as tmp:
+        solver = make_lenet_solver(tmp)
+        net = os.path.join(REPO, "models", "generated",
+                           "lenet_train_val.prototxt")
+        env = dict(os.environ, CAFFE_SYN_SHAPE="1x28x28x10")
+        out = subprocess.run(
+            [CAFFE, "train", f"-solver={solver}"],
+            capture_output=True, text=True, timeout=600, env=env)
+        assert out.returncode == 0, out.stderr
+        assert "Optimization Done." in out.stderr
+        model = os.path.join(tmp, "lenet_iter_4.caffemodel")
+        assert os.path.exists(model), out.stderr
+        out = subprocess.run(
+            [CAFFE, "test", f"-model={net}", f"-weights={model}",
+             "-iterations=2"],
+            capture_output=True, text=True, timeout=600, env=env)
+        assert out.returncode == 0, out.stderr
+        scores = dict(line.split(" = ") for line in
+                      out.stdout.strip().splitlines() if " = " in line)
+        assert "accuracy" in scores and "loss" in scores, out.stdout
+        assert 0.0 <= float(scores["accuracy"]) <= 1.0
+        assert float(scores["loss"]) > 0.0
+
+
 def test_cli_binary_time_and_device_query():
     # `caffe time` on the LeNet model, CPU
     with tempfile.TemporaryDirectory() as tmp:
