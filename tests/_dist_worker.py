@@ -13,6 +13,7 @@ sys.path.insert(0, os.path.join(REPO, "caffe-mpi.github.io_amd"))
 import caffe_amd as ca  # noqa: E402
 
 SOLVER_TEXT = """
+snapshot_prefix: "SNAPPFX"
 base_lr: 0.05
 lr_policy: "fixed"
 momentum: 0.9
@@ -62,13 +63,19 @@ def main():
     ap.add_argument("--rank-data", choices=["same", "shard", "combined"],
                     default="same")
     ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--restore-at", type=int, default=-1,
+                    help="snapshot at this iter, rebuild the solver, "
+                         "restore, re-attach the comm, continue")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
 
     ca.set_mode("cpu")
-    solver = ca.Solver(text=SOLVER_TEXT % (args.batch, args.batch))
+    tmp = tempfile.mkdtemp()
+    text = (SOLVER_TEXT % (args.batch, args.batch)).replace(
+        "SNAPPFX", os.path.join(tmp, f"r{rank}"))
+    solver = ca.Solver(text=text)
     net = solver.net
 
     if world > 1:
@@ -91,6 +98,21 @@ def main():
 
     pool_b = 16
     for it in range(args.iters):
+        if it == args.restore_at:
+            # mid-training restart: snapshot, tear the solver down, build a
+            # fresh one, Restore (params + momentum history + iter), and
+            # re-attach the collective — the 8-GPU resume path
+            # (solver.cpp:542 Snapshot / :604 Restore semantics)
+            assert ca._lib.caffe_solver_snapshot(solver._h) == 0
+            state = os.path.join(tmp, f"r{rank}_iter_{solver.iter}"
+                                      ".solverstate")
+            assert os.path.exists(state), state
+            solver = ca.Solver(text=text)
+            net = solver.net
+            assert ca._lib.caffe_solver_restore(solver._h,
+                                                state.encode()) == 0
+            if world > 1:
+                solver.set_allreduce_callback(reduce_fn, world)
         rng = np.random.default_rng(1000 + it)
         data = rng.standard_normal((pool_b, 3, 8, 8)).astype(np.float32)
         labels = rng.integers(0, 5, pool_b).astype(np.float32)

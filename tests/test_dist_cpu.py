@@ -67,6 +67,21 @@ def test_two_ranks_same_data_match_single():
     assert np.allclose(p1, p2, rtol=1e-5, atol=1e-6), (p1 - p2)
 
 
+def test_two_ranks_restore_mid_training_matches_uninterrupted():
+    # rank snapshots at iter 2, rebuilds the solver, restores (params +
+    # momentum history + iter), re-attaches the gloo comm, and continues —
+    # must land on exactly the same params as an uninterrupted run
+    # (reference solver.cpp:542 Snapshot / :604 Restore under P2PSync)
+    plain = run_dist(2, ["--iters", "4", "--rank-data", "shard",
+                         "--batch", "8"])
+    resumed = run_dist(2, ["--iters", "4", "--rank-data", "shard",
+                           "--batch", "8", "--restore-at", "2"])
+    p_plain = parse_params(plain[0])
+    p_resumed = parse_params(resumed[0])
+    assert np.allclose(p_plain, p_resumed, rtol=1e-6, atol=1e-7), \
+        np.abs(p_plain - p_resumed).max()
+
+
 def test_two_ranks_sharded_match_combined_batch():
     # rank r trains on shard r; equivalent single-rank run feeds the
     # concatenated batch (grad averaging == big-batch gradient)
