@@ -152,6 +152,24 @@ void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
                      (const f4*)x, (const f4*)dy, n4, slope, (f4*)dx);
 }
 
+__global__ void k_axpy(const f4* __restrict__ x, long n4, float a,
+                       f4* __restrict__ y) {
+  VEC_GRID(i, n4) {
+    f4 v = y[i];
+    const f4 xv = x[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] += a * xv[j];
+    y[i] = v;
+  }
+}
+// y += a*x over the (64B-padded) gradient arena — iter_size accumulation
+void axpy(hipStream_t s, long n, float a, const float* x, float* y) {
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
+  const long n4 = (n + 3) / 4;
+  hipLaunchKernelGGL(k_axpy, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
+                     (const f4*)x, n4, a, (f4*)y);
+}
+
 // ------------------------------------------------------------ pooling
 __global__ void k_pool_max_fwd(const float* __restrict__ x, int N, int C,
                                int H, int W, int kh, int kw, int ph, int pw,

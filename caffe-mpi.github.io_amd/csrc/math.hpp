@@ -93,6 +93,9 @@ void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
                     int W, int kh, int kw, int ph, int pw, int sh, int sw,
                     int dh, int dw, int OH, int OW, long Spad, float* dx);
 
+// y += a*x (iter_size diff accumulation over the padded arena)
+void axpy(hipStream_t s, long n, float a, const float* x, float* y);
+
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y);
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
               float slope, float* dx);

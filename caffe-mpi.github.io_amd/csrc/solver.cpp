@@ -46,6 +46,39 @@ Solver::~Solver() {
   if (history_)
     Engine::get().dalloc.release(history_,
                                  sizeof(float) * net_->learnable_count());
+  if (acc_)
+    Engine::get().dalloc.release(acc_,
+                                 sizeof(float) * net_->learnable_count());
+}
+
+// acc += diff (merge_back=false, after each non-final sub-pass) or
+// diff += acc then zero acc (merge_back=true, after the final sub-pass)
+void Solver::accumulate_diffs(bool merge_back) {
+  Engine& E = Engine::get();
+  const long total = net_->learnable_count();
+  if (E.mode == Mode::GPU) {
+    if (!acc_) {
+      acc_ = (float*)E.dalloc.alloc(sizeof(float) * total);
+      HIP_CHECK(hipMemsetAsync(acc_, 0, sizeof(float) * total, E.stream));
+    }
+    if (!merge_back) {
+      gpu::axpy(E.stream, total, 1.f, net_->diff_arena(), acc_);
+    } else {
+      gpu::axpy(E.stream, total, 1.f, acc_, net_->diff_arena());
+      HIP_CHECK(hipMemsetAsync(acc_, 0, sizeof(float) * total, E.stream));
+    }
+  } else {
+    if (host_acc_.empty()) host_acc_.assign(total, 0.f);
+    for (auto& p : net_->learnable_params()) {
+      float* diff = p.blob->mutable_cpu_diff();
+      float* acc = host_acc_.data() + p.offset;
+      if (!merge_back)
+        cpu::axpy(p.count, 1.f, diff, acc);
+      else
+        cpu::axpy(p.count, 1.f, acc, diff);
+    }
+    if (merge_back) std::fill(host_acc_.begin(), host_acc_.end(), 0.f);
+  }
 }
 
 float Solver::GetLearningRate() const {
@@ -221,6 +254,10 @@ void Solver::Step(int iters) {
   const long test_interval = param_->inum("test_interval", 0);
   const long test_iter = param_->inum("test_iter", 1);
   const long snap_interval = param_->inum("snapshot", 0);
+  // gradient accumulation: iter_size fwd/bwd passes per update, diffs
+  // accumulate in the arena, one reduce+update scaled by 1/iter_size
+  // (reference solver.cpp:279-297 Step loop + SGDSolver::Normalize)
+  const long iter_size = std::max<long>(1, param_->inum("iter_size", 1));
   // this fork always runs a 1-iter test at iter 0 regardless of
   // test_initialization (reference solver.cpp:243-248)
   if (iter_ == 0 && test_interval > 0 && test_net()) TestAll(1);
@@ -238,15 +275,33 @@ void Solver::Step(int iters) {
     if (test_interval > 0 && iter_ > 0 && iter_ % test_interval == 0 &&
         test_net())
       TestAll(test_iter);
-    E.data_iter = (uint64_t)iter_;
     cur_lr_ = GetLearningRate();
     cur_mom_ = GetMomentum();
-    grad_scale_ = comm_ && comm_->world() > 1
-                      ? 1.f / (float)comm_->world()
-                      : 1.f;
+    const float world =
+        comm_ && comm_->world() > 1 ? (float)comm_->world() : 1.f;
+    grad_scale_ = 1.f / (world * (float)iter_size);
     reducer_.start_iteration();
-    net_->Forward();
-    net_->Backward(&reducer_);
+    if (iter_size == 1) {
+      E.data_iter = (uint64_t)iter_;
+      net_->Forward();
+      net_->Backward(&reducer_);
+    } else {
+      // layer backwards overwrite their param diffs, so the accumulation
+      // runs hookless with a side buffer; the reducer is driven once at
+      // the end (overlap is moot — sub-passes serialize by construction)
+      for (long sub = 0; sub < iter_size; ++sub) {
+        // sub-iteration granularity: each pass sees a fresh synthetic batch
+        E.data_iter = (uint64_t)(iter_ * iter_size + sub);
+        net_->Forward();
+        net_->Backward(nullptr);
+        if (sub + 1 < iter_size) accumulate_diffs(false);
+      }
+      accumulate_diffs(true);
+      E.sync();  // diffs final on E.stream before comm_stream consumes them
+      const int nparams = (int)net_->learnable_params().size();
+      for (int k = 0; k < nparams; ++k) reducer_.param_ready(k, nullptr);
+      reducer_.iteration_end(nullptr);
+    }
     ++iter_;
     if (snap_interval > 0 && iter_ % snap_interval == 0) Snapshot();
     if (display > 0 && iter_ % display == 0) {

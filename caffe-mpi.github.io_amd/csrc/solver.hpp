@@ -106,6 +106,12 @@ class Solver {
   std::vector<float> seg_lr_mult_, seg_decay_mult_;
   float table_lr_ = -1.f, table_decay_ = -1.f;
   void ensure_seg_table();
+  // iter_size accumulation buffer (diff-arena layout; only allocated when
+  // iter_size > 1 — layer backwards overwrite their param diffs, so cross-
+  // pass accumulation happens here instead of in every wgrad kernel)
+  float* acc_ = nullptr;
+  std::vector<float> host_acc_;
+  void accumulate_diffs(bool merge_back);
   // perf-report bookkeeping
   double perf_seconds_ = 0.0;
   long perf_iters_ = 0;
