@@ -271,7 +271,17 @@ void Solver::Step(int iters) {
     E.sync();
     tstep0 = clock::now();
   }
+  int done_iters = 0;
   for (int i = 0; i < iters; ++i) {
+    if (action_fn_) {
+      const SolverAction a = action_fn_();
+      if (a == SolverAction::SNAPSHOT) {
+        Snapshot();
+      } else if (a == SolverAction::STOP) {
+        early_exit_ = true;
+        break;
+      }
+    }
     if (test_interval > 0 && iter_ > 0 && iter_ % test_interval == 0 &&
         test_net())
       TestAll(test_iter);
@@ -303,6 +313,7 @@ void Solver::Step(int iters) {
       reducer_.iteration_end(nullptr);
     }
     ++iter_;
+    ++done_iters;
     if (snap_interval > 0 && iter_ % snap_interval == 0) Snapshot();
     if (display > 0 && iter_ % display == 0) {
       const float l = net_->loss();
@@ -314,9 +325,9 @@ void Solver::Step(int iters) {
     E.sync();
     perf_seconds_ +=
         std::chrono::duration<double>(clock::now() - tstep0).count();
-    perf_iters_ += iters;
+    perf_iters_ += done_iters;
   } else {
-    skipped_iters_ += iters;
+    skipped_iters_ += done_iters;
   }
 }
 

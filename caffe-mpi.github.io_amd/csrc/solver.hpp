@@ -34,6 +34,12 @@ std::unique_ptr<Comm> make_callback_comm(HostAllreduceFn fn, void* ud,
 
 class Solver;
 
+// reference SolverAction (include/caffe/solver.hpp + util/signal_handler):
+// polled once per iteration inside Step — NONE continue, STOP break out
+// (requested_early_exit_), SNAPSHOT snapshot and continue
+enum class SolverAction { NONE = 0, STOP = 1, SNAPSHOT = 2 };
+using ActionRequestFn = SolverAction (*)();
+
 class Reducer : public ReduceHook {
  public:
   Reducer(Solver* s) : solver_(s) {}
@@ -74,6 +80,9 @@ class Solver {
   float last_loss() { return net_->loss(); }
   const PMsgPtr& param() const { return param_; }
 
+  void set_action_request(ActionRequestFn fn) { action_fn_ = fn; }
+  bool early_exit() const { return early_exit_; }
+
   void set_comm(std::unique_ptr<Comm> c) { comm_ = std::move(c); }
   Comm* comm() { return comm_.get(); }
   float* history() { return history_; }
@@ -92,6 +101,8 @@ class Solver {
   std::unique_ptr<Net> test_net_;
   std::unique_ptr<Comm> comm_;
   Reducer reducer_{this};
+  ActionRequestFn action_fn_ = nullptr;
+  bool early_exit_ = false;
   long iter_ = 0;
   mutable int current_step_ = 0;
   float* history_ = nullptr;  // device arena, diff-arena layout

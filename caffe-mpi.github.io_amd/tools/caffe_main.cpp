@@ -7,6 +7,7 @@
 // training runs one process per GPU through bench.py / the C ABI
 //  — the reference's in-process thread-per-GPU P2PManager is
 // replaced by that launcher model.
+#include <csignal>
 #include <cstdio>
 #include <cstring>
 #include <map>
@@ -15,6 +16,57 @@
 #include "../csrc/solver.hpp"
 
 using namespace camd;
+
+// ---- signal → SolverAction plumbing (reference util/signal_handler.cpp:
+// handlers only set flags; Solver polls via the action-request callback
+// once per iteration).  Defaults: -sigint_effect=stop -sighup_effect=snapshot
+// (tools/caffe.cpp:31-36).
+static volatile sig_atomic_t g_got_sigint = 0;
+static volatile sig_atomic_t g_got_sighup = 0;
+static SolverAction g_sigint_action = SolverAction::STOP;
+static SolverAction g_sighup_action = SolverAction::SNAPSHOT;
+
+static void handle_signal(int sig) {
+  if (sig == SIGINT) g_got_sigint = 1;
+  else if (sig == SIGHUP) g_got_sighup = 1;
+}
+
+static SolverAction parse_effect(const std::string& s, SolverAction dflt) {
+  if (s == "stop") return SolverAction::STOP;
+  if (s == "snapshot") return SolverAction::SNAPSHOT;
+  if (s == "none") return SolverAction::NONE;
+  if (!s.empty())
+    fprintf(stderr, "unknown signal effect '%s' (stop|snapshot|none)\n",
+            s.c_str());
+  return dflt;
+}
+
+static SolverAction action_request() {
+  if (g_got_sigint) {
+    g_got_sigint = 0;
+    return g_sigint_action;
+  }
+  if (g_got_sighup) {
+    g_got_sighup = 0;
+    return g_sighup_action;
+  }
+  return SolverAction::NONE;
+}
+
+static void install_signal_handlers(
+    const std::map<std::string, std::string>& flags) {
+  auto it = flags.find("sigint_effect");
+  if (it != flags.end())
+    g_sigint_action = parse_effect(it->second, SolverAction::STOP);
+  it = flags.find("sighup_effect");
+  if (it != flags.end())
+    g_sighup_action = parse_effect(it->second, SolverAction::SNAPSHOT);
+  struct sigaction sa;
+  memset(&sa, 0, sizeof(sa));
+  sa.sa_handler = handle_signal;
+  sigaction(SIGINT, &sa, nullptr);
+  sigaction(SIGHUP, &sa, nullptr);
+}
 
 static std::map<std::string, std::string> parse_flags(int argc, char** argv,
                                                       int start) {
@@ -84,7 +136,9 @@ int main(int argc, char** argv) {
     if (cmd == "train") {
       CHECK_(flags.count("solver")) << "train needs -solver";
       setup_device(flags);
+      install_signal_handlers(flags);
       Solver solver(parse_prototxt_file(flags["solver"]));
+      solver.set_action_request(&action_request);
       if (flags.count("snapshot") && !flags["snapshot"].empty())
         solver.Restore(flags["snapshot"]);
       else if (flags.count("weights") && !flags["weights"].empty())
@@ -98,6 +152,8 @@ int main(int argc, char** argv) {
       solver.print_perf_report();
       if (solver.param()->boolean("snapshot_after_train", true))
         solver.Snapshot();
+      if (solver.early_exit())
+        fprintf(stderr, "Optimization stopped early.\n");
       fprintf(stderr, "Optimization Done.\n");
       return 0;
     }
