@@ -276,7 +276,7 @@ void Solver::Step(int iters) {
     if (action_fn_) {
       const SolverAction a = action_fn_();
       if (a == SolverAction::SNAPSHOT) {
-        Snapshot();
+        if (snapshot_enabled_) Snapshot();
       } else if (a == SolverAction::STOP) {
         early_exit_ = true;
         break;
@@ -314,7 +314,9 @@ void Solver::Step(int iters) {
     }
     ++iter_;
     ++done_iters;
-    if (snap_interval > 0 && iter_ % snap_interval == 0) Snapshot();
+    if (snap_interval > 0 && iter_ % snap_interval == 0 &&
+        snapshot_enabled_)
+      Snapshot();
     if (display > 0 && iter_ % display == 0) {
       const float l = net_->loss();
       fprintf(stderr, "[caffe_amd] Iteration %ld, loss = %g, lr = %g\n",
@@ -334,6 +336,17 @@ void Solver::Step(int iters) {
 std::shared_ptr<Solver> create_solver_from_file(const std::string& path,
                                                 int batch_override) {
   return std::make_shared<Solver>(parse_prototxt_file(path), batch_override);
+}
+
+double Solver::perf_img_per_sec() const {
+  if (perf_iters_ == 0 || perf_seconds_ <= 0) return 0.0;
+  long batch = 0;
+  for (auto& l : net_->layers())
+    if (auto* d = dynamic_cast<const DataLayer*>(l.get())) {
+      batch = d->batch_;
+      break;
+    }
+  return perf_iters_ / perf_seconds_ * batch;
 }
 
 void Solver::print_perf_report() const {
@@ -366,6 +379,9 @@ Net* Solver::test_net() {
 }
 
 void Solver::TestAll(long iters) {
+  // reference: the test pass runs on the root solver only (solver.cpp:439;
+  // cross-rank SharedScores aggregation is multi-node machinery)
+  if (Engine::get().rank != 0) return;
   Net* tn = test_net();
   if (!tn) return;
   // average every loss-weighted / Accuracy top over the test iterations

@@ -82,6 +82,14 @@ class Solver {
 
   void set_action_request(ActionRequestFn fn) { action_fn_ = fn; }
   bool early_exit() const { return early_exit_; }
+  // reference: only the root solver snapshots (rank 0's moving averages /
+  // params are the ones persisted) — non-root CLI ranks disable the
+  // snapshot-interval and signal-triggered snapshots; direct Snapshot()
+  // calls (C ABI) are unaffected
+  void set_snapshot_enabled(bool e) { snapshot_enabled_ = e; }
+  bool snapshot_enabled() const { return snapshot_enabled_; }
+  // measured throughput of the timed Step calls (img/sec), 0 before any
+  double perf_img_per_sec() const;
 
   void set_comm(std::unique_ptr<Comm> c) { comm_ = std::move(c); }
   Comm* comm() { return comm_.get(); }
@@ -103,6 +111,7 @@ class Solver {
   Reducer reducer_{this};
   ActionRequestFn action_fn_ = nullptr;
   bool early_exit_ = false;
+  bool snapshot_enabled_ = true;
   long iter_ = 0;
   mutable int current_step_ = 0;
   float* history_ = nullptr;  // device arena, diff-arena layout

@@ -10,8 +10,13 @@
 #include <csignal>
 #include <cstdio>
 #include <cstring>
+#include <sys/stat.h>
+#include <sys/wait.h>
+#include <unistd.h>
+
 #include <map>
 #include <string>
+#include <vector>
 
 #include "../csrc/solver.hpp"
 
@@ -102,6 +107,188 @@ namespace camd {
 extern int g_syn_shape[3];
 }
 
+// ---- multi-GPU train (`-gpu=all` / `-gpu=0,1,...`): the reference's
+// in-process thread-per-GPU P2PManager (parallel.cpp) is replaced by the
+// MI355X-native one-process-per-GPU model — the parent forks one worker per
+// device BEFORE any HIP call (HIP init + fork is unsafe), the RCCL unique id
+// rendezvous is a file in a private tmp dir (the reference used MPI_Bcast;
+// single-node needs neither MPI nor a network), and the parent prints the
+// reference's overall line (parallel.cpp:85) from per-rank perf files.
+
+// device count probed in a throwaway child so the parent never inits HIP
+static int device_count_scout() {
+  int fds[2];
+  if (pipe(fds) != 0) return 0;
+  const pid_t pid = fork();
+  if (pid == 0) {
+    int n = 0;
+    if (hipGetDeviceCount(&n) != hipSuccess) n = 0;
+    (void)!write(fds[1], &n, sizeof(n));
+    _exit(0);
+  }
+  close(fds[1]);
+  int n = 0;
+  if (read(fds[0], &n, sizeof(n)) != sizeof(n)) n = 0;
+  close(fds[0]);
+  waitpid(pid, nullptr, 0);
+  return n;
+}
+
+// total train-phase Data batch from the net message — the reference splits
+// the prototxt batch across solvers (parallel.cpp:284-348), remainder to
+// the first ranks
+static long train_batch_size(const PMsgPtr& sp) {
+  PMsgPtr net_msg;
+  if (sp->has("net"))
+    net_msg = parse_prototxt_file(sp->str("net"));
+  else
+    net_msg = sp->sub("net_param");
+  if (!net_msg) return 0;
+  for (auto& lm : net_msg->subs("layer")) {
+    bool test_only = false;
+    for (auto& inc : lm->subs("include"))
+      if (inc->str("phase") == "TEST") test_only = true;
+    if (test_only || lm->str("type") != "Data") continue;
+    if (auto dp = lm->sub("data_param")) return dp->inum("batch_size", 0);
+    return 0;
+  }
+  return 0;
+}
+
+static int run_train_rank(std::map<std::string, std::string> flags, int dev,
+                          int rank, int world,
+                          const std::string& rdv_dir) {
+  Engine& E = Engine::get();
+  if (dev >= 0)
+    E.set_mode_gpu(dev);
+  else
+    E.mode = Mode::CPU;
+  E.rank = rank;  // fillers + synthetic data use seed+rank
+  install_signal_handlers(flags);
+  auto sp = parse_prototxt_file(flags["solver"]);
+  int batch_override = 0;
+  if (world > 1) {
+    const long total = train_batch_size(sp);
+    if (total > 0)
+      batch_override =
+          (int)(total / world + (rank < total % world ? 1 : 0));
+  }
+  Solver solver(sp, batch_override);
+  solver.set_action_request(&action_request);
+  if (rank != 0) solver.set_snapshot_enabled(false);
+  if (world > 1) {
+    uint8_t uid[128];
+    const std::string uid_path = rdv_dir + "/rccl_uid";
+    if (rank == 0) {
+      rccl_unique_id(uid);
+      const std::string tmp = uid_path + ".tmp";
+      FILE* f = fopen(tmp.c_str(), "wb");
+      CHECK_(f) << "cannot write " << tmp;
+      fwrite(uid, 1, sizeof(uid), f);
+      fclose(f);
+      CHECK_EQ_(rename(tmp.c_str(), uid_path.c_str()), 0);
+    } else {
+      bool got = false;
+      for (int i = 0; i < 6000 && !got; ++i) {  // up to 60 s
+        if (FILE* f = fopen(uid_path.c_str(), "rb")) {
+          got = fread(uid, 1, sizeof(uid), f) == sizeof(uid);
+          fclose(f);
+        }
+        if (!got) usleep(10000);
+      }
+      CHECK_(got) << "rank " << rank << ": no RCCL uid rendezvous";
+    }
+    solver.set_comm(make_rccl_comm(rank, world, uid));
+    solver.bcast_weights();  // initial weight bcast, parallel.cpp:208-227
+  }
+  if (flags.count("snapshot") && !flags["snapshot"].empty())
+    solver.Restore(flags["snapshot"]);
+  else if (flags.count("weights") && !flags["weights"].empty())
+    solver.LoadWeights(flags["weights"]);
+  const long max_iter = solver.param()->inum("max_iter", 0);
+  long todo = max_iter - solver.iter();
+  if (flags.count("iterations")) todo = atol(flags["iterations"].c_str());
+  CHECK_GT_(todo, 0);
+  // perf accounting is Step-call-granular (first call untimed): warm up
+  // with 2 iters then time the rest — the reference's skip-iters-0..1
+  // rule (solver.cpp:299)
+  const long warm = std::min<long>(2, todo);
+  solver.Step((int)warm);
+  if (todo > warm) solver.Step((int)(todo - warm));
+  solver.print_perf_report();
+  if (!rdv_dir.empty()) {
+    const std::string pf = rdv_dir + "/perf_" + std::to_string(rank);
+    if (FILE* f = fopen(pf.c_str(), "w")) {
+      fprintf(f, "%.17g\n", solver.perf_img_per_sec());
+      fclose(f);
+    }
+  }
+  if (rank == 0 &&
+      solver.param()->boolean("snapshot_after_train", true))
+    solver.Snapshot();
+  if (solver.early_exit())
+    fprintf(stderr, "Optimization stopped early.\n");
+  if (rank == 0) fprintf(stderr, "Optimization Done.\n");
+  return 0;
+}
+
+static int run_multi_gpu_train(
+    const std::map<std::string, std::string>& flags,
+    const std::vector<int>& devs) {
+  char tmpl[] = "/tmp/caffe_amd_rdv_XXXXXX";
+  CHECK_(mkdtemp(tmpl)) << "mkdtemp failed";
+  const std::string rdv_dir = tmpl;
+  // children own SIGINT/SIGHUP (delivered group-wide by the terminal);
+  // the parent only waits
+  signal(SIGINT, SIG_IGN);
+  signal(SIGHUP, SIG_IGN);
+  const int world = (int)devs.size();
+  std::vector<pid_t> pids;
+  for (int rank = 0; rank < world; ++rank) {
+    const pid_t pid = fork();
+    CHECK_GE_(pid, 0) << "fork failed";
+    if (pid == 0) {
+      int rc = 1;
+      try {
+        rc = run_train_rank(flags, devs[rank], rank, world, rdv_dir);
+      } catch (const std::exception& e) {
+        fprintf(stderr, "FATAL (rank %d): %s\n", rank, e.what());
+      }
+      _exit(rc);
+    }
+    pids.push_back(pid);
+  }
+  int rc = 0;
+  for (pid_t pid : pids) {
+    int status = 0;
+    waitpid(pid, &status, 0);
+    if (!WIFEXITED(status) || WEXITSTATUS(status) != 0) rc = 1;
+  }
+  if (rc == 0) {
+    double total = 0;
+    bool any = false;
+    for (int rank = 0; rank < world; ++rank) {
+      const std::string pf = rdv_dir + "/perf_" + std::to_string(rank);
+      if (FILE* f = fopen(pf.c_str(), "r")) {
+        double v = 0;
+        if (fscanf(f, "%lg", &v) == 1 && v > 0) {
+          total += v;
+          any = true;
+        }
+        fclose(f);
+      }
+    }
+    if (any)
+      fprintf(stderr, "Overall multi-GPU performance: %.1f img/sec\n",
+              total);
+  }
+  for (int rank = 0; rank < world; ++rank)
+    unlink((rdv_dir + "/perf_" + std::to_string(rank)).c_str());
+  unlink((rdv_dir + "/rccl_uid").c_str());
+  rmdir(rdv_dir.c_str());
+  return rc;
+}
+
 int main(int argc, char** argv) {
   // synthetic data shape for dataset-less runs: CAFFE_SYN_SHAPE=CxHxW[xK]
   if (const char* ss = getenv("CAFFE_SYN_SHAPE")) {
@@ -135,27 +322,30 @@ int main(int argc, char** argv) {
     }
     if (cmd == "train") {
       CHECK_(flags.count("solver")) << "train needs -solver";
-      setup_device(flags);
-      install_signal_handlers(flags);
-      Solver solver(parse_prototxt_file(flags["solver"]));
-      solver.set_action_request(&action_request);
-      if (flags.count("snapshot") && !flags["snapshot"].empty())
-        solver.Restore(flags["snapshot"]);
-      else if (flags.count("weights") && !flags["weights"].empty())
-        solver.LoadWeights(flags["weights"]);
-      const long max_iter = solver.param()->inum("max_iter", 0);
-      long todo = max_iter - solver.iter();
-      if (flags.count("iterations"))
-        todo = atol(flags["iterations"].c_str());
-      CHECK_GT_(todo, 0);
-      solver.Step((int)todo);
-      solver.print_perf_report();
-      if (solver.param()->boolean("snapshot_after_train", true))
-        solver.Snapshot();
-      if (solver.early_exit())
-        fprintf(stderr, "Optimization stopped early.\n");
-      fprintf(stderr, "Optimization Done.\n");
-      return 0;
+      const std::string gpuflag =
+          flags.count("gpu") ? flags["gpu"] : std::string();
+      if (gpuflag == "all" || gpuflag.find(',') != std::string::npos) {
+        std::vector<int> devs;
+        if (gpuflag == "all") {
+          const int n = device_count_scout();
+          CHECK_GT_(n, 0) << "-gpu=all but no HIP device visible";
+          for (int i = 0; i < n; ++i) devs.push_back(i);
+        } else {
+          size_t pos = 0;
+          while (pos < gpuflag.size()) {
+            size_t next = gpuflag.find(',', pos);
+            if (next == std::string::npos) next = gpuflag.size();
+            devs.push_back(atoi(gpuflag.substr(pos, next - pos).c_str()));
+            pos = next + 1;
+          }
+        }
+        return run_multi_gpu_train(flags, devs);
+      }
+      // `-gpu` with no value selects device 0 (old setup_device rule);
+      // absence of the flag means CPU mode
+      const int dev =
+          flags.count("gpu") ? atoi(gpuflag.c_str()) : -1;
+      return run_train_rank(flags, dev, 0, 1, std::string());
     }
     if (cmd == "test") {
       CHECK_(flags.count("model")) << "test needs -model";
