@@ -246,9 +246,11 @@ def main():
         solver.bcast_weights()
 
     def barrier():
+        # local device drained first, THEN the cross-rank barrier: t0/t1 on
+        # every rank brackets only finished GPU work (contract ordering)
+        ca.device_synchronize()
         if dist:
             dist.barrier()
-        ca.device_synchronize()
 
     solver.step(args.warmup)
     barrier()
