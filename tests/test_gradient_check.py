@@ -206,6 +206,22 @@ def test_grad_relu():
     check_gradients(net, "out")
 
 
+def test_grad_fanout_split_accumulation():
+    # one blob consumed by TWO layers: its diff must accumulate both
+    # consumers' contributions (the reference's insert_splits semantics,
+    # net.cpp InsertSplits — the GoogLeNet inception fan-out pattern)
+    net = build_net("""layer { name: "a" type: "InnerProduct"
+  bottom: "in0" top: "t1" inner_product_param { num_output: 4
+  weight_filler { type: "gaussian" std: 0.3 } } }
+layer { name: "b" type: "InnerProduct" bottom: "in0" top: "t2"
+  inner_product_param { num_output: 4
+  weight_filler { type: "gaussian" std: 0.3 } } }
+layer { name: "e" type: "Eltwise" bottom: "t1" bottom: "t2" top: "out"
+  eltwise_param { operation: SUM } }""", [(3, 6)])
+    net.set_blob("in0", seeded((3, 6)))
+    check_gradients(net, "out")
+
+
 def test_grad_softmax_with_loss():
     net = build_net("""layer { name: "loss" type: "SoftmaxWithLoss"
   bottom: "in0" bottom: "in1" top: "out" }""",
