@@ -11,11 +11,7 @@ import numpy as np
 from engine_util import TOL, relerr, net_from_text
 import caffe_amd as ca
 
-import sys
-import os
-sys.path.insert(0, os.path.join(
-    os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "oracle"))
-import oracle  # noqa: E402
+from oracle import oracle  # noqa: E402  (engine_util puts REPO on sys.path)
 
 NET = """name: "chain"
 layer {
