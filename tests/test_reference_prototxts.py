@@ -57,6 +57,27 @@ def test_reference_googlenet_train_val():
 
 
 @needs_ref
+def test_reference_resnet50_solver_file():
+    # the reference's own solver.prototxt verbatim (poly policy, comments,
+    # relative net path — resolved from the reference root as cwd; nothing
+    # is written: snapshot interval 2.5M is never reached and the python
+    # API path does not snapshot after train)
+    ca.set_mode("cpu")
+    ca.set_synthetic_shape(3, 224, 224, 1000)
+    cwd = os.getcwd()
+    os.chdir(REF)
+    try:
+        solver = ca.Solver(
+            path=os.path.join(REF, "models/resnet50/solver.prototxt"),
+            batch_override=2)
+        solver.step(2)
+        loss = solver.loss()
+    finally:
+        os.chdir(cwd)
+    assert np.isfinite(loss) and loss > 0, loss
+
+
+@needs_ref
 def test_reference_lenet_train_test():
     train_steps(
         os.path.join(REF, "examples/mnist/lenet_train_test.prototxt"),
