@@ -57,6 +57,25 @@ def test_reference_googlenet_train_val():
 
 
 @needs_ref
+@pytest.mark.parametrize("name,shape,classes", [
+    ("alexnet_bn", (3, 227, 227), 1000),     # BN variant (engine: CAFFE
+                                             # subfield tolerated)
+    ("alexnet_owt", (3, 224, 224), 1000),    # one-weird-trick AlexNet
+    ("bvlc_reference_caffenet", (3, 227, 227), 1000),  # LRN CaffeNet
+])
+def test_reference_extra_model_families(name, shape, classes):
+    train_steps(os.path.join(REF, "models", name, "train_val.prototxt"),
+                shape, classes, steps=1)
+
+
+@needs_ref
+def test_reference_cifar10_nv():
+    train_steps(os.path.join(
+        REF, "models/cifar10_nv/cifar10_nv_train_test.prototxt"),
+        (3, 28, 28), 10, steps=1, batch=4)
+
+
+@needs_ref
 def test_reference_resnet50_solver_file():
     # the reference's own solver.prototxt verbatim (poly policy, comments,
     # relative net path — resolved from the reference root as cwd; nothing
