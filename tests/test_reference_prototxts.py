@@ -62,6 +62,10 @@ def test_reference_googlenet_train_val():
                                              # subfield tolerated)
     ("alexnet_owt", (3, 224, 224), 1000),    # one-weird-trick AlexNet
     ("bvlc_reference_caffenet", (3, 227, 227), 1000),  # LRN CaffeNet
+    ("resnet18", (3, 224, 224), 1000),       # msra filler
+    ("vgg16", (3, 224, 224), 1000),
+    ("inception_v2", (3, 224, 224), 1000),   # standalone Softmax aux head
+    ("inception_v3", (3, 299, 299), 1000),
 ])
 def test_reference_extra_model_families(name, shape, classes):
     train_steps(os.path.join(REF, "models", name, "train_val.prototxt"),
