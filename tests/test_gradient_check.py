@@ -127,6 +127,20 @@ def setup_function(_):
     ca.set_random_seed(17)
 
 
+@pytest.mark.gpu
+@pytest.mark.parametrize("case", ["conv", "bn", "pool", "ip"])
+def test_grad_gpu(case):
+    # same FD harness through the HIP kernels: pins each GPU backward
+    # against its own GPU forward, independent of the CPU oracle
+    ca.set_mode("gpu")
+    ca.set_random_seed(17)
+    {"conv": test_grad_conv,
+     "bn": test_grad_batchnorm,
+     "pool": test_grad_pool_max_and_ave,
+     "ip": test_grad_inner_product}[case]()
+    ca.set_mode("cpu")
+
+
 def test_grad_conv():
     net = build_net("""layer { name: "c" type: "Convolution" bottom: "in0"
   top: "out" convolution_param { num_output: 4 kernel_size: 3 pad: 1

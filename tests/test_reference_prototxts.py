@@ -101,6 +101,27 @@ def test_reference_resnet50_solver_file():
 
 
 @needs_ref
+@pytest.mark.parametrize("deploy,prob", [
+    ("models/bvlc_alexnet/deploy.prototxt", "prob"),
+    ("models/bvlc_reference_rcnn_ilsvrc13/deploy.prototxt",
+     "fc-rcnn"),  # ends in an InnerProduct score head
+])
+def test_reference_deploy_inference(deploy, prob):
+    # deploy nets: Input layer + Softmax/score head, pure inference
+    ca.set_mode("cpu")
+    net = ca.Net.from_file(os.path.join(REF, deploy), phase=1)
+    rng = np.random.default_rng(11)
+    shape = net.blob_shape("data")
+    net.set_blob("data", rng.standard_normal(shape).astype(np.float32))
+    net.forward()
+    out = net.blob(prob)
+    assert np.all(np.isfinite(out))
+    if prob == "prob":  # softmax head sums to 1 per image
+        s = out.reshape(out.shape[0], -1).sum(1)
+        assert np.allclose(s, 1.0, atol=1e-4), s
+
+
+@needs_ref
 def test_reference_lenet_train_test():
     train_steps(
         os.path.join(REF, "examples/mnist/lenet_train_test.prototxt"),
