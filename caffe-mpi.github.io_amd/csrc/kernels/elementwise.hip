@@ -349,9 +349,49 @@ __global__ void k_bn_fwd_stats(const float* __restrict__ x, int N, int C,
   if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
 }
 
+// f4 variant (S % 4 == 0: every ResNet spatial stage except 7x7=49):
+// quarters the load count and the per-element index divisions
+__global__ void k_bn_fwd_stats_v4(const f4* __restrict__ x4, int N, int C,
+                                  int S4, int nb, double2* __restrict__ out) {
+  const int c = blockIdx.x % C;
+  const int slice = blockIdx.x / C;
+  const int n0 = (int)((long)N * slice / nb);
+  const int n1 = (int)((long)N * (slice + 1) / nb);
+  const int span = (n1 - n0) * S4;
+  double s1 = 0, s2 = 0;
+  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+    const int n = n0 + i / S4;
+    const int sp = i - (n - n0) * S4;
+    const f4 v = x4[((long)n * C + c) * S4 + sp];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      s1 += v[j];
+      s2 += (double)v[j] * v[j];
+    }
+  }
+  __shared__ double sh1[TPB], sh2[TPB];
+  sh1[threadIdx.x] = s1;
+  sh2[threadIdx.x] = s2;
+  __syncthreads();
+  for (int off = TPB / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sh1[threadIdx.x] += sh1[threadIdx.x + off];
+      sh2[threadIdx.x] += sh2[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
+}
+
 void bn_fwd_stats(hipStream_t s, const float* x, int N, int C, long S,
                   int nb, void* partials) {
   PerfScope perf(PERF_CLASS("bn"), s, 0, 4.0 * N * C * S);
+  if (S % 4 == 0) {
+    hipLaunchKernelGGL(k_bn_fwd_stats_v4, dim3(C * nb), dim3(TPB), 0, s,
+                       (const f4*)x, N, C, (int)(S / 4), nb,
+                       (double2*)partials);
+    return;
+  }
   hipLaunchKernelGGL(k_bn_fwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, N, C,
                      S, nb, (double2*)partials);
 }
@@ -518,11 +558,63 @@ __global__ void k_bn_bwd_stats(const float* __restrict__ x,
   }
   if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
 }
+__global__ void k_bn_bwd_stats_v4(const f4* __restrict__ x4,
+                                  const f4* __restrict__ dy4,
+                                  const float* __restrict__ mean,
+                                  const float* __restrict__ inv_std, int N,
+                                  int C, int S4, int nb,
+                                  const float* __restrict__ scale,
+                                  const float* __restrict__ bias, int frelu,
+                                  double2* __restrict__ out) {
+  const int c = blockIdx.x % C;
+  const int slice = blockIdx.x / C;
+  const int n0 = (int)((long)N * slice / nb);
+  const int n1 = (int)((long)N * (slice + 1) / nb);
+  const int span = (n1 - n0) * S4;
+  const float m = mean[c], inv = inv_std[c];
+  const float sc = scale ? scale[c] : 1.f;
+  const float bi = bias ? bias[c] : 0.f;
+  double s_dy = 0, s_dyxn = 0;
+  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+    const int n = n0 + i / S4;
+    const long off = ((long)n * C + c) * S4 + (i - (n - n0) * S4);
+    const f4 xv = x4[off];
+    const f4 dv = dy4[off];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float xn = (xv[j] - m) * inv;
+      double d = dv[j];
+      if (frelu && xn * sc + bi <= 0.f) d = 0.0;
+      s_dy += d;
+      s_dyxn += d * (double)xn;
+    }
+  }
+  __shared__ double sh1[TPB], sh2[TPB];
+  sh1[threadIdx.x] = s_dy;
+  sh2[threadIdx.x] = s_dyxn;
+  __syncthreads();
+  for (int off = TPB / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      sh1[threadIdx.x] += sh1[threadIdx.x + off];
+      sh2[threadIdx.x] += sh2[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) out[(long)c * nb + slice] = {sh1[0], sh2[0]};
+}
+
 void bn_bwd_stats(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std, int N, int C,
                   long S, int nb, const float* scale, const float* bias,
                   int frelu, void* partials) {
   PerfScope perf(PERF_CLASS("bn"), s, 0, 8.0 * N * C * S);
+  if (S % 4 == 0) {
+    hipLaunchKernelGGL(k_bn_bwd_stats_v4, dim3(C * nb), dim3(TPB), 0, s,
+                       (const f4*)x, (const f4*)dy, mean, inv_std, N, C,
+                       (int)(S / 4), nb, scale, bias, frelu,
+                       (double2*)partials);
+    return;
+  }
   hipLaunchKernelGGL(k_bn_bwd_stats, dim3(C * nb), dim3(TPB), 0, s, x, dy,
                      mean, inv_std, N, C, S, nb, scale, bias, frelu,
                      (double2*)partials);
