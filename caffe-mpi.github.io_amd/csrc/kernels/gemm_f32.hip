@@ -461,6 +461,42 @@ __global__ void k_splitk_reduce(const float* __restrict__ slab, long MN,
   }
 }
 
+// MN % 4 == 0 fast path (wgrad MN = Cout x C*kh*kw — always a multiple of
+// 4 in practice): float4 rows + 4-way slab unroll keeps 16 loads in
+// flight per thread instead of 1 — the scalar loop is latency-bound, not
+// bandwidth-bound.  Fixed-shape summation (deterministic).
+__global__ void k_splitk_reduce_v4(const float* __restrict__ slab, long MN4,
+                                   long MN, int SK, float4* __restrict__ C) {
+  const float4* s4 = (const float4*)slab;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < MN4;
+       i += (long)gridDim.x * blockDim.x) {
+    const long mn4 = MN >> 2;  // f4 elements per slab
+    float4 a0 = {0, 0, 0, 0}, a1 = {0, 0, 0, 0};
+    float4 a2 = {0, 0, 0, 0}, a3 = {0, 0, 0, 0};
+    int s = 0;
+    for (; s + 4 <= SK; s += 4) {
+      const float4 v0 = s4[(long)s * mn4 + i];
+      const float4 v1 = s4[(long)(s + 1) * mn4 + i];
+      const float4 v2 = s4[(long)(s + 2) * mn4 + i];
+      const float4 v3 = s4[(long)(s + 3) * mn4 + i];
+      a0.x += v0.x; a0.y += v0.y; a0.z += v0.z; a0.w += v0.w;
+      a1.x += v1.x; a1.y += v1.y; a1.z += v1.z; a1.w += v1.w;
+      a2.x += v2.x; a2.y += v2.y; a2.z += v2.z; a2.w += v2.w;
+      a3.x += v3.x; a3.y += v3.y; a3.z += v3.z; a3.w += v3.w;
+    }
+    for (; s < SK; ++s) {
+      const float4 v = s4[(long)s * mn4 + i];
+      a0.x += v.x; a0.y += v.y; a0.z += v.z; a0.w += v.w;
+    }
+    float4 r;
+    r.x = (a0.x + a1.x) + (a2.x + a3.x);
+    r.y = (a0.y + a1.y) + (a2.y + a3.y);
+    r.z = (a0.z + a1.z) + (a2.z + a3.z);
+    r.w = (a0.w + a1.w) + (a2.w + a3.w);
+    C[i] = r;
+  }
+}
+
 void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
           float alpha, const float* A, long lda, const float* B, long ldb,
           float beta, float* C, long ldc, const GemmEpi* epi,
@@ -538,6 +574,13 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
       hipLaunchKernelGGL((k_gemm_f32<true, true, true>), grid, block, 0, s,
                          g);
     const long MN = M * N;
+    if (MN % 4 == 0) {
+      const long MN4 = MN / 4;
+      const int blocks = (int)std::min<long>((MN4 + 255) / 256, 2048);
+      hipLaunchKernelGGL(k_splitk_reduce_v4, dim3(blocks), dim3(256), 0, s,
+                         slab, MN4, MN, SK, (float4*)C);
+      return;
+    }
     const int blocks = (int)std::min<long>((MN + 255) / 256, 2048);
     hipLaunchKernelGGL(k_splitk_reduce, dim3(blocks), dim3(256), 0, s, slab,
                        MN, SK, C);
