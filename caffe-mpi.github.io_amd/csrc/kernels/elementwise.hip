@@ -430,6 +430,7 @@ __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                               const float* __restrict__ scale,
                               const float* __restrict__ bias, int sb, int C,
                               int S, int frelu, long n4,
+                              const f4* __restrict__ add,
                               f4* __restrict__ y) {
   VEC_GRID(i, n4) {
     const long e0 = i * 4;
@@ -441,34 +442,40 @@ __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
       const float mu = mean[c0], inv = inv_std[c0];
       const float sc = sb ? scale[c0] : 1.f, bi = sb ? bias[c0] : 0.f;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        float t = (v[j] - mu) * inv * sc + bi;
-        v[j] = frelu ? fmaxf(t, 0.f) : t;
-      }
+      for (int j = 0; j < 4; ++j) v[j] = (v[j] - mu) * inv * sc + bi;
     } else {  // pack crosses a channel boundary
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const long e = e0 + j;
         const int cc = (int)((e / S) % C);
-        float t = (v[j] - mean[cc]) * inv_std[cc] * (sb ? scale[cc] : 1.f) +
-                  (sb ? bias[cc] : 0.f);
-        v[j] = frelu ? fmaxf(t, 0.f) : t;
+        v[j] = (v[j] - mean[cc]) * inv_std[cc] * (sb ? scale[cc] : 1.f) +
+               (sb ? bias[cc] : 0.f);
       }
+    }
+    if (add) {  // fused residual: bn(x) + other (ResNet block pattern)
+      const f4 a = add[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += a[j];
+    }
+    if (frelu) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
     }
     y[i] = v;
   }
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
-                 int sb, int N, int C, long S, float* y, int fuse_relu) {
+                 int sb, int N, int C, long S, float* y, int fuse_relu,
+                 const float* add) {
   const long total = (long)N * C * S;
-  PerfScope perf(PERF_CLASS("bn"), s, 0, 8.0 * total);
+  PerfScope perf(PERF_CLASS("bn"), s, 0, (add ? 12.0 : 8.0) * total);
   // blobs are 16-float padded: the final partial pack reads/writes pad
   // space with wrapped (but in-bounds) channel indices — harmless
   const long n4 = (total + 3) / 4;
   hipLaunchKernelGGL(k_bn_fwd_norm, dim3(nblocks(n4, 4)), dim3(TPB), 0, s,
                      (const f4*)x, mean, inv_std, scale, bias, sb, C,
-                     (int)S, fuse_relu, n4, (f4*)y);
+                     (int)S, fuse_relu, n4, (const f4*)add, (f4*)y);
 }
 
 __global__ void k_bn_moving_avg(const float* __restrict__ mean,

@@ -228,6 +228,13 @@ class BatchNormLayer : public Layer {
   float maf_ = 0.999f, eps_ = 1e-5f;
   bool scale_bias_ = false, use_global_ = false;
   bool fuse_relu_ = false;  // GPU graph fusion: BN+ReLU forward in one pass
+  // GPU graph fusion: a following Eltwise(SUM) absorbed into the norm
+  // epilogue — forward writes fuse_add_out_ = bn(x) + fuse_add_other_
+  // (bit-identical to the separate add; backward is untouched, the
+  // eltwise aliases the sum's diff onto this layer's top)
+  Blob* fuse_add_other_ = nullptr;
+  Blob* fuse_add_out_ = nullptr;
+  bool fuse_add_relu_ = false;
   int C_ = 0;
   long iter_ = 0;
   Blob mean_, var_, inv_std_, m_dy_, m_dyxn_, partials_;
@@ -272,6 +279,7 @@ class EltwiseLayer : public Layer {
   std::vector<float> coeffs_;
   std::string op_ = "SUM";
   bool fuse_relu_ = false;  // GPU graph fusion: SUM+ReLU forward in one pass
+  bool fused_away_ = false;  // producing BN writes bn(x)+other directly
 };
 
 class LRNLayer : public Layer {

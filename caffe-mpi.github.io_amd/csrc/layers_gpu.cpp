@@ -286,8 +286,15 @@ void BatchNormLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   gpu::bn_fwd_finalize(E.stream, parts, nb, C_, (long)N * S, eps_,
                        mean_.mutable_gpu_data(), var_.mutable_gpu_data(),
                        inv_std_.mutable_gpu_data());
+  const float* add = nullptr;
+  int frelu = fuse_relu_;
+  if (fuse_add_out_) {  // fused residual add: write the sum's top directly
+    y = fuse_add_out_->mutable_gpu_data();
+    add = fuse_add_other_->gpu_data();
+    frelu = fuse_add_relu_;
+  }
   gpu::bn_fwd_norm(E.stream, x, mean_.gpu_data(), inv_std_.gpu_data(), sc,
-                   bi, scale_bias_, N, C_, S, y, fuse_relu_);
+                   bi, scale_bias_, N, C_, S, y, frelu, add);
   gpu::bn_moving_avg(E.stream, mean_.gpu_data(), var_.gpu_data(), C_, maf_,
                      iter_ <= 1 ? 1 : 0, blobs_[0]->mutable_gpu_data(),
                      blobs_[1]->mutable_gpu_data());
@@ -352,6 +359,7 @@ void ReLULayer::Backward_gpu(const std::vector<Blob*>& top,
 // --------------------------------------------------------------- Eltwise
 void EltwiseLayer::Forward_gpu(const std::vector<Blob*>& bottom,
                                const std::vector<Blob*>& top) {
+  if (fused_away_) return;  // the producing BN already wrote bn(x)+other
   Engine& E = Engine::get();
   const long n = top[0]->count();
   float* y = top[0]->mutable_gpu_data();

@@ -125,7 +125,7 @@ void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
                  int scale_bias, int N, int C, long S, float* y,
-                 int fuse_relu = 0);
+                 int fuse_relu = 0, const float* add = nullptr);
 void bn_moving_avg(hipStream_t s, const float* mean, const float* var, int C,
                    float maf, int copy_only, float* gmean, float* gvar);
 void bn_fwd_test(hipStream_t s, const float* x, const float* gmean,

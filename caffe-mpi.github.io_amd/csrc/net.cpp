@@ -227,6 +227,39 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
         }
       }
     }
+    // residual-add fusion (TRAIN only; the TEST BN path has no epilogue):
+    // Eltwise(SUM, coeffs 1) whose first-hand operand is produced by the
+    // immediately preceding BatchNorm, and that operand has no other
+    // consumer — the BN's norm epilogue writes bn(x)+other straight into
+    // the sum's top (the ResNet block pattern; kills one full
+    // read-add-write pass over the activation per block)
+    if (phase_ == Phase::TRAIN) {
+      for (size_t i = 1; i < layers_.size(); ++i) {
+        auto* el = dynamic_cast<EltwiseLayer*>(layers_[i].get());
+        if (!el || el->fused_away_) continue;
+        if (el->op_ != "SUM" || bottoms_[i].size() != 2) continue;
+        bool ones = true;
+        for (float c : el->coeffs_) ones = ones && c == 1.f;
+        if (!ones) continue;
+        auto* bn = dynamic_cast<BatchNormLayer*>(layers_[i - 1].get());
+        if (!bn || bn->fuse_relu_ || bn->fuse_add_out_) continue;
+        if (tops_[i - 1].size() != 1) continue;
+        Blob* bnout = tops_[i - 1][0];
+        if (bnout == bottoms_[i - 1][0]) continue;  // in-place BN: keep
+        int idx = bnout == bottoms_[i][0] ? 0
+                  : bnout == bottoms_[i][1] ? 1
+                                            : -1;
+        if (idx < 0) continue;
+        int consumers = 0;
+        for (size_t k = 0; k < layers_.size(); ++k)
+          for (auto* b : bottoms_[k]) consumers += b == bnout;
+        if (consumers != 1) continue;
+        bn->fuse_add_other_ = bottoms_[i][1 - idx];
+        bn->fuse_add_out_ = tops_[i][0];
+        bn->fuse_add_relu_ = el->fuse_relu_;
+        el->fused_away_ = true;
+      }
+    }
     setup_arena();
   }
 }
