@@ -245,6 +245,10 @@ class ReLULayer : public Layer {
   using Layer::Layer;
   bool fused_away_ = false;  // producer applies the ReLU in its epilogue
   bool bwd_fused_ = false;   // producer also absorbs the backward mask
+  // non-in-place fused ReLU whose bottom (the pre-activation) is never
+  // materialized: backward masks by the TOP's sign — identical for
+  // slope 0 (top > 0 <=> bottom > 0)
+  bool bwd_from_top_ = false;
   void Reshape(const std::vector<Blob*>& b,
                const std::vector<Blob*>& t) override {
     if (b[0] != t[0]) t[0]->ReshapeLike(*b[0]);

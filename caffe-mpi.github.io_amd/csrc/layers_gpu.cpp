@@ -351,9 +351,13 @@ void ReLULayer::Backward_gpu(const std::vector<Blob*>& top,
   if (fused_away_ && bwd_fused_) return;  // absorbed into BN backward
   auto rp = param_->sub("relu_param");
   const float slope = rp ? (float)rp->num("negative_slope", 0) : 0.f;
-  gpu::relu_bwd(Engine::get().stream, bottom[0]->gpu_data(),
-                top[0]->gpu_diff(), bottom[0]->count(), slope,
-                bottom[0]->mutable_gpu_diff());
+  // bwd_from_top_: the pre-activation was never materialized (fused
+  // producer wrote post-ReLU into our top); for slope 0 the top's sign
+  // is the same mask
+  const float* ref =
+      bwd_from_top_ ? top[0]->gpu_data() : bottom[0]->gpu_data();
+  gpu::relu_bwd(Engine::get().stream, ref, top[0]->gpu_diff(),
+                bottom[0]->count(), slope, bottom[0]->mutable_gpu_diff());
 }
 
 // --------------------------------------------------------------- Eltwise

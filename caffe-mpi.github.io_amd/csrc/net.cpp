@@ -258,6 +258,30 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
         bn->fuse_add_out_ = tops_[i][0];
         bn->fuse_add_relu_ = el->fuse_relu_;
         el->fused_away_ = true;
+        // absorb a following non-in-place slope-0 ReLU too: the epilogue
+        // writes relu(bn(x)+other) straight into the ReLU's top and the
+        // sum is never materialized (its only consumer is the ReLU, whose
+        // backward masks by the top's sign instead)
+        if (i + 1 < layers_.size()) {
+          auto* relu = dynamic_cast<ReLULayer*>(layers_[i + 1].get());
+          if (relu && !relu->fused_away_ && bottoms_[i + 1].size() == 1 &&
+              tops_[i + 1].size() == 1 &&
+              bottoms_[i + 1][0] == tops_[i][0] &&
+              bottoms_[i + 1][0] != tops_[i + 1][0]) {
+            auto rp = relu->param()->sub("relu_param");
+            const bool slope0 = !rp || rp->num("negative_slope", 0) == 0;
+            Blob* sum = tops_[i][0];
+            int sum_consumers = 0;
+            for (size_t k = 0; k < layers_.size(); ++k)
+              for (auto* b : bottoms_[k]) sum_consumers += b == sum;
+            if (slope0 && sum_consumers == 1) {
+              bn->fuse_add_out_ = tops_[i + 1][0];
+              bn->fuse_add_relu_ = true;
+              relu->fused_away_ = true;
+              relu->bwd_from_top_ = true;
+            }
+          }
+        }
       }
     }
     setup_arena();
