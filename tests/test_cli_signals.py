@@ -15,13 +15,58 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 CAFFE = os.path.join(REPO, "caffe-mpi.github.io_amd", "caffe")
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 
-from test_cli_snapshot import make_lenet_solver, run_env  # noqa: E402
+from test_cli_snapshot import run_env  # noqa: E402
+
+# a deliberately tiny net (one small IP) so a single iteration stays in the
+# millisecond range even on a heavily loaded machine — the signal tests
+# bound how long the process may take to REACT, which is measured in
+# iterations
+TINY_SOLVER = """base_lr: 0.01
+lr_policy: "fixed"
+snapshot_prefix: "{prefix}"
+max_iter: 100000000
+net_param {{
+  name: "tiny"
+  layer {{
+    name: "input"
+    type: "Input"
+    top: "in0"
+    top: "in1"
+    input_param {{
+      shape {{ dim: 4 dim: 8 }}
+      shape {{ dim: 4 }}
+    }}
+  }}
+  layer {{
+    name: "ip"
+    type: "InnerProduct"
+    bottom: "in0"
+    top: "fc"
+    inner_product_param {{ num_output: 3 }}
+  }}
+  layer {{
+    name: "loss"
+    type: "SoftmaxWithLoss"
+    bottom: "fc"
+    bottom: "in1"
+    top: "loss"
+  }}
+}}
+"""
+
+
+def make_tiny_solver(tmp, extra=""):
+    path = os.path.join(tmp, "solver.prototxt")
+    with open(path, "w") as f:
+        f.write(TINY_SOLVER.format(prefix=os.path.join(tmp, "lenet")))
+        f.write(extra + "\n")
+    return path
 
 
 def start_train(solver, extra=()):
     env = dict(run_env(), CAFFE_SYN_SHAPE="1x28x28x10")
     return subprocess.Popen(
-        [CAFFE, "train", f"-solver={solver}", "-iterations=1000000",
+        [CAFFE, "train", f"-solver={solver}", "-iterations=1000000000",
          *extra],
         env=env, cwd=REPO,
         stdout=subprocess.PIPE, stderr=subprocess.PIPE)
@@ -29,11 +74,11 @@ def start_train(solver, extra=()):
 
 def test_sigint_stops_training():
     with tempfile.TemporaryDirectory() as tmp:
-        solver = make_lenet_solver(tmp, extra="test_interval: 0")
+        solver = make_tiny_solver(tmp)
         p = start_train(solver)
         time.sleep(3)  # past model build, mid-Step
         p.send_signal(signal.SIGINT)
-        _, err = p.communicate(timeout=120)
+        _, err = p.communicate(timeout=300)
         err = err.decode()
         assert p.returncode == 0, err
         assert "Optimization stopped early." in err, err
@@ -44,7 +89,7 @@ def test_sigint_stops_training():
 
 def test_sighup_snapshots_and_continues():
     with tempfile.TemporaryDirectory() as tmp:
-        solver = make_lenet_solver(tmp, extra="test_interval: 0")
+        solver = make_tiny_solver(tmp)
         p = start_train(solver)
         time.sleep(3)
         p.send_signal(signal.SIGHUP)
@@ -59,17 +104,17 @@ def test_sighup_snapshots_and_continues():
         assert snaps, "no snapshot after SIGHUP"
         assert p.poll() is None, "SIGHUP must not stop training"
         p.send_signal(signal.SIGINT)
-        _, err = p.communicate(timeout=120)
+        _, err = p.communicate(timeout=300)
         assert p.returncode == 0, err.decode()
 
 
 def test_sigint_effect_none_ignores_signal():
     with tempfile.TemporaryDirectory() as tmp:
-        solver = make_lenet_solver(tmp, extra="test_interval: 0")
+        solver = make_tiny_solver(tmp)
         p = start_train(solver, extra=("-sigint_effect=none",))
         time.sleep(3)
         p.send_signal(signal.SIGINT)
         time.sleep(2)
         assert p.poll() is None, "SIGINT with effect=none must be ignored"
         p.kill()
-        p.communicate(timeout=60)
+        p.communicate(timeout=300)
