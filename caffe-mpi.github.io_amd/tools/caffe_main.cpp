@@ -341,10 +341,18 @@ int main(int argc, char** argv) {
         }
         return run_multi_gpu_train(flags, devs);
       }
-      // `-gpu` with no value selects device 0 (old setup_device rule);
-      // absence of the flag means CPU mode
-      const int dev =
-          flags.count("gpu") ? atoi(gpuflag.c_str()) : -1;
+      // `-gpu` with no value selects device 0 (old setup_device rule).
+      // Without the flag the solver file decides: solver_mode GPU (the
+      // reference default, caffe.proto SolverParameter) + device_id —
+      // tools/caffe.cpp:154 honors solver_param.solver_mode the same way
+      int dev = -1;
+      if (flags.count("gpu")) {
+        dev = atoi(gpuflag.c_str());
+      } else {
+        auto sp = parse_prototxt_file(flags["solver"]);
+        if (sp->str("solver_mode", "CPU") == "GPU")
+          dev = (int)sp->inum("device_id", 0);
+      }
       return run_train_rank(flags, dev, 0, 1, std::string());
     }
     if (cmd == "test") {
