@@ -533,9 +533,15 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // split-K when the output grid cannot fill the chip (wgrad shapes);
   // plain-C calls only (no epilogue) — fused-epilogue GEMMs have huge N
   int SK = 1;
-  if (!epi && beta == 0.f && g.tiles < 1024 && K > 4 * BK) {
+  // split-K block target: enough tiles*SK to oversubscribe the 256 CUs
+  // ~2x at 2 waves/SIMD occupancy (A/B-able via CAFFE_SK_TARGET)
+  static const long sk_target = [] {
+    const char* e = getenv("CAFFE_SK_TARGET");
+    return e ? atol(e) : 1024L;
+  }();
+  if (!epi && beta == 0.f && g.tiles < sk_target && K > 4 * BK) {
     SK = (int)std::min<long>(
-        {1024 / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 256});
+        {sk_target / g.tiles + 1, (K + 4 * BK - 1) / (4 * BK), 256});
   }
   const char* pcls = !transA ? (!transB ? "gemm_nn" : "gemm_nt")
                             : (!transB ? "gemm_tn" : "gemm_tt");
