@@ -65,6 +65,57 @@ def test_conv_vs_torch(case):
     assert relerr(net.param(1, diff=True), bt.grad.numpy().ravel()) < TOL
 
 
+@pytest.mark.parametrize("kh,kw,ph,pw", [(1, 7, 0, 3), (7, 1, 3, 0),
+                                         (3, 5, 1, 2)])
+def test_conv_asymmetric_vs_torch(kh, kw, ph, pw):
+    # kernel_h/kernel_w + pad_h/pad_w — the inception_v3 1x7/7x1 pattern
+    n, cin, h, w, cout = 2, 3, 9, 9, 4
+    rng = np.random.default_rng(kh * 10 + kw)
+    x = rng.standard_normal((n, cin, h, w)).astype(np.float32)
+    wgt = rng.standard_normal((cout, cin, kh, kw)).astype(np.float32) * 0.4
+    b = rng.standard_normal(cout).astype(np.float32) * 0.1
+
+    xt = torch.tensor(x, requires_grad=True)
+    wt = torch.tensor(wgt, requires_grad=True)
+    bt = torch.tensor(b, requires_grad=True)
+    yt = torch.nn.functional.conv2d(xt, wt, bt, padding=(ph, pw))
+    dy = rng.standard_normal(tuple(yt.shape)).astype(np.float32)
+    yt.backward(torch.tensor(dy))
+
+    body = f"""layer {{ name: "c" type: "Convolution" bottom: "in0"
+  top: "out" convolution_param {{ num_output: {cout} kernel_h: {kh}
+  kernel_w: {kw} pad_h: {ph} pad_w: {pw} }} }}"""
+    net, out = run_layer("cpu", [(n, cin, h, w)], body, [x],
+                         params=[wgt, b], top_diff=dy)
+    assert relerr(out, yt.detach().numpy()) < TOL
+    assert relerr(net.blob("in0", diff=True), xt.grad.numpy()) < TOL
+    assert relerr(net.param(0, diff=True), wt.grad.numpy().ravel()) < TOL
+    assert relerr(net.param(1, diff=True), bt.grad.numpy().ravel()) < TOL
+
+
+@pytest.mark.gpu
+def test_conv_asymmetric_gpu_vs_torch():
+    # the HIP explicit-col path with kernel_h != kernel_w, vs torch-CPU
+    n, cin, h, w, cout, kh, kw, ph, pw = 2, 3, 9, 9, 4, 1, 7, 0, 3
+    rng = np.random.default_rng(3)
+    x = rng.standard_normal((n, cin, h, w)).astype(np.float32)
+    wgt = rng.standard_normal((cout, cin, kh, kw)).astype(np.float32) * 0.4
+    b = rng.standard_normal(cout).astype(np.float32) * 0.1
+    xt = torch.tensor(x, requires_grad=True)
+    wt = torch.tensor(wgt, requires_grad=True)
+    bt = torch.tensor(b, requires_grad=True)
+    yt = torch.nn.functional.conv2d(xt, wt, bt, padding=(ph, pw))
+    dy = rng.standard_normal(tuple(yt.shape)).astype(np.float32)
+    yt.backward(torch.tensor(dy))
+    body = f"""layer {{ name: "c" type: "Convolution" bottom: "in0"
+  top: "out" convolution_param {{ num_output: {cout} kernel_h: {kh}
+  kernel_w: {kw} pad_h: {ph} pad_w: {pw} }} }}"""
+    net, out = run_layer("gpu", [(n, cin, h, w)], body, [x],
+                         params=[wgt, b], top_diff=dy)
+    assert relerr(out, yt.detach().numpy()) < TOL
+    assert relerr(net.param(0, diff=True), wt.grad.numpy().ravel()) < TOL
+
+
 @pytest.mark.parametrize("shape", [(2, 3, 6, 6), (4, 8, 5, 5), (1, 16, 3, 3)])
 def test_batchnorm_vs_torch(shape):
     n, c, h, w = shape
