@@ -63,6 +63,7 @@ def main():
     ap.add_argument("--rank-data", choices=["same", "shard", "combined"],
                     default="same")
     ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--iter-size", type=int, default=1)
     ap.add_argument("--restore-at", type=int, default=-1,
                     help="snapshot at this iter, rebuild the solver, "
                          "restore, re-attach the comm, continue")
@@ -75,6 +76,8 @@ def main():
     tmp = tempfile.mkdtemp()
     text = (SOLVER_TEXT % (args.batch, args.batch)).replace(
         "SNAPPFX", os.path.join(tmp, f"r{rank}"))
+    if args.iter_size > 1:
+        text = f"iter_size: {args.iter_size}\n" + text
     solver = ca.Solver(text=text)
     net = solver.net
 

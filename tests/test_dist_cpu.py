@@ -82,6 +82,20 @@ def test_two_ranks_restore_mid_training_matches_uninterrupted():
         np.abs(p_plain - p_resumed).max()
 
 
+def test_two_ranks_iter_size_matches_combined_batch():
+    # rank r: batch 4 shard, iter_size 2 (same input both sub-passes) —
+    # accumulated mean grad == rank r's batch-4 mean; all-reduce average
+    # == the batch-8 combined gradient.  Covers the iter_size
+    # accumulate-then-reduce drive with a comm attached.
+    double = run_dist(2, ["--iters", "3", "--rank-data", "shard",
+                          "--batch", "4", "--iter-size", "2"])
+    single = run_dist(1, ["--iters", "3", "--rank-data", "combined",
+                          "--batch", "8"])
+    p2 = parse_params(double[0])
+    p1 = parse_params(single[0])
+    assert np.allclose(p1, p2, rtol=1e-4, atol=1e-5), np.abs(p1 - p2).max()
+
+
 def test_two_ranks_sharded_match_combined_batch():
     # rank r trains on shard r; equivalent single-rank run feeds the
     # concatenated batch (grad averaging == big-batch gradient)
