@@ -270,6 +270,10 @@ class EltwiseLayer : public Layer {
                   const std::vector<Blob*>&) override;
   void Reshape(const std::vector<Blob*>& b,
                const std::vector<Blob*>& t) override {
+    // eltwise_layer.cpp Reshape: all bottoms must match exactly
+    for (size_t i = 1; i < b.size(); ++i)
+      CHECK_(b[i]->shape() == b[0]->shape())
+          << "Eltwise bottoms must have identical shapes";
     t[0]->ReshapeLike(*b[0]);
   }
   void Forward_cpu(const std::vector<Blob*>&,

@@ -1,0 +1,73 @@
+"""Reference-test property restatements + error-path behavior.
+
+Properties are the reference's own test assertions restated as data-free
+invariants (test_batch_norm_layer.cpp mean/var checks, softmax
+normalization); error paths check the glog-CHECK-to-status-code
+conversion at the C ABI (SURVEY §8b error convention).
+"""
+import numpy as np
+import pytest
+
+import caffe_amd as ca
+from engine_util import net_from_text, input_net, run_layer
+
+
+def test_bn_output_normalized():
+    # test_batch_norm_layer.cpp: per-channel mean ~ 0, variance ~ 1
+    rng = np.random.default_rng(0)
+    x = (rng.standard_normal((6, 4, 7, 7)) * 3 + 5).astype(np.float32)
+    body = """layer { name: "bn" type: "BatchNorm" bottom: "in0"
+  top: "out" }"""
+    _, out = run_layer("cpu", [(6, 4, 7, 7)], body, [x])
+    per_c = out.transpose(1, 0, 2, 3).reshape(4, -1)
+    assert np.abs(per_c.mean(1)).max() < 1e-4
+    assert np.abs(per_c.var(1) - 1).max() < 1e-2
+
+
+def test_softmax_rows_sum_to_one():
+    rng = np.random.default_rng(1)
+    x = (rng.standard_normal((5, 11)) * 10).astype(np.float32)
+    body = """layer { name: "s" type: "Softmax" bottom: "in0"
+  top: "out" }"""
+    _, out = run_layer("cpu", [(5, 11)], body, [x])
+    assert np.allclose(out.sum(1), 1.0, atol=1e-5)
+    assert out.min() >= 0
+
+
+def test_dropout_test_phase_identity():
+    rng = np.random.default_rng(2)
+    x = rng.standard_normal((3, 4, 5, 5)).astype(np.float32)
+    body = """layer { name: "d" type: "Dropout" bottom: "in0" top: "out"
+  dropout_param { dropout_ratio: 0.5 } }"""
+    _, out = run_layer("cpu", [(3, 4, 5, 5)], body, [x], phase=1)
+    assert np.array_equal(out, x)
+
+
+def test_unknown_layer_type_raises():
+    ca.set_mode("cpu")
+    with pytest.raises(Exception) as e:
+        net_from_text(input_net([(1, 2)], """layer { name: "x"
+  type: "FancyUnsupportedLayer" bottom: "in0" top: "out" }"""))
+    assert "FancyUnsupportedLayer" in str(e.value)
+
+
+def test_missing_solver_file_raises():
+    ca.set_mode("cpu")
+    with pytest.raises(Exception):
+        ca.Solver(path="/nonexistent/solver.prototxt")
+
+
+def test_shape_mismatch_raises():
+    # Eltwise with mismatched bottoms must fail loudly, not corrupt
+    ca.set_mode("cpu")
+    with pytest.raises(Exception):
+        net_from_text(input_net(
+            [(2, 3, 4, 4), (2, 3, 5, 5)],
+            """layer { name: "e" type: "Eltwise" bottom: "in0"
+  bottom: "in1" top: "out" }"""))
+
+
+def test_malformed_prototxt_raises():
+    ca.set_mode("cpu")
+    with pytest.raises(Exception):
+        net_from_text("layer { name: \"x\" type: ")
