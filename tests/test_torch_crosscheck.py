@@ -144,6 +144,25 @@ def test_batchnorm_vs_torch(shape):
     assert relerr(net.param(1, diff=True), bt.grad.numpy()) < TOL
 
 
+@pytest.mark.parametrize("size,alpha,beta", [(5, 1e-4, 0.75), (3, 0.5, 0.5)])
+def test_lrn_vs_torch(size, alpha, beta):
+    # caffe cross-channel LRN == torch local_response_norm (both use
+    # 1 + alpha/size * sum within the clipped channel window, k=1)
+    shape = (2, 8, 6, 6)
+    rng = np.random.default_rng(size)
+    x = rng.standard_normal(shape).astype(np.float32)
+    dy = rng.standard_normal(shape).astype(np.float32)
+    xt = torch.tensor(x, requires_grad=True)
+    yt = torch.nn.functional.local_response_norm(xt, size, alpha=alpha,
+                                                 beta=beta, k=1.0)
+    yt.backward(torch.tensor(dy))
+    body = f"""layer {{ name: "l" type: "LRN" bottom: "in0" top: "out"
+  lrn_param {{ local_size: {size} alpha: {alpha} beta: {beta} }} }}"""
+    net, out = run_layer("cpu", [shape], body, [x], top_diff=dy)
+    assert relerr(out, yt.detach().numpy()) < TOL
+    assert relerr(net.blob("in0", diff=True), xt.grad.numpy()) < 5 * TOL
+
+
 @pytest.mark.parametrize("pool,shape,k,s", [
     ("MAX", (2, 3, 9, 9), 3, 2), ("AVE", (2, 3, 8, 8), 2, 2),
     ("MAX", (1, 4, 7, 7), 2, 1)])
