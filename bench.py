@@ -303,7 +303,10 @@ def main():
             for k, v in sorted(perf.items(), key=lambda kv: -kv[1]["ns"])}
 
         cpu_baseline = None
-        if world == 1 and not args.no_cpu_baseline:
+        # the committed oracle walker restates ResNet-50 only; pairing it
+        # with another model's metric would be wrong — omit instead
+        if world == 1 and not args.no_cpu_baseline and \
+                args.model == "resnet50":
             bcpu = 4
             t_cpu = oracle_resnet50_step(bcpu)
             cpu_baseline = {
