@@ -55,7 +55,7 @@ def objective(net, out):
 
 
 def check_gradients(net, out, perturb_inputs=("in0",), n_samples=12,
-                    loss_top=False):
+                    loss_top=False, eps=EPS, thresh=THRESH):
     rng = np.random.default_rng(7)
     net.forward()
     y = net.blob(out)
@@ -80,17 +80,17 @@ def check_gradients(net, out, perturb_inputs=("in0",), n_samples=12,
                           replace=False)
         for i in idxs:
             xi = x[i]
-            x[i] = xi + EPS
+            x[i] = xi + eps
             net.set_blob(name, x)
             op = obj()
-            x[i] = xi - EPS
+            x[i] = xi - eps
             net.set_blob(name, x)
             om = obj()
             x[i] = xi
             net.set_blob(name, x)
-            num = (op - om) / (2 * EPS)
+            num = (op - om) / (2 * eps)
             a = analytic[i]
-            assert abs(a - num) <= THRESH * max(1.0, abs(a), abs(num)), \
+            assert abs(a - num) <= thresh * max(1.0, abs(a), abs(num)), \
                 (name, i, a, num)
     # params (skip lr_mult-0 BN stats: indices with 'bn' stats are still
     # checked numerically-zero-safe since their diffs are zero)
@@ -103,17 +103,17 @@ def check_gradients(net, out, perturb_inputs=("in0",), n_samples=12,
         idxs = rng.choice(cnt, size=min(n_samples, cnt), replace=False)
         for i in idxs:
             wi = w[i]
-            w[i] = wi + EPS
+            w[i] = wi + eps
             net.set_param(pidx, w)
             op = obj()
-            w[i] = wi - EPS
+            w[i] = wi - eps
             net.set_param(pidx, w)
             om = obj()
             w[i] = wi
             net.set_param(pidx, w)
-            num = (op - om) / (2 * EPS)
+            num = (op - om) / (2 * eps)
             a = analytic[i]
-            assert abs(a - num) <= THRESH * max(1.0, abs(a), abs(num)), \
+            assert abs(a - num) <= thresh * max(1.0, abs(a), abs(num)), \
                 (lname, bi, i, a, num)
 
 
