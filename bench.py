@@ -208,9 +208,18 @@ def main():
 
     import caffe_amd as ca
 
-    # initialize HIP through /opt/rocm's runtime BEFORE torch gets a chance
-    # to load its bundled one (same soname; first load wins process-wide)
-    ca.set_mode("gpu", local_rank)
+    # CAFFE_BENCH_CPU=1: CI dry-run of THIS script's multi-rank plumbing
+    # (gloo rendezvous, barriers, max-over-ranks timing, JSON emission) on
+    # CPU with no RCCL — tests/test_bench_dryrun.py runs it at world 2.
+    # Never used for reported numbers.
+    cpu_dry = os.environ.get("CAFFE_BENCH_CPU") == "1"
+    if cpu_dry:
+        ca.set_mode("cpu")
+    else:
+        # initialize HIP through /opt/rocm's runtime BEFORE torch gets a
+        # chance to load its bundled one (same soname; first load wins
+        # process-wide)
+        ca.set_mode("gpu", local_rank)
 
     dist = None
     if world > 1:
@@ -237,7 +246,7 @@ def main():
                 time.sleep(0.5)
     solver = ca.Solver(path=gen, batch_override=args.batch)
 
-    if world > 1:
+    if world > 1 and not cpu_dry:
         # ncclUniqueId from rank 0 over the gloo store (the reference used
         # MPI_Bcast, parallel.cpp:45)
         obj = [solver.comm_unique_id() if rank == 0 else None]
@@ -271,7 +280,7 @@ def main():
     # the step — a rank-0-only step would deadlock the communicator).
     ca.perf_reset()
     ca.set_perf_timing(True)
-    solver.step(3)
+    solver.step(1 if cpu_dry else 3)
     ca.device_synchronize()
     ca.set_perf_timing(False)
     result = None
@@ -306,7 +315,7 @@ def main():
         # the committed oracle walker restates ResNet-50 only; pairing it
         # with another model's metric would be wrong — omit instead
         if world == 1 and not args.no_cpu_baseline and \
-                args.model == "resnet50":
+                args.model == "resnet50" and not cpu_dry:
             bcpu = 4
             t_cpu = oracle_resnet50_step(bcpu)
             cpu_baseline = {
