@@ -311,6 +311,25 @@ void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
                      s, dy, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
 }
 
+// y = x * a[c] (+ b[c]) — Scale layer forward (b = bias or null) and its
+// backward-data (a = scale, b = null), reference scale_layer.cpp
+__global__ void k_chan_affine(const float* __restrict__ x,
+                              const float* __restrict__ a,
+                              const float* __restrict__ b, int C, long S,
+                              long total, float* __restrict__ y) {
+  GRID_STRIDE(i, total) {
+    const int c = (int)((i / S) % C);
+    y[i] = x[i] * a[c] + (b ? b[c] : 0.f);
+  }
+}
+void chan_affine(hipStream_t s, const float* x, const float* a,
+                 const float* b, int N, int C, long S, float* y) {
+  const long total = (long)N * C * S;
+  PerfScope perf(PERF_CLASS("eltwise"), s, 0, 8.0 * total);
+  hipLaunchKernelGGL(k_chan_affine, dim3(nblocks(total, 8)), dim3(TPB), 0,
+                     s, x, a, b, C, S, total, y);
+}
+
 // ------------------------------------------------------------ batchnorm
 // partials layout: double2[C][nb] {sum, sumsq} (fwd) / {sum_dy, sum_dyxn}
 // (bwd).  Deterministic: fixed block→slice mapping, in-block tree reduce.

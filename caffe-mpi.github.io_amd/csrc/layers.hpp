@@ -240,6 +240,36 @@ class BatchNormLayer : public Layer {
   Blob mean_, var_, inv_std_, m_dy_, m_dyxn_, partials_;
 };
 
+// Scale / Bias (reference layers/scale_layer.cpp, bias_layer.cpp): the
+// BVLC-style companion of a stat-only BatchNorm — y = x * scale[c]
+// (+ bias[c]).  Channel axis only (axis 1, num_axes 1), single bottom
+// with learnable params — the pattern every public BVLC-format ResNet
+// prototxt uses; the fork's own models use batch_norm_param.scale_bias
+// instead (csrc BatchNormLayer).
+class ScaleLayer : public Layer {
+ public:
+  using Layer::Layer;
+  void LayerSetUp(const std::vector<Blob*>&,
+                  const std::vector<Blob*>&) override;
+  void Reshape(const std::vector<Blob*>& b,
+               const std::vector<Blob*>& t) override {
+    if (b[0] != t[0]) t[0]->ReshapeLike(*b[0]);
+  }
+  void Forward_cpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override;
+  void Forward_gpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override;
+  void Backward_cpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override;
+  void Backward_gpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override;
+  bool bias_ = false;  // scale_param.bias_term
+  int C_ = 0;
+  Blob partials_;  // per-channel-slice double2 {sum dy*x, sum dy}
+  Blob temp_;      // in-place input copy for backward (scale_layer.cpp)
+  Blob zo_;        // GPU consts+scratch: [zeros C][ones C][scratch 2C]
+};
+
 class ReLULayer : public Layer {
  public:
   using Layer::Layer;

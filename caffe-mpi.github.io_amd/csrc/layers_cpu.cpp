@@ -1087,7 +1087,104 @@ REGISTER_LAYER("DummyData", DataLayer)
 REGISTER_LAYER("Convolution", ConvolutionLayer)
 REGISTER_LAYER("InnerProduct", InnerProductLayer)
 REGISTER_LAYER("Pooling", PoolingLayer)
+// ---------------------------------------------------------------- Scale
+// Reference layers/scale_layer.cpp (+ bias inside, :121-134): channel-wise
+// y = x * scale[c] (+ bias[c]); the BVLC BatchNorm+Scale prototxt pattern.
+// Single-bottom learnable form only (axis 1, num_axes 1 — what every
+// public BVLC-format ResNet uses); the two-bottom form is unsupported.
+void ScaleLayer::LayerSetUp(const std::vector<Blob*>& bottom,
+                            const std::vector<Blob*>&) {
+  CHECK_EQ_((long)bottom.size(), 1)
+      << "Scale: only the single-bottom learnable form is supported";
+  auto sp = param_->sub("scale_param");
+  const long axis = sp ? sp->inum("axis", 1) : 1;
+  const long num_axes = sp ? sp->inum("num_axes", 1) : 1;
+  CHECK_EQ_(axis, 1);
+  CHECK_EQ_(num_axes, 1);
+  bias_ = sp && sp->boolean("bias_term", false);
+  C_ = bottom[0]->channels();
+  if (blobs_.empty()) {
+    Engine& E = Engine::get();
+    blobs_.emplace_back(new Blob({C_}));
+    auto filler = sp ? sp->sub("filler") : nullptr;
+    if (filler) {
+      fill_blob(*blobs_[0], filler, E.cpu_rng);
+    } else {  // reference default: scale = 1 (scale_layer.cpp:39-44)
+      float* p = blobs_[0]->mutable_cpu_data();
+      for (int c = 0; c < C_; ++c) p[c] = 1.f;
+    }
+    if (bias_) {
+      blobs_.emplace_back(new Blob({C_}));
+      fill_blob(*blobs_[1], sp->sub("bias_filler"), E.cpu_rng);
+    }
+  }
+}
+
+void ScaleLayer::Forward_cpu(const std::vector<Blob*>& bottom,
+                             const std::vector<Blob*>& top) {
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  if (bottom[0] == top[0]) {  // in-place: keep x for backward
+    temp_.ReshapeLike(*bottom[0]);
+    memcpy(temp_.mutable_cpu_data(), bottom[0]->cpu_data(),
+           sizeof(float) * bottom[0]->count());
+  }
+  const float* x = bottom[0]->cpu_data();
+  const float* sc = blobs_[0]->cpu_data();
+  const float* bi = bias_ ? blobs_[1]->cpu_data() : nullptr;
+  float* y = top[0]->mutable_cpu_data();
+  for (int n = 0; n < N; ++n)
+    for (int c = 0; c < C_; ++c) {
+      const float s = sc[c], b = bi ? bi[c] : 0.f;
+      const float* xp = x + ((long)n * C_ + c) * S;
+      float* yp = y + ((long)n * C_ + c) * S;
+      for (long i = 0; i < S; ++i) yp[i] = xp[i] * s + b;
+    }
+}
+
+void ScaleLayer::Backward_cpu(const std::vector<Blob*>& top,
+                              const std::vector<bool>& prop_down,
+                              const std::vector<Blob*>& bottom) {
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  const float* dy = top[0]->cpu_diff();
+  const float* x = bottom[0] == top[0] ? temp_.cpu_data()
+                                       : bottom[0]->cpu_data();
+  const float* sc = blobs_[0]->cpu_data();
+  float* dsc = blobs_[0]->mutable_cpu_diff();
+  float* dbi = bias_ ? blobs_[1]->mutable_cpu_diff() : nullptr;
+  // in-place safe: param sums read (x, dy) before dx overwrites
+  std::vector<double> s_dx(C_, 0.0), s_dy(C_, 0.0);
+  for (int n = 0; n < N; ++n)
+    for (int c = 0; c < C_; ++c) {
+      const float* xp = x + ((long)n * C_ + c) * S;
+      const float* dp = dy + ((long)n * C_ + c) * S;
+      double a = 0, b = 0;
+      for (long i = 0; i < S; ++i) {
+        a += (double)dp[i] * xp[i];
+        b += dp[i];
+      }
+      s_dx[c] += a;
+      s_dy[c] += b;
+    }
+  for (int c = 0; c < C_; ++c) {
+    dsc[c] = (float)s_dx[c];
+    if (dbi) dbi[c] = (float)s_dy[c];
+  }
+  if (prop_down[0]) {
+    float* dx = bottom[0]->mutable_cpu_diff();
+    for (int n = 0; n < N; ++n)
+      for (int c = 0; c < C_; ++c) {
+        const float s = sc[c];
+        const float* dp = dy + ((long)n * C_ + c) * S;
+        float* op = dx + ((long)n * C_ + c) * S;
+        for (long i = 0; i < S; ++i) op[i] = dp[i] * s;
+      }
+  }
+}
+
 REGISTER_LAYER("BatchNorm", BatchNormLayer)
+REGISTER_LAYER("Scale", ScaleLayer)
 REGISTER_LAYER("ReLU", ReLULayer)
 REGISTER_LAYER("Eltwise", EltwiseLayer)
 REGISTER_LAYER("LRN", LRNLayer)

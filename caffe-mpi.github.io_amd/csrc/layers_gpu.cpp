@@ -334,6 +334,55 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
                       bottom[0]->mutable_gpu_diff());
 }
 
+// ----------------------------------------------------------------- Scale
+void ScaleLayer::Forward_gpu(const std::vector<Blob*>& bottom,
+                             const std::vector<Blob*>& top) {
+  Engine& E = Engine::get();
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  if (bottom[0] == top[0]) {  // in-place: keep x for backward
+    temp_.ReshapeLike(*bottom[0]);
+    gpu::copy(E.stream, bottom[0]->count(), bottom[0]->gpu_data(),
+              temp_.mutable_gpu_data());
+  }
+  gpu::chan_affine(E.stream, bottom[0]->gpu_data(), blobs_[0]->gpu_data(),
+                   bias_ ? blobs_[1]->gpu_data() : nullptr, N, C_, S,
+                   top[0]->mutable_gpu_data());
+}
+
+void ScaleLayer::Backward_gpu(const std::vector<Blob*>& top,
+                              const std::vector<bool>& prop_down,
+                              const std::vector<Blob*>& bottom) {
+  Engine& E = Engine::get();
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  const float* x = bottom[0] == top[0] ? temp_.gpu_data()
+                                       : bottom[0]->gpu_data();
+  const float* dy = top[0]->gpu_diff();
+  // per-channel {sum dy, sum dy*x} via the BN bwd-stats reduction with
+  // mean 0 / inv_std 1 (xn == x there); finalize writes dscale/dbias and
+  // scratch means we ignore.  zo_ layout: [zeros C][ones C][scratch 2C]
+  if (zo_.count() != 4 * C_) {
+    zo_.Reshape({4 * C_});
+    gpu::set_const(E.stream, C_, 0.f, zo_.mutable_gpu_data());
+    gpu::set_const(E.stream, 3L * C_, 1.f, zo_.mutable_gpu_data() + C_);
+  }
+  const int nb = gpu::bn_blocks_per_channel(N, C_);
+  partials_.Reshape({(int)(C_ * nb * 4)});
+  void* parts = partials_.mutable_gpu_data();
+  float* zo = zo_.mutable_gpu_data();
+  gpu::bn_bwd_stats(E.stream, x, dy, /*mean=*/zo, /*inv_std=*/zo + C_, N,
+                    C_, S, nb, nullptr, nullptr, 0, parts);
+  gpu::bn_bwd_finalize(E.stream, parts, nb, C_, (long)N * S,
+                       blobs_[0]->gpu_data(), 1,
+                       blobs_[0]->mutable_gpu_diff(),
+                       bias_ ? blobs_[1]->mutable_gpu_diff() : zo + 2 * C_,
+                       zo + 2 * C_, zo + 3 * C_);
+  if (prop_down[0])
+    gpu::chan_affine(E.stream, dy, blobs_[0]->gpu_data(), nullptr, N, C_,
+                     S, bottom[0]->mutable_gpu_diff());
+}
+
 // ------------------------------------------------------------------ ReLU
 void ReLULayer::Forward_gpu(const std::vector<Blob*>& bottom,
                             const std::vector<Blob*>& top) {

@@ -96,6 +96,10 @@ void col2im_batched(hipStream_t s, const float* dcol, int Nimg, int C, int H,
 // y += a*x (iter_size diff accumulation over the padded arena)
 void axpy(hipStream_t s, long n, float a, const float* x, float* y);
 
+// y = x * a[c] (+ b[c]) — Scale layer fwd / bwd-data
+void chan_affine(hipStream_t s, const float* x, const float* a,
+                 const float* b, int N, int C, long S, float* y);
+
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y);
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
               float slope, float* dx);
