@@ -270,6 +270,29 @@ class ScaleLayer : public Layer {
   Blob zo_;        // GPU consts+scratch: [zeros C][ones C][scratch 2C]
 };
 
+// Bias (reference layers/bias_layer.cpp): y = x + bias[c], channel axis,
+// single-bottom learnable form — Scale's standalone sibling
+class BiasLayer : public Layer {
+ public:
+  using Layer::Layer;
+  void LayerSetUp(const std::vector<Blob*>&,
+                  const std::vector<Blob*>&) override;
+  void Reshape(const std::vector<Blob*>& b,
+               const std::vector<Blob*>& t) override {
+    if (b[0] != t[0]) t[0]->ReshapeLike(*b[0]);
+  }
+  void Forward_cpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override;
+  void Forward_gpu(const std::vector<Blob*>&,
+                   const std::vector<Blob*>&) override;
+  void Backward_cpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override;
+  void Backward_gpu(const std::vector<Blob*>&, const std::vector<bool>&,
+                    const std::vector<Blob*>&) override;
+  int C_ = 0;
+  Blob partials_, zo_;  // per-channel reduction workspace (GPU)
+};
+
 class ReLULayer : public Layer {
  public:
   using Layer::Layer;

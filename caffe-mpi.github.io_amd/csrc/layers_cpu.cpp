@@ -1183,8 +1183,65 @@ void ScaleLayer::Backward_cpu(const std::vector<Blob*>& top,
   }
 }
 
+// ----------------------------------------------------------------- Bias
+// Reference layers/bias_layer.cpp: y = x + bias[c] (channel axis,
+// learnable single-bottom form)
+void BiasLayer::LayerSetUp(const std::vector<Blob*>& bottom,
+                           const std::vector<Blob*>&) {
+  CHECK_EQ_((long)bottom.size(), 1)
+      << "Bias: only the single-bottom learnable form is supported";
+  auto bp = param_->sub("bias_param");
+  CHECK_EQ_(bp ? bp->inum("axis", 1) : 1, 1);
+  CHECK_EQ_(bp ? bp->inum("num_axes", 1) : 1, 1);
+  C_ = bottom[0]->channels();
+  if (blobs_.empty()) {
+    blobs_.emplace_back(new Blob({C_}));
+    fill_blob(*blobs_[0], bp ? bp->sub("filler") : nullptr,
+              Engine::get().cpu_rng);
+  }
+}
+
+void BiasLayer::Forward_cpu(const std::vector<Blob*>& bottom,
+                            const std::vector<Blob*>& top) {
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  const float* x = bottom[0]->cpu_data();
+  const float* bi = blobs_[0]->cpu_data();
+  float* y = top[0]->mutable_cpu_data();
+  for (int n = 0; n < N; ++n)
+    for (int c = 0; c < C_; ++c) {
+      const float b = bi[c];
+      const float* xp = x + ((long)n * C_ + c) * S;
+      float* yp = y + ((long)n * C_ + c) * S;
+      for (long i = 0; i < S; ++i) yp[i] = xp[i] + b;
+    }
+}
+
+void BiasLayer::Backward_cpu(const std::vector<Blob*>& top,
+                             const std::vector<bool>& prop_down,
+                             const std::vector<Blob*>& bottom) {
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  const float* dy = top[0]->cpu_diff();
+  float* dbi = blobs_[0]->mutable_cpu_diff();
+  for (int c = 0; c < C_; ++c) {
+    double acc = 0;
+    for (int n = 0; n < N; ++n) {
+      const float* dp = dy + ((long)n * C_ + c) * S;
+      for (long i = 0; i < S; ++i) acc += dp[i];
+    }
+    dbi[c] = (float)acc;
+  }
+  if (prop_down[0]) {
+    float* dx = bottom[0]->mutable_cpu_diff();
+    if (dx != dy)
+      memcpy(dx, dy, sizeof(float) * bottom[0]->count());
+  }
+}
+
 REGISTER_LAYER("BatchNorm", BatchNormLayer)
 REGISTER_LAYER("Scale", ScaleLayer)
+REGISTER_LAYER("Bias", BiasLayer)
 REGISTER_LAYER("ReLU", ReLULayer)
 REGISTER_LAYER("Eltwise", EltwiseLayer)
 REGISTER_LAYER("LRN", LRNLayer)

@@ -383,6 +383,38 @@ void ScaleLayer::Backward_gpu(const std::vector<Blob*>& top,
                      S, bottom[0]->mutable_gpu_diff());
 }
 
+// ------------------------------------------------------------------ Bias
+void BiasLayer::Forward_gpu(const std::vector<Blob*>& bottom,
+                            const std::vector<Blob*>& top) {
+  Engine& E = Engine::get();
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  // y = x*1 + b[c]: the ones live in zo_ ([zeros C][ones C][scratch 2C])
+  if (zo_.count() != 4 * C_) {
+    zo_.Reshape({4 * C_});
+    gpu::set_const(E.stream, C_, 0.f, zo_.mutable_gpu_data());
+    gpu::set_const(E.stream, 3L * C_, 1.f, zo_.mutable_gpu_data() + C_);
+  }
+  gpu::chan_affine(E.stream, bottom[0]->gpu_data(),
+                   zo_.mutable_gpu_data() + C_, blobs_[0]->gpu_data(), N,
+                   C_, S, top[0]->mutable_gpu_data());
+}
+
+void BiasLayer::Backward_gpu(const std::vector<Blob*>& top,
+                             const std::vector<bool>& prop_down,
+                             const std::vector<Blob*>& bottom) {
+  Engine& E = Engine::get();
+  const int N = bottom[0]->num();
+  const long S = bottom[0]->count() / ((long)N * C_);
+  const float* dy = top[0]->gpu_diff();
+  // db[c] = sum dy — reuse the deterministic conv bias-grad reduction
+  gpu::bias_grad(E.stream, dy, N, C_, S, blobs_[0]->mutable_gpu_diff());
+  if (prop_down[0]) {
+    float* dx = bottom[0]->mutable_gpu_diff();
+    if (dx != dy) gpu::copy(E.stream, bottom[0]->count(), dy, dx);
+  }
+}
+
 // ------------------------------------------------------------------ ReLU
 void ReLULayer::Forward_gpu(const std::vector<Blob*>& bottom,
                             const std::vector<Blob*>& top) {

@@ -128,6 +128,44 @@ net_param {
     assert any(n == "scale1" for n, _, _ in p.values()), p
 
 
+def test_bias_vs_torch():
+    shape = (2, 4, 5, 5)
+    rng = np.random.default_rng(21)
+    x = rng.standard_normal(shape).astype(np.float32)
+    bi = (0.3 * rng.standard_normal(4)).astype(np.float32)
+    dy = rng.standard_normal(shape).astype(np.float32)
+    xt = torch.tensor(x, requires_grad=True)
+    bt = torch.tensor(bi, requires_grad=True)
+    yt = xt + bt.view(1, -1, 1, 1)
+    yt.backward(torch.tensor(dy))
+    body = """layer { name: "b" type: "Bias" bottom: "in0"
+  top: "out" }"""
+    net, out = run_layer("cpu", [shape], body, [x], params=[bi],
+                         top_diff=dy)
+    assert relerr(out, yt.detach().numpy()) < TOL
+    assert relerr(net.blob("in0", diff=True), xt.grad.numpy()) < TOL
+    assert relerr(net.param(0, diff=True), bt.grad.numpy()) < TOL
+
+
+@pytest.mark.gpu
+def test_bias_gpu_vs_cpu():
+    shape = (2, 5, 4, 4)
+    rng = np.random.default_rng(22)
+    x = rng.standard_normal(shape).astype(np.float32)
+    bi = (0.3 * rng.standard_normal(5)).astype(np.float32)
+    dy = rng.standard_normal(shape).astype(np.float32)
+    body = """layer { name: "b" type: "Bias" bottom: "in0"
+  top: "out" }"""
+    res = {}
+    for mode in ("cpu", "gpu"):
+        net, out = run_layer(mode, [shape], body, [x], params=[bi],
+                             top_diff=dy)
+        res[mode] = (out, net.blob("in0", diff=True),
+                     net.param(0, diff=True))
+    for a, b in zip(res["cpu"], res["gpu"]):
+        assert relerr(b, a) < TOL
+
+
 @pytest.mark.gpu
 def test_scale_gpu_vs_cpu():
     shape = (3, 6, 5, 5)
