@@ -46,6 +46,42 @@ net_param {
 """
 
 
+def test_momentum_history_two_steps():
+    # h_t = m*h_{t-1} + lr*g_t ; w -= h_t  (sgd_solver.cpp:143-252):
+    # with the SAME batch both steps (=> same loss surface point drifting),
+    # replay the recurrence from captured per-step gradients
+    ca.set_mode("cpu")
+    text = SOLVER.replace("momentum: 0.0", "momentum: 0.9").replace(
+        "weight_decay: 0.01", "weight_decay: 0.0")
+    rng = np.random.default_rng(6)
+    x = rng.standard_normal((4, 6)).astype(np.float32)
+    labels = np.array([2, 1, 0, 1], np.float32)
+
+    # gradient probes: run a shadow solver to harvest g1 at w0 and g2 at w1
+    solver = ca.Solver(text=text)
+    net = solver.net
+    net.set_blob("in0", x)
+    net.set_blob("in1", labels)
+    w0 = net.param(0).copy()
+    net.forward()
+    net.backward()
+    g1 = net.param(0, diff=True).copy()
+    solver.step(1)
+    w1 = net.param(0).copy()
+    net.forward()
+    net.backward()
+    g2 = net.param(0, diff=True).copy()
+    solver.step(1)
+    w2 = net.param(0).copy()
+
+    lr_mult = 2.0  # SOLVER's ip weight param
+    h1 = 0.1 * lr_mult * g1
+    assert np.allclose(w1, w0 - h1, rtol=1e-5, atol=1e-6)
+    h2 = 0.9 * h1 + 0.1 * lr_mult * g2
+    assert np.allclose(w2, w1 - h2, rtol=1e-5, atol=1e-6), \
+        np.abs(w2 - (w1 - h2)).max()
+
+
 def test_lr_and_decay_multipliers():
     ca.set_mode("cpu")
     solver = ca.Solver(text=SOLVER)
