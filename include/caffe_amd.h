@@ -49,7 +49,9 @@ void caffe_solver_free(caffe_solver_t s);
  * bucketed all-reduce overlapped on the side stream, fused SGD update */
 int caffe_solver_step(caffe_solver_t s, int iters);
 long caffe_solver_iter(caffe_solver_t s);
-/* smoothed display loss (solver.cpp:606-617 reads the net loss) — syncs */
+/* current net loss (the reference additionally smooths over the display
+ * window, solver.cpp:606-617; this returns the raw last-iteration loss)
+ * — syncs */
 float caffe_solver_loss(caffe_solver_t s);
 caffe_net_t caffe_solver_net(caffe_solver_t s);
 
