@@ -112,7 +112,10 @@ EXPORT void orc_gemm(int transA, int transB, int M, int N, int K, float alpha,
                      const float *A, const float *B, float beta, float *C) {
   const long lda = transA ? M : K; /* leading dim of stored A */
   const long ldb = transB ? K : N;
-#pragma omp parallel for schedule(static)
+  /* collapse(2): conv GEMMs have M as small as 64 (stage-2 Cout) — row-only
+   * parallelism starves a 256-thread host.  Each output element keeps its
+   * serial k-order, so results are bit-identical to the row-parallel form. */
+#pragma omp parallel for collapse(2) schedule(static)
   for (int m = 0; m < M; ++m) {
     for (int n = 0; n < N; ++n) {
       double acc = 0.0;
