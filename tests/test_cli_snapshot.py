@@ -126,3 +126,24 @@ def test_cli_binary_time_and_device_query():
             env=dict(os.environ, CAFFE_SYN_SHAPE="1x28x28x10"))
         assert out.returncode == 0, out.stderr
         assert "TOTAL" in out.stderr
+        # per-layer forward/backward lines for every LeNet layer
+        for lname in ("conv1", "pool1", "ip1", "loss"):
+            assert lname in out.stderr, (lname, out.stderr)
+
+
+def test_cli_usage_and_unknown_command():
+    out = subprocess.run([CAFFE], capture_output=True, text=True,
+                         timeout=60)
+    assert out.returncode == 1
+    assert "usage:" in out.stderr
+    out = subprocess.run([CAFFE, "frobnicate"], capture_output=True,
+                         text=True, timeout=60)
+    assert out.returncode == 1
+    assert "unknown command" in out.stderr
+
+
+def test_cli_train_missing_solver_flag():
+    out = subprocess.run([CAFFE, "train"], capture_output=True, text=True,
+                         timeout=60)
+    assert out.returncode == 1
+    assert "solver" in out.stderr
