@@ -75,6 +75,58 @@ def decode_blob(buf):
     return shape, arr
 
 
+def test_solverstate_independent_decode():
+    # SolverState { iter=1, learned_net=2, history=3 (BlobProto) } —
+    # caffe.proto:303-308, decoded with the same independent walker
+    ca.set_mode("cpu")
+    solver = ca.Solver(text="""base_lr: 0.05
+lr_policy: "fixed"
+momentum: 0.9
+random_seed: 13
+snapshot_prefix: "SNAP"
+net_param {
+  name: "s"
+  layer { name: "input" type: "Input" top: "in0" top: "in1"
+    input_param { shape { dim: 2 dim: 4 } shape { dim: 2 } } }
+  layer { name: "ip" type: "InnerProduct" bottom: "in0" top: "fc"
+    inner_product_param { num_output: 3
+      weight_filler { type: "gaussian" std: 0.2 } } }
+  layer { name: "loss" type: "SoftmaxWithLoss" bottom: "fc"
+    bottom: "in1" top: "loss" }
+}
+""".replace("SNAP", tempfile.mkdtemp() + "/s"))
+    rng = np.random.default_rng(1)
+    solver.net.set_blob("in0", rng.standard_normal((2, 4)).astype(
+        np.float32))
+    solver.net.set_blob("in1", np.array([0, 2], np.float32))
+    solver.step(3)
+    import caffe_amd as ca2
+    assert ca2._lib.caffe_solver_snapshot(solver._h) == 0
+    prefix = solver._h  # path known from text:
+    # find the state file
+    import glob
+    states = glob.glob(os.path.join(tempfile.gettempdir(), "**",
+                                    "s_iter_3.solverstate"),
+                       recursive=True)
+    assert states, "no solverstate written"
+    raw = open(max(states, key=os.path.getmtime), "rb").read()
+    it = None
+    learned = None
+    hist = []
+    for fnum, wt, v in walk(raw):
+        if fnum == 1 and wt == 0:
+            it = v
+        elif fnum == 2 and wt == 2:
+            learned = v.decode()
+        elif fnum == 3 and wt == 2:
+            hist.append(decode_blob(v))
+    assert it == 3
+    assert learned and learned.endswith("s_iter_3.caffemodel")
+    # history blobs mirror the learnable params (ip W and b)
+    assert len(hist) == solver.net.num_params()
+    assert hist[0][1].size in (12, 3)  # W (3x4) or bias depending on order
+
+
 def test_caffemodel_independent_decode():
     ca.set_mode("cpu")
     ca.set_random_seed(41)
