@@ -131,6 +131,30 @@ def test_cli_binary_time_and_device_query():
             assert lname in out.stderr, (lname, out.stderr)
 
 
+def test_cli_finetune_weights_resets_iter():
+    # `caffe train -weights=model` (finetune flow, tools/caffe.cpp:176):
+    # params load from the .caffemodel but the iteration counter starts
+    # at 0 (unlike -snapshot, which restores iter + momentum history)
+    with tempfile.TemporaryDirectory() as tmp:
+        solver = make_lenet_solver(tmp)
+        env = dict(os.environ, CAFFE_SYN_SHAPE="1x28x28x10")
+        out = subprocess.run([CAFFE, "train", f"-solver={solver}"],
+                             capture_output=True, text=True, timeout=600,
+                             env=env)
+        assert out.returncode == 0, out.stderr
+        model = os.path.join(tmp, "lenet_iter_4.caffemodel")
+        assert os.path.exists(model)
+        out = subprocess.run(
+            [CAFFE, "train", f"-solver={solver}", f"-weights={model}",
+             "-iterations=2"],
+            capture_output=True, text=True, timeout=600, env=env)
+        assert out.returncode == 0, out.stderr
+        # finetune run counts from 0: snapshot_after_train writes iter_2
+        assert os.path.exists(os.path.join(tmp,
+                                           "lenet_iter_2.caffemodel")), \
+            out.stderr
+
+
 def test_cli_usage_and_unknown_command():
     out = subprocess.run([CAFFE], capture_output=True, text=True,
                          timeout=60)
