@@ -102,8 +102,7 @@ net_param {
     solver.step(3)
     import caffe_amd as ca2
     assert ca2._lib.caffe_solver_snapshot(solver._h) == 0
-    prefix = solver._h  # path known from text:
-    # find the state file
+    # find the state file (prefix was a tmpdir)
     import glob
     states = glob.glob(os.path.join(tempfile.gettempdir(), "**",
                                     "s_iter_3.solverstate"),
