@@ -71,3 +71,19 @@ def test_malformed_prototxt_raises():
     ca.set_mode("cpu")
     with pytest.raises(Exception):
         net_from_text("layer { name: \"x\" type: ")
+
+
+def test_truncation_fuzz_never_crashes():
+    # every prefix-truncation of a valid net must either parse+build or
+    # raise a clean error through the C ABI — never take the process down
+    ca.set_mode("cpu")
+    full = input_net([(1, 2, 4, 4)], """layer { name: "c"
+  type: "Convolution" bottom: "in0" top: "mid" convolution_param {
+  num_output: 2 kernel_size: 3 pad: 1 } }
+layer { name: "r" type: "ReLU" bottom: "mid" top: "out" }""")
+    rng = np.random.default_rng(5)
+    for cut in sorted(rng.integers(1, len(full), size=40).tolist()):
+        try:
+            net_from_text(full[:cut])
+        except Exception:
+            pass  # clean failure is the contract
