@@ -122,6 +122,25 @@ def test_reference_deploy_inference(deploy, prob):
 
 
 @needs_ref
+def test_reference_lenet_solver_file():
+    # the official MNIST solver verbatim (inv policy, comments, relative
+    # net path; BASELINE configs[0] is the LeNet CPU case)
+    ca.set_mode("cpu")
+    ca.set_synthetic_shape(1, 28, 28, 10)
+    cwd = os.getcwd()
+    os.chdir(REF)
+    try:
+        solver = ca.Solver(
+            path=os.path.join(REF, "examples/mnist/lenet_solver.prototxt"),
+            batch_override=8)
+        solver.step(3)
+        loss = solver.loss()
+    finally:
+        os.chdir(cwd)
+    assert np.isfinite(loss) and loss > 0, loss
+
+
+@needs_ref
 def test_reference_lenet_train_test():
     train_steps(
         os.path.join(REF, "examples/mnist/lenet_train_test.prototxt"),
