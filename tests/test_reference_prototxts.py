@@ -141,6 +141,25 @@ def test_reference_lenet_solver_file():
 
 
 @needs_ref
+def test_reference_cifar10_nv_solver_file():
+    # inline-comment-heavy solver with trailing comments after values
+    ca.set_mode("cpu")
+    ca.set_synthetic_shape(3, 28, 28, 10)
+    cwd = os.getcwd()
+    os.chdir(REF)
+    try:
+        solver = ca.Solver(
+            path=os.path.join(
+                REF, "models/cifar10_nv/cifar10_nv_solver.prototxt"),
+            batch_override=4)
+        solver.step(2)
+        loss = solver.loss()
+    finally:
+        os.chdir(cwd)
+    assert np.isfinite(loss) and loss > 0, loss
+
+
+@needs_ref
 def test_reference_lenet_train_test():
     train_steps(
         os.path.join(REF, "examples/mnist/lenet_train_test.prototxt"),
