@@ -15,7 +15,10 @@ import numpy as np
 import torch
 import torch.nn.functional as F
 
-HERE = os.path.dirname(os.path.abspath(__file__))
+# output dir overridable so tests can verify the committed fixtures are
+# bit-reproducible without touching them
+HERE = os.environ.get("GOLDEN_OUT",
+                      os.path.dirname(os.path.abspath(__file__)))
 rng = np.random.default_rng(1371)
 
 

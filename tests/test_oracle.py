@@ -169,3 +169,23 @@ def test_accuracy():
     lab = np.array([1, 1, 1], np.float32)
     assert abs(orc.accuracy(pred, lab, 3, 2, 1, 1) - 2.0 / 3) < 1e-6
     assert orc.accuracy(pred, lab, 3, 2, 1, 2) == 1.0
+
+
+def test_golden_fixtures_bit_reproducible(tmp_path):
+    # the committed .npz fixtures must regenerate bit-identically from
+    # the committed generator (guards against silent fixture drift)
+    import subprocess
+    import sys
+    env = dict(os.environ, GOLDEN_OUT=str(tmp_path))
+    subprocess.check_call(
+        [sys.executable, os.path.join(GOLD, "gen_golden.py")], env=env,
+        stdout=subprocess.DEVNULL)
+    regen = sorted(p.name for p in tmp_path.glob("*.npz"))
+    committed = sorted(f for f in os.listdir(GOLD) if f.endswith(".npz"))
+    assert regen == committed
+    for name in regen:
+        a = np.load(tmp_path / name)
+        b = np.load(os.path.join(GOLD, name))
+        assert set(a.files) == set(b.files), name
+        for k in a.files:
+            assert np.array_equal(a[k], b[k]), (name, k)
