@@ -826,7 +826,13 @@ void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
 }
 
 // ------------------------------------------------------------ softmax
-// one wave per sample row (C up to 1000): wave-parallel max/sum reductions
+// one wave per sample row (C up to 1000): wave-parallel max/sum reductions.
+// Non-finite inputs PROPAGATE as NaN exactly like the CPU path: a NaN or
+// +inf logit makes the whole row NaN (CPU: exp(NaN-mx)=NaN contaminates the
+// sum), while -inf logits are benign (exp(-inf-mx)=0).  fmaxf silently
+// drops NaN operands and __expf may launder NaN, so the contamination is
+// detected explicitly — a divergence here once masked an all-NaN logits
+// tensor as a clean uniform distribution.
 __global__ void k_softmax_fwd(const float* __restrict__ x, int outer, int C,
                               int inner, float* __restrict__ y) {
   const long rows = (long)outer * inner;
@@ -840,10 +846,16 @@ __global__ void k_softmax_fwd(const float* __restrict__ x, int outer, int C,
     const float* xp = x + (long)o * C * inner + sp;
     float* yp = y + (long)o * C * inner + sp;
     float mx = -3.402823466e38f;
-    for (int c = lane; c < C; c += 64)
-      mx = fmaxf(mx, xp[(long)c * inner]);
-    for (int off = 32; off > 0; off >>= 1)
+    int bad = 0;  // any NaN / +inf in the row
+    for (int c = lane; c < C; c += 64) {
+      const float v = xp[(long)c * inner];
+      mx = fmaxf(mx, v);
+      bad |= (v != v) | (v > 3.402823466e38f);
+    }
+    for (int off = 32; off > 0; off >>= 1) {
       mx = fmaxf(mx, __shfl_xor(mx, off, 64));
+      bad |= __shfl_xor(bad, off, 64);
+    }
     float sum = 0.f;
     for (int c = lane; c < C; c += 64) {
       const float e = __expf(xp[(long)c * inner] - mx);
@@ -852,7 +864,9 @@ __global__ void k_softmax_fwd(const float* __restrict__ x, int outer, int C,
     }
     for (int off = 32; off > 0; off >>= 1) sum += __shfl_xor(sum, off, 64);
     const float inv = 1.f / sum;
-    for (int c = lane; c < C; c += 64) yp[(long)c * inner] *= inv;
+    const float fix = bad ? __builtin_nanf("") : 0.f;
+    for (int c = lane; c < C; c += 64)
+      yp[(long)c * inner] = yp[(long)c * inner] * inv + fix;
   }
 }
 void softmax_fwd(hipStream_t s, const float* x, int outer, int C, int inner,
@@ -875,9 +889,13 @@ __global__ void k_sm_loss(const float* __restrict__ prob,
     const int o = (int)(r / inner);
     const int sp = (int)(r - (long)o * inner);
     const int lv = (int)label[(long)o * inner + sp];
-    const float p =
-        fmaxf(prob[((long)o * C + lv) * inner + sp], 1.175494351e-38f);
-    acc -= (double)__logf(p);
+    const float p = prob[((long)o * C + lv) * inner + sp];
+    // NaN prob must surface as NaN loss (CPU parity): fmaxf(NaN, FLT_MIN)
+    // would clamp it to a clean finite number and hide the divergence
+    if (p != p)
+      acc += (double)p;
+    else
+      acc -= (double)__logf(fmaxf(p, 1.175494351e-38f));
   }
   __shared__ double sh[TPB];
   sh[threadIdx.x] = acc;
@@ -1105,15 +1123,17 @@ void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
 }
 
 // segmented fused SGD: one launch covers a whole bucket's params.
-// segs: sorted arena offsets; per-seg weight base pointer and lr/decay.
+// segs: sorted arena offsets; per-seg weight base pointer and lr/decay
+// MULTIPLIERS (uploaded once — the per-iteration lr/decay ride in as
+// kernel args, so no H2D traffic in the iteration loop).
 // Elements in inter-param padding update harmlessly (their g is 0).
 __global__ void k_sgd_seg(long lo, long hi, float* __restrict__ g_arena,
                           float* __restrict__ h_arena,
                           const long* __restrict__ seg_off,
                           float* const* __restrict__ w_ptrs,
-                          const float* __restrict__ lrs,
-                          const float* __restrict__ decays, int nseg,
-                          float mom, float gscale) {
+                          const float* __restrict__ lr_mults,
+                          const float* __restrict__ decay_mults, int nseg,
+                          float mom, float lr, float decay, float gscale) {
   for (long i = lo + blockIdx.x * (long)blockDim.x + threadIdx.x; i < hi;
        i += (long)gridDim.x * blockDim.x) {
     // binary search: largest k with seg_off[k] <= i
@@ -1126,21 +1146,21 @@ __global__ void k_sgd_seg(long lo, long hi, float* __restrict__ g_arena,
         b = m - 1;
     }
     float* w = w_ptrs[a] + (i - seg_off[a]);
-    float gi = g_arena[i] * gscale + decays[a] * *w;
-    gi = h_arena[i] = mom * h_arena[i] + lrs[a] * gi;
+    float gi = g_arena[i] * gscale + (decay * decay_mults[a]) * *w;
+    gi = h_arena[i] = mom * h_arena[i] + (lr * lr_mults[a]) * gi;
     *w -= gi;
     g_arena[i] = 0.f;
   }
 }
 void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
                           float* h_arena, const long* seg_off,
-                          float* const* w_ptrs, const float* lrs,
-                          const float* decays, int nseg, float mom,
-                          float gscale) {
+                          float* const* w_ptrs, const float* lr_mults,
+                          const float* decay_mults, int nseg, float mom,
+                          float lr, float decay, float gscale) {
   PerfScope perf(PERF_CLASS("sgd"), s, 0, 20.0 * (hi - lo));
   hipLaunchKernelGGL(k_sgd_seg, dim3(nblocks(hi - lo, 4)), dim3(TPB), 0, s,
-                     lo, hi, g_arena, h_arena, seg_off, w_ptrs, lrs, decays,
-                     nseg, mom, gscale);
+                     lo, hi, g_arena, h_arena, seg_off, w_ptrs, lr_mults,
+                     decay_mults, nseg, mom, lr, decay, gscale);
 }
 
 // ------------------------------------------------------------ synthetic

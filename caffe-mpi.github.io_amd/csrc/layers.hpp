@@ -238,6 +238,18 @@ class BatchNormLayer : public Layer {
   int C_ = 0;
   long iter_ = 0;
   Blob mean_, var_, inv_std_, m_dy_, m_dyxn_, partials_;
+  // in-place BN (top==bottom, the standard BVLC prototxt idiom): backward
+  // recomputes x̂ from the ORIGINAL input, which the in-place forward
+  // overwrites — Forward saves a copy into saved_x_ for that case only
+  Blob saved_x_;
+  // the input tensor backward must normalize against (bottom data, or
+  // saved_x_ when in-place) — set by Forward each iteration
+  const float* bwd_x(const std::vector<Blob*>& bottom,
+                     const std::vector<Blob*>& top, bool gpu) {
+    if (top[0] != bottom[0]) return gpu ? bottom[0]->gpu_data()
+                                        : bottom[0]->cpu_data();
+    return gpu ? saved_x_.gpu_data() : saved_x_.cpu_data();
+  }
 };
 
 // Scale / Bias (reference layers/scale_layer.cpp, bias_layer.cpp): the

@@ -540,6 +540,13 @@ void BatchNormLayer::Forward_cpu(const std::vector<Blob*>& bottom,
   const long S = bottom[0]->count() / ((long)N * C_);
   const float* x = bottom[0]->cpu_data();
   float* y = top[0]->mutable_cpu_data();
+  if (top[0] == bottom[0] && phase_ == Phase::TRAIN) {
+    // in-place BN: backward needs the original x (it recomputes x̂) —
+    // save it before the normalize overwrites the blob
+    saved_x_.ReshapeLike(*bottom[0]);
+    memcpy(saved_x_.mutable_cpu_data(), x,
+           sizeof(float) * bottom[0]->count());
+  }
   const float* sc = scale_bias_ ? blobs_[3]->cpu_data() : nullptr;
   const float* bi = scale_bias_ ? blobs_[4]->cpu_data() : nullptr;
   if (phase_ == Phase::TEST) {
@@ -610,7 +617,7 @@ void BatchNormLayer::Backward_cpu(const std::vector<Blob*>& top,
                                   const std::vector<Blob*>& bottom) {
   const int N = bottom[0]->num();
   const long S = bottom[0]->count() / ((long)N * C_);
-  const float* x = bottom[0]->cpu_data();
+  const float* x = bwd_x(bottom, top, false);  // saved copy when in-place
   const float* dy = top[0]->cpu_diff();
   const float* mean = mean_.cpu_data();
   const float* inv = inv_std_.cpu_data();
@@ -1044,11 +1051,16 @@ void AccuracyLayer::Forward_cpu(const std::vector<Blob*>& bottom,
     for (int s = 0; s < inner; ++s) {
       const int lv = (int)label[(long)o * inner + s];
       const float pv = pred[((long)o * C + lv) * inner + s];
+      ++total;
+      // NaN predictions count as INCORRECT: the strictly-greater rank test
+      // is vacuously 0 for NaN and would report accuracy 1.0 on a fully
+      // divergent (all-NaN) net — the reference shares the rank form but an
+      // engine must not present NaN as a perfect score
+      if (pv != pv) continue;
       int rank = 0;
       for (int c = 0; c < C; ++c)
         if (pred[((long)o * C + c) * inner + s] > pv) ++rank;
       if (rank < top_k_) ++correct;
-      ++total;
     }
   top[0]->mutable_cpu_data()[0] = total ? (float)correct / total : 0.f;
 }

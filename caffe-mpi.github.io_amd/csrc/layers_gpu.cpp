@@ -273,6 +273,12 @@ void BatchNormLayer::Forward_gpu(const std::vector<Blob*>& bottom,
   float* y = top[0]->mutable_gpu_data();
   const float* sc = scale_bias_ ? blobs_[3]->gpu_data() : nullptr;
   const float* bi = scale_bias_ ? blobs_[4]->gpu_data() : nullptr;
+  if (top[0] == bottom[0] && phase_ == Phase::TRAIN) {
+    // in-place BN: keep the original x for backward's x̂ recompute
+    saved_x_.ReshapeLike(*bottom[0]);
+    gpu::copy(E.stream, bottom[0]->count(), x, saved_x_.mutable_gpu_data());
+    x = saved_x_.gpu_data();  // stats/norm read the stable copy
+  }
   if (phase_ == Phase::TEST) {
     gpu::bn_fwd_test(E.stream, x, blobs_[0]->gpu_data(),
                      blobs_[1]->gpu_data(), sc, bi, scale_bias_, N, C_, S,
@@ -307,7 +313,7 @@ void BatchNormLayer::Backward_gpu(const std::vector<Blob*>& top,
   Engine& E = Engine::get();
   const int N = bottom[0]->num();
   const long S = bottom[0]->count() / ((long)N * C_);
-  const float* x = bottom[0]->gpu_data();
+  const float* x = bwd_x(bottom, top, true);  // saved copy when in-place
   const float* dy = top[0]->gpu_diff();
   const int nb = gpu::bn_blocks_per_channel(N, C_);
   partials_.Reshape({(int)(C_ * nb * 4)});

@@ -197,9 +197,9 @@ void sgd_update(hipStream_t s, long n, float* g, float* w, float* h,
 // lrs/decays already folded with the per-param multipliers by the caller
 void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
                           float* h_arena, const long* seg_off,
-                          float* const* w_ptrs, const float* lrs,
-                          const float* decays, int nseg, float mom,
-                          float gscale);
+                          float* const* w_ptrs, const float* lr_mults,
+                          const float* decay_mults, int nseg, float mom,
+                          float lr, float decay, float gscale);
 
 void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
                   float lo, float hi, float* y);

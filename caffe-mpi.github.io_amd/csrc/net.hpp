@@ -43,6 +43,21 @@ class Net {
   };
   std::vector<LParam>& learnable_params() { return params_; }
   long learnable_count() const { return arena_count_; }
+  // params_ is arena (reverse-layer) order; the reference serializes SGD
+  // history in FORWARD learnable-param order (sgd_solver.cpp:262-353) —
+  // snapshot/restore must walk this permutation for wire compatibility
+  std::vector<int> forward_param_order() const {
+    std::map<const Layer*, int> pos;
+    for (size_t i = 0; i < layers_.size(); ++i) pos[layers_[i].get()] = (int)i;
+    std::vector<int> idx(params_.size());
+    for (size_t i = 0; i < idx.size(); ++i) idx[i] = (int)i;
+    std::stable_sort(idx.begin(), idx.end(), [&](int a, int b) {
+      const int la = pos.at(params_[a].layer), lb = pos.at(params_[b].layer);
+      if (la != lb) return la < lb;
+      return params_[a].blob_idx < params_[b].blob_idx;
+    });
+    return idx;
+  }
   float* diff_arena() { return diff_arena_; }
   float* data_arena() { return nullptr; }  // weights stay per-blob this round
 

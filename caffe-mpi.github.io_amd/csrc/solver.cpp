@@ -150,40 +150,35 @@ void Reducer::param_ready(int param_id, hipEvent_t done) {
 }
 
 void Solver::ensure_seg_table() {
-  if (d_seg_off_) {
-    // refresh lr/decay rows only when the per-iteration coefficients moved
-    if (table_lr_ == cur_lr_ && table_decay_ == weight_decay_) return;
-  }
+  // the table holds per-param lr/decay MULTIPLIERS (immutable after net
+  // build) — per-iteration lr/decay are kernel arguments, so the table is
+  // uploaded exactly once, synchronously (no pageable-async lifetime
+  // hazard, no H2D copies in the iteration loop)
+  if (d_seg_off_) return;
   Engine& E = Engine::get();
   auto& params = net_->learnable_params();
   const int n = (int)params.size();
   std::vector<long> offs(n);
   std::vector<float*> wps(n);
-  std::vector<float> lrs(n), decays(n);
+  std::vector<float> lrms(n), dcms(n);
   for (int i = 0; i < n; ++i) {
     offs[i] = params[i].offset;
     wps[i] = params[i].blob->mutable_gpu_data();
-    lrs[i] = cur_lr_ * params[i].lr_mult;
-    decays[i] = weight_decay_ * params[i].decay_mult;
+    lrms[i] = params[i].lr_mult;
+    dcms[i] = params[i].decay_mult;
   }
-  if (!d_seg_off_) {
-    d_seg_off_ = (long*)E.dalloc.alloc(n * sizeof(long));
-    d_w_ptrs_ = (float**)E.dalloc.alloc(n * sizeof(float*));
-    d_lrs_ = (float*)E.dalloc.alloc(n * sizeof(float));
-    d_decays_ = (float*)E.dalloc.alloc(n * sizeof(float));
-    HIP_CHECK(hipMemcpyAsync(d_seg_off_, offs.data(), n * sizeof(long),
-                             hipMemcpyHostToDevice, E.comm_stream));
-    HIP_CHECK(hipMemcpyAsync(d_w_ptrs_, wps.data(), n * sizeof(float*),
-                             hipMemcpyHostToDevice, E.comm_stream));
-  }
-  // pageable async H2D stages through a bounce buffer at call time, so the
-  // local vectors need not outlive the call and no stream sync is needed
-  HIP_CHECK(hipMemcpyAsync(d_lrs_, lrs.data(), n * sizeof(float),
-                           hipMemcpyHostToDevice, E.comm_stream));
-  HIP_CHECK(hipMemcpyAsync(d_decays_, decays.data(), n * sizeof(float),
-                           hipMemcpyHostToDevice, E.comm_stream));
-  table_lr_ = cur_lr_;
-  table_decay_ = weight_decay_;
+  d_seg_off_ = (long*)E.dalloc.alloc(n * sizeof(long));
+  d_w_ptrs_ = (float**)E.dalloc.alloc(n * sizeof(float*));
+  d_lrs_ = (float*)E.dalloc.alloc(n * sizeof(float));
+  d_decays_ = (float*)E.dalloc.alloc(n * sizeof(float));
+  HIP_CHECK(hipMemcpy(d_seg_off_, offs.data(), n * sizeof(long),
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(d_w_ptrs_, wps.data(), n * sizeof(float*),
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(d_lrs_, lrms.data(), n * sizeof(float),
+                      hipMemcpyHostToDevice));
+  HIP_CHECK(hipMemcpy(d_decays_, dcms.data(), n * sizeof(float),
+                      hipMemcpyHostToDevice));
 }
 
 void Reducer::flush(hipEvent_t ev) {
@@ -206,8 +201,8 @@ void Reducer::flush(hipEvent_t ev) {
     gpu::sgd_update_segmented(
         E.comm_stream, first.offset, first.offset + count,
         S.net().diff_arena(), S.history(), S.d_seg_off_, S.d_w_ptrs_,
-        S.d_lrs_, S.d_decays_, (int)params.size(), S.cur_mom_,
-        S.grad_scale_);
+        S.d_lrs_, S.d_decays_, (int)params.size(), S.cur_mom_, S.cur_lr_,
+        S.weight_decay_, S.grad_scale_);
   } else {
     if (S.comm_ && S.comm_->world() > 1) {
       // CPU buckets: gather the param diffs into one flat range is
@@ -392,9 +387,16 @@ void Solver::TestAll(long iters) {
     tn->Forward();
     for (auto& kv : tn->scores()) scores[kv.first] += kv.second;
   }
-  for (auto& kv : scores)
+  for (auto& kv : scores) {
+    const double v = kv.second / iters;
     fprintf(stderr, "[caffe_amd] Test net output: %s = %g\n",
-            kv.first.c_str(), kv.second / iters);
+            kv.first.c_str(), v);
+    if (!std::isfinite(v))
+      fprintf(stderr,
+              "[caffe_amd] WARNING: non-finite test score for %s — the net "
+              "is diverging (or untrained BN stats at iter 0)\n",
+              kv.first.c_str());
+  }
 }
 
 void Solver::Snapshot() {
@@ -411,7 +413,12 @@ void Solver::Snapshot() {
   sw.vint(1, iter_);
   sw.str(2, model);
   std::vector<float> host;
-  for (auto& p : net_->learnable_params()) {
+  // history blobs go out in FORWARD learnable-param order — the order the
+  // reference writes (sgd_solver.cpp:262-353); the arena itself is
+  // reverse-layer order, so walk the permutation
+  auto& aparams = net_->learnable_params();
+  for (int pi : net_->forward_param_order()) {
+    const auto& p = aparams[pi];
     wire::Writer bw;
     if (E.mode == Mode::GPU) {
       host.resize(p.count);
@@ -450,20 +457,24 @@ void Solver::Restore(const std::string& path) {
   std::string learned;
   size_t hidx = 0;
   auto& params = net_->learnable_params();
+  // the wire carries history in forward learnable-param order (reference
+  // sgd_solver.cpp:307-353) — map each blob back to its arena slot
+  const std::vector<int> order = net_->forward_param_order();
   while (r.next(&fld)) {
     if (fld.num == 1 && fld.wt == 0) iter_ = (long)fld.vint;
     else if (fld.num == 2 && fld.wt == 2) learned.assign(fld.data, fld.len);
     else if (fld.num == 4 && fld.wt == 0) current_step_ = (int)fld.vint;
     else if (fld.num == 3 && fld.wt == 2) {
       auto b = wire::parse_blob(fld.data, fld.len);
-      CHECK_LT_(hidx, params.size());
-      CHECK_EQ_((long)b.data.size(), params[hidx].count);
+      CHECK_LT_(hidx, order.size());
+      const auto& p = params[order[hidx]];
+      CHECK_EQ_((long)b.data.size(), p.count);
       if (E.mode == Mode::GPU) {
-        HIP_CHECK(hipMemcpy(history_ + params[hidx].offset, b.data.data(),
+        HIP_CHECK(hipMemcpy(history_ + p.offset, b.data.data(),
                             sizeof(float) * b.data.size(),
                             hipMemcpyHostToDevice));
       } else {
-        memcpy(host_history_.data() + params[hidx].offset, b.data.data(),
+        memcpy(host_history_.data() + p.offset, b.data.data(),
                sizeof(float) * b.data.size());
       }
       ++hidx;

@@ -121,10 +121,8 @@ class Solver {
   // flush time by scaling the per-iteration coefficients)
   long* d_seg_off_ = nullptr;
   float** d_w_ptrs_ = nullptr;
-  float* d_lrs_ = nullptr;
-  float* d_decays_ = nullptr;
-  std::vector<float> seg_lr_mult_, seg_decay_mult_;
-  float table_lr_ = -1.f, table_decay_ = -1.f;
+  float* d_lrs_ = nullptr;    // per-param lr MULTIPLIERS (uploaded once)
+  float* d_decays_ = nullptr; // per-param decay multipliers
   void ensure_seg_table();
   // iter_size accumulation buffer (diff-arena layout; only allocated when
   // iter_size > 1 — layer backwards overwrite their param diffs, so cross-
