@@ -41,6 +41,8 @@ struct GemmArgs {
   const float* bias;
   int bias_per_col;
   int relu;
+  long Srow;               // dest per-row stride (== S unless strided)
+  int OWo, osh, osw, Wd;   // strided scatter (OWo>0)
   // operand views (spad==0 => plain)
   GemmView av, bv;
   // split-K
@@ -62,7 +64,7 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
   const float* p;
   long nvalid;  // elements valid from j=0
   if (v.spad && v.kh > 0) {
-    // implicit im2col (s1/d1): r = (c, ki, kj), sp = (oh, ow)
+    // implicit im2col (d1, any stride): r = (c, ki, kj), sp = (oh, ow)
     const long n = q / v.spad;
     const long sp = q - n * v.spad;
     if (r >= rmax || sp >= v.S || q >= qmax) {
@@ -75,27 +77,33 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
     const int ki = krem / v.kw, kj = krem - (krem / v.kw) * v.kw;
     int oh = (int)(sp / v.OW);
     int ow = (int)(sp - (long)oh * v.OW);
-    int h = oh - v.ph + ki;
-    int w = ow - v.pw + kj;
+    int h = oh * v.sh - v.ph + ki;
+    int w = ow * v.sw - v.pw + kj;
     const float* xp = P + (n * v.chan + c) * (long)v.H * v.W;
     const long smax = v.S - sp;  // elements left in this image
     // fast path: chunk stays in one output row, fully interior
     if (smax >= 16 && ow + 16 <= v.OW && h >= 0 && h < v.H && w >= 0 &&
-        w + 15 < v.W) {
-      const float* p = xp + h * v.W + w;
-      if ((((uintptr_t)p) & 15) == 0) {
-        const f32x4* p4 = (const f32x4*)p;
+        w + 15 * v.sw < v.W) {
+      if (v.sw == 1) {
+        const float* p = xp + h * v.W + w;
+        if ((((uintptr_t)p) & 15) == 0) {
+          const f32x4* p4 = (const f32x4*)p;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const f32x4 t = p4[j];
-          out[4 * j + 0] = t.x;
-          out[4 * j + 1] = t.y;
-          out[4 * j + 2] = t.z;
-          out[4 * j + 3] = t.w;
+          for (int j = 0; j < 4; ++j) {
+            const f32x4 t = p4[j];
+            out[4 * j + 0] = t.x;
+            out[4 * j + 1] = t.y;
+            out[4 * j + 2] = t.z;
+            out[4 * j + 3] = t.w;
+          }
+        } else {
+#pragma unroll
+          for (int j = 0; j < 16; ++j) out[j] = p[j];
         }
-      } else {
+      } else {  // strided row gather: 16 independent loads
+        const float* p = xp + h * v.W + w;
 #pragma unroll
-        for (int j = 0; j < 16; ++j) out[j] = p[j];
+        for (int j = 0; j < 16; ++j) out[j] = p[j * v.sw];
       }
       return;
     }
@@ -108,8 +116,9 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const bool wrapped = j >= jw;
-        const int hj = h + (wrapped ? 1 : 0);
-        const int wj = wrapped ? wreset + (j - jw) : w0 + j;
+        const int hj = h + (wrapped ? v.sh : 0);
+        const int wj = (wrapped ? wreset + (j - jw) * v.sw
+                                : w0 + j * v.sw);
         const bool ok =
             j < smax && hj >= 0 && hj < v.H && wj >= 0 && wj < v.W;
         out[j] = ok ? xp[hj * v.W + wj] : 0.f;
@@ -120,14 +129,14 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
     for (int j = 0; j < 16; ++j) {
       const bool ok = j < smax && h >= 0 && h < v.H && w >= 0 && w < v.W;
       out[j] = ok ? xp[h * v.W + w] : 0.f;
-      // next output pixel (stride 1): ow+1, wrapping to the next row
+      // next output pixel: ow+1 (input w += sw), wrapping to the next row
       if (++ow == v.OW) {
         ow = 0;
         ++oh;
-        ++h;
+        h += v.sh;
         w = -v.pw + kj;
       } else {
-        ++w;
+        w += v.sw;
       }
     }
     return;
@@ -415,13 +424,19 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
     // col is constant across the 16 regs — hoist the scatter division
     const long col = n0 + wc * 64 + tj * 32 + row_in;
     if (col >= g.N) return;
-    long col_base = 0;  // n*n_stride + sp for scatter, col for plain
+    long col_base = 0;  // n*n_stride + pix for scatter, col for plain
     bool col_ok = true;
     if (!SPLITK && g.spad > 0) {
       const long n = col / g.spad;
       const long sp = col - n * g.spad;
       col_ok = sp < g.S;
-      col_base = n * g.n_stride + sp;
+      long pix = sp;
+      if (g.OWo > 0) {  // strided scatter (1x1/s2 dgrad)
+        const int oh = (int)(sp / g.OWo);
+        const int ow = (int)(sp - (long)oh * g.OWo);
+        pix = ((long)oh * g.osh) * g.Wd + (long)ow * g.osw;
+      }
+      col_base = n * g.n_stride + pix;
     }
     if (!col_ok) return;
     const float cbias =
@@ -438,7 +453,7 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
       }
       if (g.bias) v += g.bias_per_col ? cbias : g.bias[row];
       if (g.relu) v = fmaxf(v, 0.f);
-      const long off = g.spad > 0 ? col_base + row * g.S
+      const long off = g.spad > 0 ? col_base + row * g.Srow
                                   : row * g.ldc + col;
       if (g.beta != 0.f) v += g.beta * g.C[off];
       g.C[off] = v;
@@ -524,6 +539,11 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     g.bias = epi->bias;
     g.bias_per_col = epi->bias_per_col;
     g.relu = epi->relu;
+    g.Srow = epi->Srow > 0 ? epi->Srow : epi->S;
+    g.OWo = epi->OWo;
+    g.osh = epi->osh;
+    g.osw = epi->osw;
+    g.Wd = epi->Wd;
   }
   if (aview) g.av = *aview;
   if (bview) g.bv = *bview;
@@ -651,12 +671,21 @@ __global__ void k_bias_grad_part(const float* __restrict__ dy, int N, int C,
   const int n0 = (int)((long)N * slice / nb);
   const int n1 = (int)((long)N * (slice + 1) / nb);
   const int span = (n1 - n0) * S;
-  double acc = 0;
-  for (int i = threadIdx.x; i < span; i += blockDim.x) {
+  const int B = blockDim.x;
+  auto addr = [&](int i) {
     const int n = n0 + i / S;
-    const int sp = i - (n - n0) * S;
-    acc += dy[((long)n * C + c) * S + sp];
+    return ((long)n * C + c) * S + (i - (n - n0) * S);
+  };
+  double acc = 0, acc2 = 0;
+  int i = threadIdx.x;
+  for (; i + 3 * B < span; i += 4 * B) {  // 4 loads in flight (MLP)
+    const float v0 = dy[addr(i)], v1 = dy[addr(i + B)];
+    const float v2 = dy[addr(i + 2 * B)], v3 = dy[addr(i + 3 * B)];
+    acc += (double)v0 + v1;
+    acc2 += (double)v2 + v3;
   }
+  for (; i < span; i += B) acc += dy[addr(i)];
+  acc += acc2;
   __shared__ double sh[256];
   sh[threadIdx.x] = acc;
   __syncthreads();

@@ -40,13 +40,22 @@ static bool conv_s1d1(const ConvolutionLayer& l) {
 }
 // implicit-im2col staging walks pixel chunks of 16; below OW 24 nearly
 // every chunk crosses an output row (slow masked path) — the explicit col
-// buffer wins there (measured: stage-4 3x3 at OW=14 ran at 68 vs ~90 TF)
+// buffer wins there (measured: stage-4 3x3 at OW=14 ran at 68 vs ~90 TF).
+// Round 2: strides are supported by the view (conv1 7x7s2, the 1x1/s2
+// projection shortcuts, AlexNet 11x11/s4 — kills the im2col/col2im
+// passes); dilation stays on the explicit path.
 static bool conv_implicit(const ConvolutionLayer& l) {
   static const int min_ow = [] {
     const char* e = getenv("CAFFE_IMPLICIT_MIN_OW");
     return e ? atoi(e) : 24;
   }();
-  return conv_s1d1(l) && (l.kh_ == 1 || l.OW_ >= min_ow);
+  if (l.dh_ != 1 || l.dw_ != 1) return false;
+  static const bool strided_ok = [] {
+    const char* e = getenv("CAFFE_IMPLICIT_STRIDED");
+    return e ? atoi(e) != 0 : true;
+  }();
+  if ((l.sh_ != 1 || l.sw_ != 1) && !strided_ok) return false;
+  return l.kh_ == 1 || l.OW_ >= min_ow;
 }
 static bool conv_is_1x1(const ConvolutionLayer& l) {
   return l.kh_ == 1 && l.kw_ == 1 && conv_s1d1(l) && !l.ph_ && !l.pw_ &&
@@ -75,8 +84,8 @@ void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
               S_, &epi, nullptr, &xv);
     return;
   }
-  if (conv_implicit(*this)) {  // implicit im2col view
-    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_};
+  if (conv_implicit(*this)) {  // implicit im2col view (strides in-view)
+    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_, sh_, sw_};
     for (int g = 0; g < group_; ++g) {
       epi.bias = bias_ ? blobs_[1]->gpu_data() + (long)g * (Cout_ / group_)
                        : nullptr;
@@ -139,8 +148,8 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
   }
 
   if (conv_implicit(*this)) {
-    // wgrad: dW = dY-view · (implicit col of x)ᵀ
-    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_};
+    // wgrad: dW = dY-view · (implicit col of x)ᵀ — strides live in the view
+    GemmView xv{Spad_, S_, C_, kh_, kw_, ph_, pw_, H_, W_, OW_, sh_, sw_};
     for (int g = 0; g < group_; ++g)
       gpu::gemm(E.stream, false, true, Cout_ / group_, K, NS, 1.f,
                 dy + (long)g * (Cout_ / group_) * S_, 0,
@@ -149,31 +158,68 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
                     (long)g * (Cout_ / group_) * K,
                 K, nullptr, &dyv, &xv);
     if (prop_down[0]) {
-      // dgrad = forward conv of dY with flipped/transposed weights:
-      // dx[ci] = Σ_{co,ki,kj} dy[co][h-ki+ (kh-1-ph) ...] · Wt
-      float* wt =
-          (float*)ws.get(11, sizeof(float) * blobs_[0]->count());
-      gpu::weight_flip_grouped(E.stream, w, Cout_, C_ / group_, kh_, kw_,
-                               group_, wt);
-      const long S_in = (long)H_ * W_;
-      const long spad_in = (S_in + 15) / 16 * 16;
-      const long NSin = (long)N_ * spad_in;
-      const int Kd = Cout_ / group_ * kh_ * kw_;
-      GemmEpi epi;
-      epi.spad = spad_in;
-      epi.S = S_in;
-      epi.n_stride = (long)C_ * S_in;
-      // dY viewed with the transposed-conv geometry: input dims (OH,OW),
-      // pad (k-1-p), output dims (H,W)
-      GemmView dyc{spad_in, S_in, Cout_, kh_, kw_, kh_ - 1 - ph_,
-                   kw_ - 1 - pw_, OH_, OW_, W_};
-      for (int g = 0; g < group_; ++g)
-        gpu::gemm(E.stream, false, false, C_ / group_, NSin, Kd, 1.f,
-                  wt + (long)g * (C_ / group_) * Kd, Kd,
-                  dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
-                  bottom[0]->mutable_gpu_diff() +
-                      (long)g * (C_ / group_) * S_in,
-                  S_in, &epi, nullptr, &dyc);
+      if (conv_s1d1(*this)) {
+        // dgrad = forward conv of dY with flipped/transposed weights:
+        // dx[ci] = Σ_{co,ki,kj} dy[co][h-ki+ (kh-1-ph) ...] · Wt
+        float* wt =
+            (float*)ws.get(11, sizeof(float) * blobs_[0]->count());
+        gpu::weight_flip_grouped(E.stream, w, Cout_, C_ / group_, kh_, kw_,
+                                 group_, wt);
+        const long S_in = (long)H_ * W_;
+        const long spad_in = (S_in + 15) / 16 * 16;
+        const long NSin = (long)N_ * spad_in;
+        const int Kd = Cout_ / group_ * kh_ * kw_;
+        GemmEpi epi;
+        epi.spad = spad_in;
+        epi.S = S_in;
+        epi.n_stride = (long)C_ * S_in;
+        // dY viewed with the transposed-conv geometry: input dims (OH,OW),
+        // pad (k-1-p), output dims (H,W)
+        GemmView dyc{spad_in, S_in, Cout_, kh_, kw_, kh_ - 1 - ph_,
+                     kw_ - 1 - pw_, OH_, OW_, W_};
+        for (int g = 0; g < group_; ++g)
+          gpu::gemm(E.stream, false, false, C_ / group_, NSin, Kd, 1.f,
+                    wt + (long)g * (C_ / group_) * Kd, Kd,
+                    dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+                    bottom[0]->mutable_gpu_diff() +
+                        (long)g * (C_ / group_) * S_in,
+                    S_in, &epi, nullptr, &dyc);
+      } else if (kh_ == 1 && kw_ == 1 && ph_ == 0 && pw_ == 0) {
+        // strided 1x1 (projection shortcuts): dgrad touches only the
+        // sampled pixels — zero dx, then a GEMM whose epilogue scatters
+        // column (oh, ow) to input pixel (oh*sh, ow*sw).  No dcol
+        // round-trip and no col2im kernel.
+        float* dx = bottom[0]->mutable_gpu_diff();
+        gpu::set_const(E.stream, bottom[0]->count(), 0.f, dx);
+        GemmEpi epi;
+        epi.spad = Spad_;
+        epi.S = S_;
+        epi.n_stride = (long)C_ * H_ * W_;
+        epi.Srow = (long)H_ * W_;
+        epi.OWo = OW_;
+        epi.osh = sh_;
+        epi.osw = sw_;
+        epi.Wd = W_;
+        for (int g = 0; g < group_; ++g)
+          gpu::gemm(E.stream, true, false, C_ / group_, NS, Cout_ / group_,
+                    1.f, w + (long)g * (Cout_ / group_) * K, K,
+                    dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+                    dx + (long)g * (C_ / group_) * H_ * W_, 0, &epi,
+                    nullptr, &dyv);
+      } else {
+        // strided kh>1 (conv1-style — only reached when prop_down, which
+        // the first layer never is): dcol = Wᵀ·dY, then gather col2im
+        float* dcol =
+            (float*)ws.get(1, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
+        for (int g = 0; g < group_; ++g)
+          gpu::gemm(E.stream, true, false, K, NS, Cout_ / group_, 1.f,
+                    w + (long)g * (Cout_ / group_) * K, K,
+                    dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+                    dcol + (long)g * K * NS, NS, nullptr, nullptr, &dyv);
+        gpu::col2im_batched(E.stream, dcol, N_, C_, H_, W_, kh_, kw_, ph_,
+                            pw_, sh_, sw_, dh_, dw_, OH_, OW_, Spad_,
+                            bottom[0]->mutable_gpu_diff());
+      }
     }
     return;
   }

@@ -46,6 +46,11 @@ struct GemmEpi {
   const float* bias = nullptr;
   bool bias_per_col = false;  // IP: bias indexed by column, conv: by row
   bool relu = false;
+  // strided scatter (1x1/s2 dgrad without a col2im pass): column pixel
+  // sp -> (oh, ow) over OWo maps to destination pixel oh*osh*Wd + ow*osw;
+  // per-row stride is Srow (= dest H*W).  OWo == 0 -> plain NCHW scatter.
+  long Srow = 0;
+  int OWo = 0, osh = 1, osw = 1, Wd = 0;
 };
 
 // NCHW-view operand: the GEMM axis that is contiguous in the stored tensor
@@ -65,6 +70,8 @@ struct GemmView {
   long chan = 0;  // channels of the viewed tensor (address stride term)
   int kh = 0, kw = 0, ph = 0, pw = 0;  // kh>0 => implicit im2col
   int H = 0, W = 0, OW = 0;            // input dims / output row width
+  int sh = 1, sw = 1;  // conv strides (round 2: strided implicit im2col —
+                       // conv1 7x7s2, the 1x1s2 projections, AlexNet s4)
 };
 
 // Wt[ci][co*kh*kw + ki*kw + kj] = W[co][ci][kh-1-ki][kw-1-kj] — the

@@ -435,6 +435,17 @@ ALL_CHECKS = {
                                               k=7, s=1, p=3),
     "conv_11x11s4": lambda m: check_conv(m, N=2, C=3, H=47, W=47, Co=8,
                                          k=11, s=4, p=0),  # AlexNet conv1
+    # round 2: STRIDED implicit-im2col views (strides folded into the GEMM
+    # staging walk — no im2col/col2im kernels).  Shapes chosen so OW >= 24
+    # (kh>1) / OW >= 16 (one-wrap strided walk) to hit each fast path:
+    # conv1-style 7x7s2 (strided interior gather + row wraps + dcol-free
+    # wgrad), 1x1s2 projection (zero+scatter dgrad), 11x11s4 wide.
+    "conv_7x7s2_implicit": lambda m: check_conv(m, N=2, C=3, H=55, W=55,
+                                                Co=8, k=7, s=2, p=3),
+    "conv_1x1s2_wide": lambda m: check_conv(m, N=2, C=6, H=40, W=40, Co=8,
+                                            k=1, s=2, p=0, bias=False),
+    "conv_11x11s4_wide": lambda m: check_conv(m, N=2, C=3, H=79, W=79, Co=8,
+                                              k=11, s=4, p=0),
     "ip": check_ip,
     "ip_large": lambda m: check_ip(m, M=130, K=260, Nout=140),
     "pool_max": lambda m: check_pool(m, "MAX"),
