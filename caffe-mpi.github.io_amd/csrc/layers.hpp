@@ -157,6 +157,9 @@ class ConvolutionLayer : public Layer {
                     const std::vector<Blob*>&) override;
   void Backward_gpu(const std::vector<Blob*>&, const std::vector<bool>&,
                     const std::vector<Blob*>&) override;
+  // strided 1x1 dgrad without a col2im pass (zero + strided-scatter GEMM)
+  void scatter_dgrad(const std::vector<Blob*>& top,
+                     const std::vector<Blob*>& bottom);
 
   int Cout_ = 0, kh_ = 0, kw_ = 0, sh_ = 1, sw_ = 1, ph_ = 0, pw_ = 0,
       dh_ = 1, dw_ = 1, group_ = 1;

@@ -41,25 +41,65 @@ static bool conv_s1d1(const ConvolutionLayer& l) {
 // implicit-im2col staging walks pixel chunks of 16; below OW 24 nearly
 // every chunk crosses an output row (slow masked path) — the explicit col
 // buffer wins there (measured: stage-4 3x3 at OW=14 ran at 68 vs ~90 TF).
-// Round 2: strides are supported by the view (conv1 7x7s2, the 1x1/s2
-// projection shortcuts, AlexNet 11x11/s4 — kills the im2col/col2im
-// passes); dilation stays on the explicit path.
+// Round 2: strides are supported by the view, but ONLY for conv1-like
+// shapes (kh>1, Cout/group <= 128 so the x view is staged once — a
+// strided view re-read per output tile-row pays 2x bytes per pass, which
+// measured SLOWER than the explicit col for the large-Cout 1x1/s2
+// projections).  Strided 1x1 keeps the explicit col for fwd/wgrad and a
+// zero+scatter dgrad (no dcol GEMM, no col2im).  Dilation stays explicit.
 static bool conv_implicit(const ConvolutionLayer& l) {
   static const int min_ow = [] {
     const char* e = getenv("CAFFE_IMPLICIT_MIN_OW");
     return e ? atoi(e) : 24;
   }();
   if (l.dh_ != 1 || l.dw_ != 1) return false;
+  if (l.sh_ == 1 && l.sw_ == 1) return l.kh_ == 1 || l.OW_ >= min_ow;
   static const bool strided_ok = [] {
     const char* e = getenv("CAFFE_IMPLICIT_STRIDED");
     return e ? atoi(e) != 0 : true;
   }();
-  if ((l.sh_ != 1 || l.sw_ != 1) && !strided_ok) return false;
-  return l.kh_ == 1 || l.OW_ >= min_ow;
+  return strided_ok && l.kh_ > 1 && l.OW_ >= min_ow &&
+         l.Cout_ / l.group_ <= 128;
+}
+// strided 1x1 pad-0 dgrad can always scatter (writes only the sampled
+// input pixels after a zero fill) — independent of the fwd/wgrad path
+static bool conv_scatter_dgrad(const ConvolutionLayer& l) {
+  return (l.sh_ > 1 || l.sw_ > 1) && l.kh_ == 1 && l.kw_ == 1 &&
+         l.ph_ == 0 && l.pw_ == 0 && l.dh_ == 1 && l.dw_ == 1;
 }
 static bool conv_is_1x1(const ConvolutionLayer& l) {
   return l.kh_ == 1 && l.kw_ == 1 && conv_s1d1(l) && !l.ph_ && !l.pw_ &&
          l.group_ == 1;
+}
+
+// strided 1x1 (projection shortcuts) dgrad: dgrad touches only the sampled
+// input pixels — zero dx, then a GEMM whose epilogue scatters column
+// (oh, ow) to input pixel (oh*sh, ow*sw).  No dcol round-trip, no col2im.
+void ConvolutionLayer::scatter_dgrad(const std::vector<Blob*>& top,
+                                     const std::vector<Blob*>& bottom) {
+  Engine& E = Engine::get();
+  const int K = C_ / group_ * kh_ * kw_;
+  const long NS = (long)N_ * Spad_;
+  const float* w = blobs_[0]->gpu_data();
+  const float* dy = top[0]->gpu_diff();
+  GemmView dyv{Spad_, S_, Cout_};
+  float* dx = bottom[0]->mutable_gpu_diff();
+  gpu::set_const(E.stream, bottom[0]->count(), 0.f, dx);
+  GemmEpi epi;
+  epi.spad = Spad_;
+  epi.S = S_;
+  epi.n_stride = (long)C_ * H_ * W_;
+  epi.Srow = (long)H_ * W_;
+  epi.OWo = OW_;
+  epi.osh = sh_;
+  epi.osw = sw_;
+  epi.Wd = W_;
+  for (int g = 0; g < group_; ++g)
+    gpu::gemm(E.stream, true, false, C_ / group_, NS, Cout_ / group_, 1.f,
+              w + (long)g * (Cout_ / group_) * K, K,
+              dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
+              dx + (long)g * (C_ / group_) * H_ * W_, 0, &epi, nullptr,
+              &dyv);
 }
 
 void ConvolutionLayer::Forward_gpu(const std::vector<Blob*>& bottom,
@@ -184,28 +224,8 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
                     bottom[0]->mutable_gpu_diff() +
                         (long)g * (C_ / group_) * S_in,
                     S_in, &epi, nullptr, &dyc);
-      } else if (kh_ == 1 && kw_ == 1 && ph_ == 0 && pw_ == 0) {
-        // strided 1x1 (projection shortcuts): dgrad touches only the
-        // sampled pixels — zero dx, then a GEMM whose epilogue scatters
-        // column (oh, ow) to input pixel (oh*sh, ow*sw).  No dcol
-        // round-trip and no col2im kernel.
-        float* dx = bottom[0]->mutable_gpu_diff();
-        gpu::set_const(E.stream, bottom[0]->count(), 0.f, dx);
-        GemmEpi epi;
-        epi.spad = Spad_;
-        epi.S = S_;
-        epi.n_stride = (long)C_ * H_ * W_;
-        epi.Srow = (long)H_ * W_;
-        epi.OWo = OW_;
-        epi.osh = sh_;
-        epi.osw = sw_;
-        epi.Wd = W_;
-        for (int g = 0; g < group_; ++g)
-          gpu::gemm(E.stream, true, false, C_ / group_, NS, Cout_ / group_,
-                    1.f, w + (long)g * (Cout_ / group_) * K, K,
-                    dy + (long)g * (Cout_ / group_) * S_, 0, 0.f,
-                    dx + (long)g * (C_ / group_) * H_ * W_, 0, &epi,
-                    nullptr, &dyv);
+      } else if (conv_scatter_dgrad(*this)) {
+        scatter_dgrad(top, bottom);
       } else {
         // strided kh>1 (conv1-style — only reached when prop_down, which
         // the first layer never is): dcol = Wᵀ·dY, then gather col2im
@@ -236,6 +256,10 @@ void ConvolutionLayer::Backward_gpu(const std::vector<Blob*>& top,
                   (long)g * (Cout_ / group_) * K,
               K, nullptr, &dyv);
   if (prop_down[0]) {
+    if (conv_scatter_dgrad(*this)) {  // strided 1x1: no dcol/col2im
+      scatter_dgrad(top, bottom);
+      return;
+    }
     float* dcol =
         (float*)ws.get(1, sizeof(float) * (size_t)C_ * kh_ * kw_ * NS);
     for (int g = 0; g < group_; ++g)
