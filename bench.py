@@ -293,13 +293,32 @@ def main():
                 g["ns"] += v["ns"]
                 g["launches"] += v["launches"]
         achieved = g["flops"] / max(g["ns"], 1) * 1e9  # FLOP/s
+        # measured HBM bytes per GEMM-class launch, from the committed PMC
+        # capture (tools/pmc_traffic.py, FETCH_SIZE x2-corrected +
+        # WRITE_SIZE per MI355X_MICROARCH.md §HBM); null if not captured
+        traffic = None
+        pmc_path = os.path.join(REPO, "profiles",
+                                f"pmc_traffic_{args.model}.json")
+        if os.path.exists(pmc_path):
+            try:
+                pmc = json.load(open(pmc_path))
+                gc = pmc["classes"].get("gemm")
+                if gc and gc.get("launches"):
+                    traffic = ((gc["fetch_bytes"] + gc["write_bytes"])
+                               / gc["launches"])
+            except Exception:
+                traffic = None
         roofline = {
             "bound": "mfma",
             "achieved": achieved,
             "peak": PEAK_F32_MFMA,
             "unit": "TFLOP/s",
             "frac": achieved / PEAK_F32_MFMA,
-            "traffic": None,
+            "traffic": traffic,
+            "traffic_note": ("HBM bytes/launch of the GEMM class, "
+                            "profiles/pmc_traffic_*.json "
+                            "(FETCH_SIZE x2 + WRITE_SIZE)")
+                           if traffic is not None else None,
             "kernel": "gemm_f32 (conv/IP contractions)",
             "kernel_time_frac": None,
         }
