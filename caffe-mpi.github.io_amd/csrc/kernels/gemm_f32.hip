@@ -465,6 +465,258 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
   epi_tile(acc11, 1, 1);
 }
 
+// ==================================================================
+// k_gemm2 — glds (global_load_lds) staged fp32 MFMA GEMM, BK=64.
+//
+// Round-2 load-path redesign (cdna_hip_programming.md "glds vs register
+// staging"): the v1 kernel is register-staged at 223 VGPRs / 2 waves per
+// SIMD with ~50% issue stall; glds removes the ~32 staging VGPRs and 3/4
+// of the LDS instructions, and BK=64 halves the barrier count.  LDS
+// images are LINEAR in the exact lane order glds writes (wave-uniform
+// base + lane*16B — the swizzle lives in the per-lane GLOBAL source
+// address, guide rule 21):
+//   A (TA=false, A[M][K] row-major): m-major [128][64], element (m,k) at
+//     word m*64 + ((k/4 ^ (m&15))*4 + k%4) — the XOR spread makes the
+//     MFMA column reads (consecutive m at fixed k) conflict-free;
+//   A (TA=true, A[K][M]): k-major [64][128] linear (source contiguous
+//     along m — no swizzle needed, rows of consecutive words);
+//   B (B[K][N] or channel view): k-major [64][128] linear.
+// 2 LDS buffers (2*(32+32) KB = 128 KB, 1 block/CU), glds for tile t+1
+// issued before the MFMA loop on tile t, plain __syncthreads() per tile
+// (its implicit vmcnt(0) drains the DMA).  The K-remainder tile is
+// register-staged through read16 (zero-filled — glds cannot zero).
+// Operand scope: plain or channel-view (kh==0) only; implicit-im2col
+// views (kh>0) and TB=true stay on k_gemm_f32 (their contracted axis
+// carries per-image zero padding glds cannot inject).
+// Out-of-range lanes are exec-masked off: their LDS slots keep garbage,
+// which only ever lands in output rows/cols the epilogue drops.
+constexpr int BK2 = 64;
+
+template <bool TA>
+__launch_bounds__(256, 1) __global__ void k_gemm2(GemmArgs g) {
+  extern __shared__ float smem2[];  // [2][A 128*64] [2][B 64*128]
+  auto Aimg = [&](int b) -> float* { return smem2 + b * (128 * BK2); };
+  auto Bimg = [&](int b) -> float* {
+    return smem2 + (2 + b) * (128 * BK2);
+  };
+
+  long flat = blockIdx.x;
+  {
+    const long nwg = g.tiles;
+    const long q = nwg / 8, rr = nwg % 8;
+    const long xcd = flat % 8, idx = flat / 8;
+    flat = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const long tile_m = flat / g.tn;
+  const long tile_n = flat - tile_m * g.tn;
+  const long m0 = tile_m * BM, n0 = tile_n * BN;
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  const int wr = wave & 1;
+  const int wc = (wave >> 1) & 1;
+  const int row_in = lane & 31;
+  const int ksel = lane >> 5;
+
+  const long ntiles = (g.K + BK2 - 1) / BK2;
+  const long nfull = g.K / BK2;  // glds-able full tiles
+
+  // ---- per-lane staging geometry (fixed across K-tiles)
+  // B: wave w fills k-rows [16w,16w+16), 8 instrs x 2 rows; lane: k-row
+  // 16w+2i+lane/32, 4-float chunk at q = n0 + 4*(lane%32)
+  const long bq = n0 + 4 * (lane & 31);
+  const float* bbase;
+  long bstride;
+  if (g.bv.spad) {  // channel view: q -> (n_img, sp), row stride S
+    const long n_img = bq / g.bv.spad;
+    const long sp = bq - n_img * g.bv.spad;
+    bbase = g.B + n_img * g.bv.chan * g.bv.S + sp;
+    bstride = g.bv.S;
+  } else {
+    bbase = g.B + bq;
+    bstride = g.ldb;
+  }
+  // A (TA=false): wave w fills m-rows [32w,32w+32), 8 instrs x 4 rows;
+  // lane: m-row 32w+4i+lane/16, k-chunk kc = (lane%16) ^ (m&15)
+  // A (TA=true): like B with lda stride: k-row 16w+2i+lane/32, m-chunk
+  // m0 + 4*(lane%32)
+  const long am_chunk = TA ? m0 + 4 * (lane & 31) : 0;
+
+  auto stage_glds = [&](int buf, long kt) {
+    const long k0 = kt * BK2;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      {  // B instr i
+        const int krow = 16 * wave + 2 * i + (lane >> 5);
+        if (bq < g.N) {
+          const float* src = bbase + (k0 + krow) * bstride;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) uint32_t*)(const
+                                                                  void*)src,
+              (__attribute__((address_space(3))) uint32_t*)
+                  &Bimg(buf)[(16 * wave + 2 * i) * 128],
+              16, 0, 0);
+        }
+      }
+      if (!TA) {  // A instr i: 4 m-rows, swizzled k source
+        const int mrow = 32 * wave + 4 * i + (lane >> 4);
+        const int kc = (lane & 15) ^ (mrow & 15);
+        if (m0 + mrow < g.M) {
+          const float* src = g.A + (m0 + mrow) * g.lda + k0 + 4 * kc;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) uint32_t*)(const
+                                                                  void*)src,
+              (__attribute__((address_space(3))) uint32_t*)
+                  &Aimg(buf)[(32 * wave + 4 * i) * BK2],
+              16, 0, 0);
+        }
+      } else {  // A instr i: 2 k-rows, contiguous m source
+        const int krow = 16 * wave + 2 * i + (lane >> 5);
+        if (am_chunk < g.M) {
+          const float* src = g.A + (k0 + krow) * g.lda + am_chunk;
+          __builtin_amdgcn_global_load_lds(
+              (const __attribute__((address_space(1))) uint32_t*)(const
+                                                                  void*)src,
+              (__attribute__((address_space(3))) uint32_t*)
+                  &Aimg(buf)[(16 * wave + 2 * i) * 128],
+              16, 0, 0);
+        }
+      }
+    }
+  };
+
+  // K-remainder tile: register-staged with zero fill (read16 masks).
+  auto stage_tail = [&](int buf, long kt) {
+    const long k0 = kt * BK2;
+    float r[16];
+    // B: thread t covers k-row (t&63), n-chunks (t>>6)*32 + {0,16}
+    const int bk = t & 63;
+    const int nb = (t >> 6) * 32;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      read16(g.B, k0 + bk, n0 + nb + 16 * h, g.ldb, g.K, g.N, g.bv, r);
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        Bimg(buf)[bk * 128 + nb + 16 * h + j] = r[j];
+    }
+    if (!TA) {
+      // A: thread t covers m-row (t&127), k-chunks (t>>7)*32 + {0,16},
+      // written through the m-swizzle
+      const int m = t & 127;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int kb = (t >> 7) * 32 + 16 * h;
+        read16(g.A, m, k0 + kb, g.lda, g.M, g.K, g.av, r);
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          const int k = kb + j;
+          Aimg(buf)[m * BK2 + (((k >> 2) ^ (m & 15)) << 2) + (k & 3)] =
+              r[j];
+        }
+      }
+    } else {
+      const int ak = t & 63;
+      const int mb = (t >> 6) * 32;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        read16(g.A, k0 + ak, m0 + mb + 16 * h, g.lda, g.K, g.M, g.av, r);
+#pragma unroll
+        for (int j = 0; j < 16; ++j)
+          Aimg(buf)[ak * 128 + mb + 16 * h + j] = r[j];
+      }
+    }
+  };
+
+  auto stage = [&](int buf, long kt) {
+    if (kt < nfull)
+      stage_glds(buf, kt);
+    else
+      stage_tail(buf, kt);
+  };
+
+  f32x16 acc00 = {}, acc01 = {}, acc10 = {}, acc11 = {};
+
+  stage(0, 0);
+  __syncthreads();
+
+  // per-lane read bases
+  const int a0base = TA ? (wr * 64 + row_in) : (wr * 64 + row_in) * BK2;
+  const int bbase_r = wc * 64 + row_in;
+  const int aswz = row_in & 15;
+
+  for (long tt = 0; tt < ntiles; ++tt) {
+    if (tt + 1 < ntiles) stage((tt + 1) & 1, tt + 1);
+    {
+      const float* Ab = Aimg(tt & 1);
+      const float* Bb = Bimg(tt & 1);
+#pragma unroll 8
+      for (int kk = 0; kk < BK2; kk += 2) {
+        const int krow = kk + ksel;
+        float a0, a1;
+        if (!TA) {
+          const int s = (((krow >> 2) ^ aswz) << 2) + (krow & 3);
+          a0 = Ab[a0base + s];
+          a1 = Ab[a0base + 32 * BK2 + s];
+        } else {
+          a0 = Ab[krow * 128 + a0base];
+          a1 = Ab[krow * 128 + a0base + 32];
+        }
+        const float b0 = Bb[krow * 128 + bbase_r];
+        const float b1 = Bb[krow * 128 + bbase_r + 32];
+        acc00 =
+            __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
+        acc01 =
+            __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc01, 0, 0, 0);
+        acc10 =
+            __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
+        acc11 =
+            __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
+      }
+    }
+    __syncthreads();  // publishes stage writes; implicit vmcnt(0) drains
+                      // the outstanding glds
+  }
+
+  // ---- epilogue (same acc->(row,col) map as k_gemm_f32, no split-K)
+  auto epi_tile = [&](const f32x16& a, int ti, int tj) {
+    const long col = n0 + wc * 64 + tj * 32 + row_in;
+    if (col >= g.N) return;
+    long col_base = 0;
+    bool col_ok = true;
+    if (g.spad > 0) {
+      const long n = col / g.spad;
+      const long sp = col - n * g.spad;
+      col_ok = sp < g.S;
+      long pix = sp;
+      if (g.OWo > 0) {
+        const int oh = (int)(sp / g.OWo);
+        const int ow = (int)(sp - (long)oh * g.OWo);
+        pix = ((long)oh * g.osh) * g.Wd + (long)ow * g.osw;
+      }
+      col_base = n * g.n_stride + pix;
+    }
+    if (!col_ok) return;
+    const float cbias = (g.bias && g.bias_per_col) ? g.bias[col] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long row =
+          m0 + wr * 64 + ti * 32 + ((r & 3) + 8 * (r >> 2) + 4 * ksel);
+      if (row >= g.M) continue;
+      float v = g.alpha * a[r];
+      if (g.bias) v += g.bias_per_col ? cbias : g.bias[row];
+      if (g.relu) v = fmaxf(v, 0.f);
+      const long off =
+          g.spad > 0 ? col_base + row * g.Srow : row * g.ldc + col;
+      if (g.beta != 0.f) v += g.beta * g.C[off];
+      g.C[off] = v;
+    }
+  };
+  epi_tile(acc00, 0, 0);
+  epi_tile(acc01, 0, 1);
+  epi_tile(acc10, 1, 0);
+  epi_tile(acc11, 1, 1);
+}
+
 // fixed-order split-K reduce: C[i] = Σ_sk slab[sk][i]
 __global__ void k_splitk_reduce(const float* __restrict__ slab, long MN,
                                 int SK, float* __restrict__ C) {
@@ -611,6 +863,35 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     hipLaunchKernelGGL(k_splitk_reduce, dim3(blocks), dim3(256), 0, s, slab,
                        MN, SK, C);
     return;
+  }
+  // glds v2 path (BK=64, 1 block/CU): NN/TN with plain or channel-view
+  // operands; implicit-im2col views (kh>0) and NT keep the register
+  // kernel (their contracted axis has per-image zero padding a DMA
+  // cannot inject).  CAFFE_GEMM_V2=0 reverts to v1 everywhere.
+  static const int v2 = [] {
+    const char* e = getenv("CAFFE_GEMM_V2");
+    return e ? atoi(e) : 1;
+  }();
+  if (v2 && !transB && M > 64 && N > 64 && !aview &&
+      (!bview || bview->kh == 0)) {
+    static const bool lds_ok = [] {
+      return hipFuncSetAttribute(
+                 (const void*)&k_gemm2<false>,
+                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                 4 * 128 * BK2 * (int)sizeof(float)) == hipSuccess &&
+             hipFuncSetAttribute(
+                 (const void*)&k_gemm2<true>,
+                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                 4 * 128 * BK2 * (int)sizeof(float)) == hipSuccess;
+    }();
+    if (lds_ok) {
+      const size_t shmem = 4 * 128 * BK2 * sizeof(float);
+      if (!transA)
+        hipLaunchKernelGGL((k_gemm2<false>), grid, block, shmem, s, g);
+      else
+        hipLaunchKernelGGL((k_gemm2<true>), grid, block, shmem, s, g);
+      return;
+    }
   }
   if (!transA && !transB)
     hipLaunchKernelGGL((k_gemm_f32<false, false, false>), grid, block, 0, s,
