@@ -36,6 +36,12 @@ void Engine::drain_events() {
 }
 
 void* DeviceAllocator::alloc(size_t bytes) {
+  // +64B safety tail on EVERY device allocation: staged GEMM loads read
+  // whole 16B lane chunks through per-image spad padding and trailing
+  // partial chunks (values are dropped or masked, but the BYTES must be
+  // mapped) — without slack the last row of a tensor can touch the next
+  // page (observed as a GPU memory-access fault under the glds kernel)
+  bytes += 64;
   {
     std::lock_guard<std::mutex> g(mu_);
     auto it = free_.find(bytes);
@@ -55,6 +61,7 @@ void* DeviceAllocator::alloc(size_t bytes) {
 }
 
 void DeviceAllocator::release(void* p, size_t bytes) {
+  bytes += 64;  // must mirror alloc()'s safety-tail key
   std::lock_guard<std::mutex> g(mu_);
   free_.emplace(bytes, p);
 }
