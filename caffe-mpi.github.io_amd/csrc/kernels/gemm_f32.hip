@@ -867,10 +867,14 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // glds v2 path (BK=64, 1 block/CU): NN/TN with plain or channel-view
   // operands; implicit-im2col views (kh>0) and NT keep the register
   // kernel (their contracted axis has per-image zero padding a DMA
-  // cannot inject).  CAFFE_GEMM_V2=0 reverts to v1 everywhere.
+  // cannot inject).  MEASURED NEGATIVE on the model shapes:
+  // 4096^3 98 vs 115 TF, ResNet shape mix ~-10% per class — at fp32's
+  // arithmetic density the 128KB/2-buffer variant's 1 block/CU loses the
+  // cross-block overlap v1 gets at 2 blocks/CU, and 3 buffers don't fit
+  // LDS at BK=64.  Kept behind CAFFE_GEMM_V2=1 as the measured record.
   static const int v2 = [] {
     const char* e = getenv("CAFFE_GEMM_V2");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   if (v2 && !transB && M > 64 && N > 64 && !aview &&
       (!bview || bview->kh == 0)) {
