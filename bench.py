@@ -20,6 +20,9 @@ sys.path.insert(0, os.path.join(REPO, "caffe-mpi.github.io_amd"))
 sys.path.insert(0, REPO)
 
 PEAK_F32_MFMA = 157.3e12  # gfx950 fp32 matrix peak (spec; 155 TF measured)
+# bf16 dense MFMA peak: 32x32x16 bf16 = 1024 FLOP/clk/SIMD x 4 x 256 CU
+# x 2.4 GHz (MI355X_MICROARCH; AMD's 5 PF headline is 2:1-sparse)
+PEAK_BF16_MFMA = 2516.6e12
 
 
 def oracle_resnet50_step(batch):
@@ -200,6 +203,9 @@ def main():
     ap.add_argument("--batch", type=int, default=128)  # per GPU
     ap.add_argument("--model", default="resnet50")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    # bf16 = mixed precision (bf16 MFMA, fp32 accumulation/storage) — a
+    # SECOND reported line; the BASELINE contract metric stays f32
+    ap.add_argument("--dtype", choices=["f32", "bf16"], default="f32")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -230,6 +236,8 @@ def main():
               "googlenet": (3, 224, 224)}
     ca.set_synthetic_shape(*shapes[args.model], 1000)
     ca.set_random_seed(1371)
+    if args.dtype == "bf16" and not cpu_dry:
+        ca.set_compute("bf16")
 
     import subprocess
     gen = os.path.join(REPO, "models", "generated",
@@ -308,12 +316,13 @@ def main():
                                / gc["launches"])
             except Exception:
                 traffic = None
+        peak = PEAK_BF16_MFMA if args.dtype == "bf16" else PEAK_F32_MFMA
         roofline = {
             "bound": "mfma",
             "achieved": achieved,
-            "peak": PEAK_F32_MFMA,
+            "peak": peak,
             "unit": "TFLOP/s",
-            "frac": achieved / PEAK_F32_MFMA,
+            "frac": achieved / peak,
             "traffic": traffic,
             "traffic_note": ("HBM bytes/launch of the GEMM class, "
                             "profiles/pmc_traffic_*.json "
@@ -358,7 +367,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "f32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "workload": f"{args.model} fp32 training, per-GPU batch "

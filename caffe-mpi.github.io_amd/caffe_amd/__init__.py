@@ -64,6 +64,7 @@ _lib.caffe_solver_iter.restype = ctypes.c_long
 _lib.caffe_net_create.restype = ctypes.c_void_p
 _lib.caffe_net_loss.restype = ctypes.c_float
 _lib.caffe_set_random_seed.argtypes = [ctypes.c_uint64]
+_lib.caffe_set_compute.argtypes = [ctypes.c_char_p]
 
 _f32p = ctypes.POINTER(ctypes.c_float)
 _lib.caffe_net_blob_get.argtypes = [_vp, ctypes.c_char_p, ctypes.c_int,
@@ -103,6 +104,12 @@ def _ckp(ptr):
 
 def set_mode(mode, device=0):
     _ck(_lib.caffe_set_mode(1 if mode == "gpu" else 0, device))
+
+
+def set_compute(dtype):
+    """GEMM compute dtype: "f32" (exact, default) or "bf16" (bf16 MFMA
+    with fp32 accumulation — mixed precision; storage stays fp32)."""
+    _ck(_lib.caffe_set_compute(dtype.encode()))
 
 
 def set_random_seed(seed):

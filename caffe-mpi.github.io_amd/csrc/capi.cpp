@@ -46,6 +46,22 @@ int caffe_set_mode(int mode, int device) {
   API_CATCH
 }
 
+// GEMM compute dtype: "f32" (exact MFMA, default) or "bf16" (bf16 MFMA,
+// fp32 accumulation — mixed precision, the reference Ftype/Btype analog).
+// Storage and every non-GEMM op stay fp32.
+int caffe_set_compute(const char* dtype) {
+  API_TRY
+  std::string d = dtype ? dtype : "";
+  if (d == "f32" || d == "fp32" || d == "float")
+    Engine::get().gemm_bf16 = false;
+  else if (d == "bf16" || d == "bfloat16")
+    Engine::get().gemm_bf16 = true;
+  else
+    CAMD_FATAL << "unknown compute dtype " << d;
+  return 0;
+  API_CATCH
+}
+
 int caffe_set_random_seed(uint64_t seed) {
   API_TRY
   Engine& E = Engine::get();

@@ -131,6 +131,12 @@ class Engine {
   int rank = 0;          // solver rank; fillers use seed+rank
   int world = 1;
 
+  // GEMM compute dtype: false = fp32 MFMA (exact, the parity/default
+  // path), true = bf16 MFMA with fp32 accumulation (mixed precision —
+  // the reference's Ftype/Btype fp16 analog).  Storage
+  // stays fp32 either way; only the conv/IP contractions change.
+  bool gemm_bf16 = false;
+
   // synthetic-data configuration (no LMDB datasets in this environment)
   bool synthetic = true;
   int syn_classes = 1000;

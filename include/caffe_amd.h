@@ -26,6 +26,12 @@ const char* caffe_last_error(void);
 
 /* Caffe::set_mode / SetDevice (common.hpp): mode 0 = CPU, 1 = GPU */
 int caffe_set_mode(int mode, int device);
+/* GEMM compute dtype: "f32" (default, exact) or "bf16" (bf16 MFMA with
+ * fp32 accumulation — the MI355X-native mixed-precision mode; replaces
+ * the reference's NetParameter default_forward_type/default_backward_type
+ * FLOAT16 machinery, net.cpp:100-156 / type.hpp:13-47, with fp32 storage
+ * and tensor-core math instead of fp16 storage). */
+int caffe_set_compute(const char* dtype);
 /* Caffe::set_random_seed (common.cpp); rank offsets seed like
  * parallel.cpp:179-187 */
 int caffe_set_random_seed(uint64_t seed);
