@@ -28,8 +28,9 @@ def bf16_mode():
 
 
 def _relerr(a, b):
-    a = np.asarray(a, np.float64)
-    b = np.asarray(b, np.float64)
+    a = np.asarray(a, np.float64).ravel()
+    b = np.asarray(b, np.float64).ravel()
+    assert a.size == b.size, (a.size, b.size)
     return np.abs(a - b).max() / max(np.abs(b).max(), 1e-6)
 
 

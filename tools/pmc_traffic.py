@@ -124,8 +124,10 @@ def main():
                       "MI355X_MICROARCH.md §HBM); WRITE_SIZE raw",
         "classes": classes,
     }
-    os.makedirs(os.path.join(REPO, "profiles"), exist_ok=True)
-    jpath = os.path.join(REPO, "profiles", f"pmc_traffic_{model}.json")
+    # written under gpurun_out/ (the only dir that travels back from the
+    # GPU box); copy into profiles/ and commit on the dev side
+    os.makedirs(os.path.join(REPO, "gpurun_out"), exist_ok=True)
+    jpath = os.path.join(REPO, "gpurun_out", f"pmc_traffic_{model}.json")
     with open(jpath, "w") as f:
         json.dump(result, f, indent=1, sort_keys=True)
     os.makedirs(os.path.join(REPO, "gpurun_out"), exist_ok=True)
