@@ -119,17 +119,13 @@ using f4 = __attribute__((ext_vector_type(4))) float;
 // 4-way unrolled grid-stride walk for the streaming kernels: the rolled
 // form keeps ONE load in flight per wave (PMC: 59-81% parked at ~3.5 TB/s
 // vs 6.3 achievable); BODY4 sees indices i0..i3, BODY1 mops up.
-// (call sites wrap bodies in ({ ... }) so their commas stay protected)
-#define VEC_GRID4(n4, BODY4, BODY1)                                  \
+#define VEC_GRID4(n4, BODY4, BODY1)                                   \
   {                                                                   \
     const long _stride = (long)gridDim.x * blockDim.x;                \
     long i0 = blockIdx.x * (long)blockDim.x + threadIdx.x;            \
     for (; i0 + 3 * _stride < (n4); i0 += 4 * _stride) {              \
       const long i1 = i0 + _stride, i2 = i0 + 2 * _stride,            \
                  i3 = i0 + 3 * _stride;                               \
-      (void)i1;                                                       \
-      (void)i2;                                                       \
-      (void)i3;                                                       \
       BODY4;                                                          \
     }                                                                 \
     for (; i0 < (n4); i0 += _stride) {                                \
@@ -144,13 +140,15 @@ __device__ __forceinline__ f4 relu4(f4 v, float slope) {
 }
 __global__ void k_relu_fwd(const f4* __restrict__ x, long n4, float slope,
                            f4* __restrict__ y) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
               y[i0] = relu4(v0, slope);
               y[i1] = relu4(v1, slope);
               y[i2] = relu4(v2, slope);
               y[i3] = relu4(v3, slope);
-            }), ({ y[i0] = relu4(x[i0], slope); }))
+            },
+            { y[i0] = relu4(x[i0], slope); })
 }
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y) {
   PerfScope perf(PERF_CLASS("relu"), s, 0, 8.0 * n);
@@ -167,14 +165,16 @@ __device__ __forceinline__ f4 relu_bwd4(f4 v, f4 d, float slope) {
 __global__ void k_relu_bwd(const f4* __restrict__ x,
                            const f4* __restrict__ dy, long n4, float slope,
                            f4* __restrict__ dx) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
               const f4 d0 = dy[i0], d1 = dy[i1], d2 = dy[i2], d3 = dy[i3];
               dx[i0] = relu_bwd4(v0, d0, slope);
               dx[i1] = relu_bwd4(v1, d1, slope);
               dx[i2] = relu_bwd4(v2, d2, slope);
               dx[i3] = relu_bwd4(v3, d3, slope);
-            }), ({ dx[i0] = relu_bwd4(x[i0], dy[i0], slope); }))
+            },
+            { dx[i0] = relu_bwd4(x[i0], dy[i0], slope); })
 }
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
               float slope, float* dx) {
@@ -191,14 +191,16 @@ __device__ __forceinline__ f4 axpy4(f4 y, f4 x, float a) {
 }
 __global__ void k_axpy(const f4* __restrict__ x, long n4, float a,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 y0 = y[i0], y1 = y[i1], y2 = y[i2], y3 = y[i3];
               const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
               y[i0] = axpy4(y0, x0, a);
               y[i1] = axpy4(y1, x1, a);
               y[i2] = axpy4(y2, x2, a);
               y[i3] = axpy4(y3, x3, a);
-            }), ({ y[i0] = axpy4(y[i0], x[i0], a); }))
+            },
+            { y[i0] = axpy4(y[i0], x[i0], a); })
 }
 // y += a*x over the (64B-padded) gradient arena — iter_size accumulation
 void axpy(hipStream_t s, long n, float a, const float* x, float* y) {
@@ -564,7 +566,8 @@ __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                               const f4* __restrict__ add,
                               f4* __restrict__ y) {
   VEC_GRID4(
-      n4, ({
+      n4,
+      {
         const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
         y[i0] = bn_norm_pack(i0, v0, mean, inv_std, scale, bias, sb, C, S,
                              frelu, add);
@@ -574,10 +577,11 @@ __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                              frelu, add);
         y[i3] = bn_norm_pack(i3, v3, mean, inv_std, scale, bias, sb, C, S,
                              frelu, add);
-      }), ({
+      },
+      {
         y[i0] = bn_norm_pack(i0, x[i0], mean, inv_std, scale, bias, sb, C,
                              S, frelu, add);
-      }))
+      })
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
@@ -860,7 +864,8 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                                const float* __restrict__ bias, int frelu,
                                int C, int S, long n4, f4* __restrict__ dx) {
   VEC_GRID4(
-      n4, ({
+      n4,
+      {
         const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
         const f4 d0 = dy[i0], d1 = dy[i1], d2 = dy[i2], d3 = dy[i3];
         dx[i0] = bn_bwd_pack(i0, x0, d0, mean, inv_std, scale, sb, m_dy,
@@ -871,10 +876,11 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                              m_dyxn, bias, frelu, C, S);
         dx[i3] = bn_bwd_pack(i3, x3, d3, mean, inv_std, scale, sb, m_dy,
                              m_dyxn, bias, frelu, C, S);
-      }), ({
+      },
+      {
         dx[i0] = bn_bwd_pack(i0, x[i0], dy[i0], mean, inv_std, scale, sb,
                              m_dy, m_dyxn, bias, frelu, C, S);
-      }))
+      })
 }
 void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std,
@@ -1125,13 +1131,15 @@ void axpby(hipStream_t s, long n, float a, const float* x, float b,
 
 __global__ void k_copy(long n4, const f4* __restrict__ x,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
               y[i0] = v0;
               y[i1] = v1;
               y[i2] = v2;
               y[i3] = v3;
-            }), ({ y[i0] = x[i0]; }))
+            },
+            { y[i0] = x[i0]; })
 }
 void copy(hipStream_t s, long n, const float* x, float* y) {
   if (x == y) return;
@@ -1159,14 +1167,16 @@ __device__ __forceinline__ f4 add_relu4(f4 a, f4 b, int frelu) {
 __global__ void k_add3(long n4, const f4* __restrict__ a,
                        const f4* __restrict__ b, int frelu,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 a0 = a[i0], a1 = a[i1], a2 = a[i2], a3 = a[i3];
               const f4 b0 = b[i0], b1 = b[i1], b2 = b[i2], b3 = b[i3];
               y[i0] = add_relu4(a0, b0, frelu);
               y[i1] = add_relu4(a1, b1, frelu);
               y[i2] = add_relu4(a2, b2, frelu);
               y[i3] = add_relu4(a3, b3, frelu);
-            }), ({ y[i0] = add_relu4(a[i0], b[i0], frelu); }))
+            },
+            { y[i0] = add_relu4(a[i0], b[i0], frelu); })
 }
 void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
           int fuse_relu) {
@@ -1178,14 +1188,16 @@ void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
 
 __global__ void k_acc(long n4, const f4* __restrict__ x,
                       f4* __restrict__ y) {
-  VEC_GRID4(n4, ({
+  VEC_GRID4(n4,
+            {
               const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
               const f4 y0 = y[i0], y1 = y[i1], y2 = y[i2], y3 = y[i3];
               y[i0] = y0 + x0;
               y[i1] = y1 + x1;
               y[i2] = y2 + x2;
               y[i3] = y3 + x3;
-            }), ({ y[i0] += x[i0]; }))
+            },
+            { y[i0] += x[i0]; })
 }
 void acc(hipStream_t s, long n, const float* x, float* y) {
   PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
