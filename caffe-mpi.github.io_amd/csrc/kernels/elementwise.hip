@@ -116,39 +116,15 @@ using f4 = __attribute__((ext_vector_type(4))) float;
 #define VEC_GRID(i, n4) \
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < (n4); \
        i += (long)gridDim.x * blockDim.x)
-// 4-way unrolled grid-stride walk for the streaming kernels: the rolled
-// form keeps ONE load in flight per wave (PMC: 59-81% parked at ~3.5 TB/s
-// vs 6.3 achievable); BODY4 sees indices i0..i3, BODY1 mops up.
-#define VEC_GRID4(n4, BODY4, BODY1)                                   \
-  {                                                                   \
-    const long _stride = (long)gridDim.x * blockDim.x;                \
-    long i0 = blockIdx.x * (long)blockDim.x + threadIdx.x;            \
-    for (; i0 + 3 * _stride < (n4); i0 += 4 * _stride) {              \
-      const long i1 = i0 + _stride, i2 = i0 + 2 * _stride,            \
-                 i3 = i0 + 3 * _stride;                               \
-      BODY4;                                                          \
-    }                                                                 \
-    for (; i0 < (n4); i0 += _stride) {                                \
-      BODY1;                                                          \
-    }                                                                 \
-  }
 // ------------------------------------------------------------ relu
-__device__ __forceinline__ f4 relu4(f4 v, float slope) {
-#pragma unroll
-  for (int j = 0; j < 4; ++j) v[j] = v[j] > 0.f ? v[j] : slope * v[j];
-  return v;
-}
 __global__ void k_relu_fwd(const f4* __restrict__ x, long n4, float slope,
                            f4* __restrict__ y) {
-  VEC_GRID4(n4,
-            {
-              const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
-              y[i0] = relu4(v0, slope);
-              y[i1] = relu4(v1, slope);
-              y[i2] = relu4(v2, slope);
-              y[i3] = relu4(v3, slope);
-            },
-            { y[i0] = relu4(x[i0], slope); })
+  VEC_GRID(i, n4) {
+    f4 v = x[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] = v[j] > 0.f ? v[j] : slope * v[j];
+    y[i] = v;
+  }
 }
 void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y) {
   PerfScope perf(PERF_CLASS("relu"), s, 0, 8.0 * n);
@@ -157,24 +133,16 @@ void relu_fwd(hipStream_t s, const float* x, long n, float slope, float* y) {
                      (const f4*)x, n4, slope, (f4*)y);
 }
 
-__device__ __forceinline__ f4 relu_bwd4(f4 v, f4 d, float slope) {
-#pragma unroll
-  for (int j = 0; j < 4; ++j) d[j] *= v[j] > 0.f ? 1.f : slope;
-  return d;
-}
 __global__ void k_relu_bwd(const f4* __restrict__ x,
                            const f4* __restrict__ dy, long n4, float slope,
                            f4* __restrict__ dx) {
-  VEC_GRID4(n4,
-            {
-              const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
-              const f4 d0 = dy[i0], d1 = dy[i1], d2 = dy[i2], d3 = dy[i3];
-              dx[i0] = relu_bwd4(v0, d0, slope);
-              dx[i1] = relu_bwd4(v1, d1, slope);
-              dx[i2] = relu_bwd4(v2, d2, slope);
-              dx[i3] = relu_bwd4(v3, d3, slope);
-            },
-            { dx[i0] = relu_bwd4(x[i0], dy[i0], slope); })
+  VEC_GRID(i, n4) {
+    const f4 v = x[i];
+    f4 d = dy[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) d[j] *= v[j] > 0.f ? 1.f : slope;
+    dx[i] = d;
+  }
 }
 void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
               float slope, float* dx) {
@@ -184,23 +152,15 @@ void relu_bwd(hipStream_t s, const float* x, const float* dy, long n,
                      (const f4*)x, (const f4*)dy, n4, slope, (f4*)dx);
 }
 
-__device__ __forceinline__ f4 axpy4(f4 y, f4 x, float a) {
-#pragma unroll
-  for (int j = 0; j < 4; ++j) y[j] += a * x[j];
-  return y;
-}
 __global__ void k_axpy(const f4* __restrict__ x, long n4, float a,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4,
-            {
-              const f4 y0 = y[i0], y1 = y[i1], y2 = y[i2], y3 = y[i3];
-              const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
-              y[i0] = axpy4(y0, x0, a);
-              y[i1] = axpy4(y1, x1, a);
-              y[i2] = axpy4(y2, x2, a);
-              y[i3] = axpy4(y3, x3, a);
-            },
-            { y[i0] = axpy4(y[i0], x[i0], a); })
+  VEC_GRID(i, n4) {
+    f4 v = y[i];
+    const f4 xv = x[i];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] += a * xv[j];
+    y[i] = v;
+  }
 }
 // y += a*x over the (64B-padded) gradient arena — iter_size accumulation
 void axpy(hipStream_t s, long n, float a, const float* x, float* y) {
@@ -523,40 +483,6 @@ void bn_fwd_finalize(hipStream_t s, const void* partials, int nb, int C,
 
 // flat float4 walk; per 4-pack ONE division recovers the channel (packs
 // never straddle a channel boundary check: split handled elementwise)
-__device__ __forceinline__ f4 bn_norm_pack(
-    long i, f4 v, const float* __restrict__ mean,
-    const float* __restrict__ inv_std, const float* __restrict__ scale,
-    const float* __restrict__ bias, int sb, int C, int S, int frelu,
-    const f4* __restrict__ add) {
-  const long e0 = i * 4;
-  const int row = (int)(e0 / S);  // n*C + c
-  const int c0 = row % C;
-  const int rem = (int)(e0 - (long)row * S);
-  if (rem + 4 <= S) {
-    const float mu = mean[c0], inv = inv_std[c0];
-    const float sc = sb ? scale[c0] : 1.f, bi = sb ? bias[c0] : 0.f;
-#pragma unroll
-    for (int j = 0; j < 4; ++j) v[j] = (v[j] - mu) * inv * sc + bi;
-  } else {  // pack crosses a channel boundary
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const long e = e0 + j;
-      const int cc = (int)((e / S) % C);
-      v[j] = (v[j] - mean[cc]) * inv_std[cc] * (sb ? scale[cc] : 1.f) +
-             (sb ? bias[cc] : 0.f);
-    }
-  }
-  if (add) {  // fused residual: bn(x) + other (ResNet block pattern)
-    const f4 a = add[i];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) v[j] += a[j];
-  }
-  if (frelu) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
-  }
-  return v;
-}
 __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                               const float* __restrict__ mean,
                               const float* __restrict__ inv_std,
@@ -565,23 +491,37 @@ __global__ void k_bn_fwd_norm(const f4* __restrict__ x,
                               int S, int frelu, long n4,
                               const f4* __restrict__ add,
                               f4* __restrict__ y) {
-  VEC_GRID4(
-      n4,
-      {
-        const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
-        y[i0] = bn_norm_pack(i0, v0, mean, inv_std, scale, bias, sb, C, S,
-                             frelu, add);
-        y[i1] = bn_norm_pack(i1, v1, mean, inv_std, scale, bias, sb, C, S,
-                             frelu, add);
-        y[i2] = bn_norm_pack(i2, v2, mean, inv_std, scale, bias, sb, C, S,
-                             frelu, add);
-        y[i3] = bn_norm_pack(i3, v3, mean, inv_std, scale, bias, sb, C, S,
-                             frelu, add);
-      },
-      {
-        y[i0] = bn_norm_pack(i0, x[i0], mean, inv_std, scale, bias, sb, C,
-                             S, frelu, add);
-      })
+  VEC_GRID(i, n4) {
+    const long e0 = i * 4;
+    const int row = (int)(e0 / S);  // n*C + c
+    const int c0 = row % C;
+    const int rem = (int)(e0 - (long)row * S);
+    f4 v = x[i];
+    if (rem + 4 <= S) {
+      const float mu = mean[c0], inv = inv_std[c0];
+      const float sc = sb ? scale[c0] : 1.f, bi = sb ? bias[c0] : 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = (v[j] - mu) * inv * sc + bi;
+    } else {  // pack crosses a channel boundary
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long e = e0 + j;
+        const int cc = (int)((e / S) % C);
+        v[j] = (v[j] - mean[cc]) * inv_std[cc] * (sb ? scale[cc] : 1.f) +
+               (sb ? bias[cc] : 0.f);
+      }
+    }
+    if (add) {  // fused residual: bn(x) + other (ResNet block pattern)
+      const f4 a = add[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] += a[j];
+    }
+    if (frelu) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
+    }
+    y[i] = v;
+  }
 }
 void bn_fwd_norm(hipStream_t s, const float* x, const float* mean,
                  const float* inv_std, const float* scale, const float* bias,
@@ -818,42 +758,6 @@ void bn_bwd_finalize(hipStream_t s, const void* partials, int nb, int C,
                      dbias, m_dy, m_dyxn);
 }
 
-__device__ __forceinline__ f4 bn_bwd_pack(
-    long i, f4 xv, f4 d, const float* __restrict__ mean,
-    const float* __restrict__ inv_std, const float* __restrict__ scale,
-    int sb, const float* __restrict__ m_dy,
-    const float* __restrict__ m_dyxn, const float* __restrict__ bias,
-    int frelu, int C, int S) {
-  const long e0 = i * 4;
-  const int row = (int)(e0 / S);
-  const int rem = (int)(e0 - (long)row * S);
-  if (rem + 4 <= S) {
-    const int c = row % C;
-    const float mu = mean[c], inv = inv_std[c];
-    const float sc = sb ? scale[c] : 1.f;
-    const float bi = (frelu && bias) ? bias[c] : 0.f;
-    const float mdy = m_dy[c], mdyxn = m_dyxn[c];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float xn = (xv[j] - mu) * inv;
-      float dj = d[j];
-      if (frelu && xn * sc + bi <= 0.f) dj = 0.f;
-      d[j] = (dj * sc - mdy - mdyxn * xn) * inv;
-    }
-  } else {
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int cc = (int)(((e0 + j) / S) % C);
-      const float xn = (xv[j] - mean[cc]) * inv_std[cc];
-      const float scc = sb ? scale[cc] : 1.f;
-      float dj = d[j];
-      if (frelu && xn * scc + ((frelu && bias) ? bias[cc] : 0.f) <= 0.f)
-        dj = 0.f;
-      d[j] = (dj * scc - m_dy[cc] - m_dyxn[cc] * xn) * inv_std[cc];
-    }
-  }
-  return d;
-}
 __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                                const f4* __restrict__ dy,
                                const float* __restrict__ mean,
@@ -863,24 +767,39 @@ __global__ void k_bn_bwd_apply(const f4* __restrict__ x,
                                const float* __restrict__ m_dyxn,
                                const float* __restrict__ bias, int frelu,
                                int C, int S, long n4, f4* __restrict__ dx) {
-  VEC_GRID4(
-      n4,
-      {
-        const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
-        const f4 d0 = dy[i0], d1 = dy[i1], d2 = dy[i2], d3 = dy[i3];
-        dx[i0] = bn_bwd_pack(i0, x0, d0, mean, inv_std, scale, sb, m_dy,
-                             m_dyxn, bias, frelu, C, S);
-        dx[i1] = bn_bwd_pack(i1, x1, d1, mean, inv_std, scale, sb, m_dy,
-                             m_dyxn, bias, frelu, C, S);
-        dx[i2] = bn_bwd_pack(i2, x2, d2, mean, inv_std, scale, sb, m_dy,
-                             m_dyxn, bias, frelu, C, S);
-        dx[i3] = bn_bwd_pack(i3, x3, d3, mean, inv_std, scale, sb, m_dy,
-                             m_dyxn, bias, frelu, C, S);
-      },
-      {
-        dx[i0] = bn_bwd_pack(i0, x[i0], dy[i0], mean, inv_std, scale, sb,
-                             m_dy, m_dyxn, bias, frelu, C, S);
-      })
+  VEC_GRID(i, n4) {
+    const long e0 = i * 4;
+    const int row = (int)(e0 / S);
+    const int rem = (int)(e0 - (long)row * S);
+    const f4 xv = x[i];
+    f4 d = dy[i];
+    if (rem + 4 <= S) {
+      const int c = row % C;
+      const float mu = mean[c], inv = inv_std[c];
+      const float sc = sb ? scale[c] : 1.f;
+      const float bi = (frelu && bias) ? bias[c] : 0.f;
+      const float mdy = m_dy[c], mdyxn = m_dyxn[c];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float xn = (xv[j] - mu) * inv;
+        float dj = d[j];
+        if (frelu && xn * sc + bi <= 0.f) dj = 0.f;
+        d[j] = (dj * sc - mdy - mdyxn * xn) * inv;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int cc = (int)(((e0 + j) / S) % C);
+        const float xn = (xv[j] - mean[cc]) * inv_std[cc];
+        const float scc = sb ? scale[cc] : 1.f;
+        float dj = d[j];
+        if (frelu && xn * scc + ((frelu && bias) ? bias[cc] : 0.f) <= 0.f)
+          dj = 0.f;
+        d[j] = (dj * scc - m_dy[cc] - m_dyxn[cc] * xn) * inv_std[cc];
+      }
+    }
+    dx[i] = d;
+  }
 }
 void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
                   const float* mean, const float* inv_std,
@@ -1131,15 +1050,7 @@ void axpby(hipStream_t s, long n, float a, const float* x, float b,
 
 __global__ void k_copy(long n4, const f4* __restrict__ x,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4,
-            {
-              const f4 v0 = x[i0], v1 = x[i1], v2 = x[i2], v3 = x[i3];
-              y[i0] = v0;
-              y[i1] = v1;
-              y[i2] = v2;
-              y[i3] = v3;
-            },
-            { y[i0] = x[i0]; })
+  VEC_GRID(i, n4) y[i] = x[i];
 }
 void copy(hipStream_t s, long n, const float* x, float* y) {
   if (x == y) return;
@@ -1156,27 +1067,17 @@ void set_const(hipStream_t s, long n, float v, float* y) {
   hipLaunchKernelGGL(k_set, dim3(nblocks(n, 8)), dim3(TPB), 0, s, n, v, y);
 }
 
-__device__ __forceinline__ f4 add_relu4(f4 a, f4 b, int frelu) {
-  f4 v = a + b;
-  if (frelu) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
-  }
-  return v;
-}
 __global__ void k_add3(long n4, const f4* __restrict__ a,
                        const f4* __restrict__ b, int frelu,
                        f4* __restrict__ y) {
-  VEC_GRID4(n4,
-            {
-              const f4 a0 = a[i0], a1 = a[i1], a2 = a[i2], a3 = a[i3];
-              const f4 b0 = b[i0], b1 = b[i1], b2 = b[i2], b3 = b[i3];
-              y[i0] = add_relu4(a0, b0, frelu);
-              y[i1] = add_relu4(a1, b1, frelu);
-              y[i2] = add_relu4(a2, b2, frelu);
-              y[i3] = add_relu4(a3, b3, frelu);
-            },
-            { y[i0] = add_relu4(a[i0], b[i0], frelu); })
+  VEC_GRID(i, n4) {
+    f4 v = a[i] + b[i];
+    if (frelu) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v[j] = fmaxf(v[j], 0.f);
+    }
+    y[i] = v;
+  }
 }
 void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
           int fuse_relu) {
@@ -1188,16 +1089,7 @@ void add3(hipStream_t s, long n, const float* a, const float* b, float* y,
 
 __global__ void k_acc(long n4, const f4* __restrict__ x,
                       f4* __restrict__ y) {
-  VEC_GRID4(n4,
-            {
-              const f4 x0 = x[i0], x1 = x[i1], x2 = x[i2], x3 = x[i3];
-              const f4 y0 = y[i0], y1 = y[i1], y2 = y[i2], y3 = y[i3];
-              y[i0] = y0 + x0;
-              y[i1] = y1 + x1;
-              y[i2] = y2 + x2;
-              y[i3] = y3 + x3;
-            },
-            { y[i0] += x[i0]; })
+  VEC_GRID(i, n4) y[i] += x[i];
 }
 void acc(hipStream_t s, long n, const float* x, float* y) {
   PerfScope perf(PERF_CLASS("eltwise"), s, 0, 12.0 * n);
