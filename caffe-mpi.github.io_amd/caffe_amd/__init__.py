@@ -106,6 +106,16 @@ def set_mode(mode, device=0):
     _ck(_lib.caffe_set_mode(1 if mode == "gpu" else 0, device))
 
 
+def set_data_iter(i):
+    """Data-stream position (LMDB cursor / synthetic counter)."""
+    _ck(_lib.caffe_set_data_iter(ctypes.c_uint64(i)))
+
+
+def set_rank_world(rank, world):
+    """Engine rank/world for the sharded data feed (no communicator)."""
+    _ck(_lib.caffe_set_rank_world(rank, world))
+
+
 def set_compute(dtype):
     """GEMM compute dtype: "f32" (exact, default) or "bf16" (bf16 MFMA
     with fp32 accumulation — mixed precision; storage stays fp32)."""

@@ -46,6 +46,28 @@ int caffe_set_mode(int mode, int device) {
   API_CATCH
 }
 
+// data-stream iteration counter (the LMDB cursor / synthetic stream
+// position; Solver::Step drives it during training — tests reset it)
+int caffe_set_data_iter(uint64_t iter) {
+  API_TRY
+  Engine::get().data_iter = iter;
+  return 0;
+  API_CATCH
+}
+
+// rank/world without a communicator (single-process tests of the
+// rank-sharded data feed; comm_init sets the same fields when a real
+// collective attaches — reference P2PSync rank plumbing)
+int caffe_set_rank_world(int rank, int world) {
+  API_TRY
+  Engine& E = Engine::get();
+  E.rank = rank;
+  E.world = world;
+  E.cpu_rng.seed(E.seed + (uint64_t)rank);
+  return 0;
+  API_CATCH
+}
+
 // GEMM compute dtype: "f32" (exact MFMA, default) or "bf16" (bf16 MFMA,
 // fp32 accumulation — mixed precision, the reference Ftype/Btype analog).
 // Storage and every non-GEMM op stay fp32.

@@ -1245,6 +1245,44 @@ void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
                      decay_mults, nseg, mom, lr, decay, gscale);
 }
 
+// ---------------------------------------------------- LMDB transform
+// uint8 -> fp32 crop/mirror/mean/scale (reference data_transformer.cu:
+// 14-100): one thread per output element; geo = per-image (ho, wo,
+// mirror); mean subtracted in SOURCE coordinates (mean_file semantics).
+__global__ void k_transform_u8(const uint8_t* __restrict__ in, int N,
+                               int C, int inH, int inW, int outH, int outW,
+                               const int* __restrict__ geo,
+                               const float* __restrict__ mean,
+                               int mean_mode, float scale,
+                               float* __restrict__ out) {
+  const long total = (long)N * C * outH * outW;
+  GRID_STRIDE(i, total) {
+    const int x = (int)(i % outW);
+    const int y = (int)((i / outW) % outH);
+    const int c = (int)((i / ((long)outW * outH)) % C);
+    const int n = (int)(i / ((long)outW * outH * C));
+    const int ho = geo[3 * n], wo = geo[3 * n + 1], mir = geo[3 * n + 2];
+    const int sx = wo + (mir ? outW - 1 - x : x);
+    const long src = (((long)n * C + c) * inH + ho + y) * inW + sx;
+    float m = 0.f;
+    if (mean_mode == 1)
+      m = mean[c];
+    else if (mean_mode == 2)
+      m = mean[((long)c * inH + ho + y) * inW + sx];
+    out[i] = ((float)in[src] - m) * scale;
+  }
+}
+void transform_u8(hipStream_t s, const uint8_t* in, int N, int C, int inH,
+                  int inW, int outH, int outW, const int* geo,
+                  const float* mean, int mean_mode, float scale,
+                  float* out) {
+  const long total = (long)N * C * outH * outW;
+  PerfScope perf(PERF_CLASS("data"), s, 0, 5.0 * total);
+  hipLaunchKernelGGL(k_transform_u8, dim3(nblocks(total, 4)), dim3(TPB), 0,
+                     s, in, N, C, inH, inW, outH, outW, geo, mean,
+                     mean_mode, scale, out);
+}
+
 // ------------------------------------------------------------ synthetic
 __global__ void k_fill_uniform(long n, uint64_t key, float lo, float hi,
                                float* __restrict__ y) {

@@ -123,7 +123,10 @@ struct Workspace {
 };
 
 // ------------------------------------------------------------- layer classes
-class DataLayer : public Layer {  // synthetic source
+struct LmdbFeed;  // data_lmdb.cpp: reader + prefetch + transform state
+
+class DataLayer : public Layer {  // synthetic or LMDB source (§8a a12,
+                                  // §8f.1)
  public:
   using Layer::Layer;
   void LayerSetUp(const std::vector<Blob*>&,
@@ -141,6 +144,15 @@ class DataLayer : public Layer {  // synthetic source
 
   int batch_ = 0, C_ = 3, H_ = 224, W_ = 224;
   uint64_t iter_ = 0;
+
+  // LMDB mode (data_param.source exists on disk — else synthetic):
+  // from-scratch reader + prefetch worker + GPU uint8 transform
+  void setup_lmdb(const std::string& source);
+  void forward_lmdb_cpu(const std::vector<Blob*>& top);
+  void forward_lmdb_gpu(const std::vector<Blob*>& top);
+  std::shared_ptr<LmdbFeed> feed_;
+  int u8_slot_ = -1, geo_slot_ = -1, mean_slot_ = -1;
+  bool mean_uploaded_ = false;
 };
 
 class ConvolutionLayer : public Layer {

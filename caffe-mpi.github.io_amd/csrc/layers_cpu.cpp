@@ -3,6 +3,8 @@
 // parity is checked against oracle/, never against this file).
 #include <cmath>
 
+#include <sys/stat.h>
+
 #include "layers.hpp"
 
 namespace camd {
@@ -129,9 +131,24 @@ Workspace::~Workspace() {
 // ---------------------------------------------------------------- Data
 void DataLayer::LayerSetUp(const std::vector<Blob*>&,
                            const std::vector<Blob*>&) {
-  Engine& E = Engine::get();
   auto dp = param_->sub("data_param");
   batch_ = dp ? (int)dp->inum("batch_size", 1) : 1;
+  // LMDB source when data_param.source EXISTS on disk (reference
+  // db_lmdb.cpp backend); a prototxt naming an absent dataset falls back
+  // to the synthetic stream so the reference model zoo runs unmodified
+  // in this dataset-less environment (round-1 drop-in contract)
+  const std::string src = dp ? dp->str("source", "") : "";
+  if (!src.empty()) {
+    struct stat st {};
+    if (stat(src.c_str(), &st) == 0) {
+      setup_lmdb(src);
+      return;
+    }
+    fprintf(stderr,
+            "[caffe_amd] DataLayer '%s': source '%s' not found — using "
+            "the synthetic stream\n",
+            name_.c_str(), src.c_str());
+  }
   auto tp = param_->sub("transform_param");
   const int crop = tp ? (int)tp->inum("crop_size", 0) : 0;
   // synthetic shape: explicit engine override, else crop_size (3 channels)
@@ -148,8 +165,6 @@ void DataLayer::LayerSetUp(const std::vector<Blob*>&,
                << "': no crop_size and no synthetic shape set "
                   "(caffe_set_synthetic_shape)";
   }
-  CHECK_(E.synthetic) << "only the synthetic data source is available in "
-                         "this build (LMDB pipeline: later round)";
 }
 int g_syn_shape[3] = {0, 0, 0};
 
@@ -161,6 +176,11 @@ void DataLayer::Reshape(const std::vector<Blob*>&,
 
 void DataLayer::Forward_cpu(const std::vector<Blob*>&,
                             const std::vector<Blob*>& top) {
+  if (feed_) {
+    forward_lmdb_cpu(top);
+    ++iter_;
+    return;
+  }
   Engine& E = Engine::get();
   const uint64_t key = h_splitmix64(E.seed ^ ((uint64_t)E.rank << 40) ^
                                   (E.data_iter << 8));

@@ -16,6 +16,11 @@ using gpu::GemmView;
 
 void DataLayer::Forward_gpu(const std::vector<Blob*>&,
                             const std::vector<Blob*>& top) {
+  if (feed_) {
+    forward_lmdb_gpu(top);
+    ++iter_;
+    return;
+  }
   Engine& E = Engine::get();
   const uint64_t key =
       h_splitmix64(E.seed ^ ((uint64_t)E.rank << 40) ^ (E.data_iter << 8));

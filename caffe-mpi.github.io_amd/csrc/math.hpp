@@ -208,6 +208,14 @@ void sgd_update_segmented(hipStream_t s, long lo, long hi, float* g_arena,
                           const float* decay_mults, int nseg, float mom,
                           float lr, float decay, float gscale);
 
+// LMDB batch transform: uint8 -> fp32 crop/mirror/mean/scale
+// (data_transformer.cu:14-100 analog); geo = per-image (ho, wo, mirror),
+// mean_mode 0 = none, 1 = per-channel, 2 = per-pixel (mean_file)
+void transform_u8(hipStream_t s, const uint8_t* in, int N, int C, int inH,
+                  int inW, int outH, int outW, const int* geo,
+                  const float* mean, int mean_mode, float scale,
+                  float* out);
+
 void fill_uniform(hipStream_t s, long n, uint64_t seed, uint64_t counter,
                   float lo, float hi, float* y);
 void fill_labels(hipStream_t s, long n, uint64_t seed, uint64_t counter,

@@ -32,6 +32,11 @@ int caffe_set_mode(int mode, int device);
  * FLOAT16 machinery, net.cpp:100-156 / type.hpp:13-47, with fp32 storage
  * and tensor-core math instead of fp16 storage). */
 int caffe_set_compute(const char* dtype);
+/* rank/world for the sharded data feed without a communicator (tests);
+ * caffe_comm_init sets the same engine fields. */
+int caffe_set_rank_world(int rank, int world);
+/* data-stream position (LMDB cursor analog; Solver::Step drives it). */
+int caffe_set_data_iter(uint64_t iter);
 /* Caffe::set_random_seed (common.cpp); rank offsets seed like
  * parallel.cpp:179-187 */
 int caffe_set_random_seed(uint64_t seed);
