@@ -329,6 +329,227 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
 }
 
 // ==================================================================
+// k_gemm_slim — small-M tile variant (round 2): TM=32 with
+// TN=256, or TM=64 with TN=128.  The 128x128 tile leaves most MFMA rows
+// empty for the inception-branch and stage-1 shapes (Cout 16..64:
+// GoogLeNet conv GEMMs measured 25% of peak, ResNet-50's
+// 64x401408x576 41 TF); here each wave owns a 32(M) x 64(N) sub-tile
+// with TWO 32x32 accumulators, waves tile (TM/32) x (4/(TM/32)) — every
+// MFMA row is a real output row.  Structure otherwise mirrors k_gemm_f32
+// (register T14 pipeline, one barrier per K-tile, same epilogue/split-K
+// slabs); at TM=64 the LDS footprint (50 KB) admits 3 blocks/CU.
+template <bool TA, bool TB, int TM, bool SPLITK>
+__launch_bounds__(256, 2) __global__ void k_gemm_slim(GemmArgs g) {
+  constexpr int TN = TM == 32 ? 256 : 128;
+  constexpr int AS_S = TM + 1;  // +1 pad: conflict-free b32 banking
+  constexpr int BS_S = TN + 1;
+  __shared__ float smem[2 * BK * AS_S + 2 * BK * BS_S];
+  auto As = [&](int buf) -> float* { return smem + buf * (BK * AS_S); };
+  auto Bs = [&](int buf) -> float* {
+    return smem + 2 * BK * AS_S + buf * (BK * BS_S);
+  };
+
+  long flat = blockIdx.x;
+  {
+    const long nwg = g.tiles;
+    const long q = nwg / 8, rr = nwg % 8;
+    const long xcd = flat % 8, idx = flat / 8;
+    flat = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + idx;
+  }
+  const long tile_m = flat / g.tn;
+  const long tile_n = flat - tile_m * g.tn;
+  const long m0 = tile_m * TM, n0 = tile_n * TN;
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wave = t >> 6;
+  constexpr int mw = TM / 32;
+  const int wr = wave % mw;
+  const int wc = wave / mw;
+  const int row_in = lane & 31;
+  const int ksel = lane >> 5;
+
+  long k_lo = 0, k_hi = g.K;
+  if (SPLITK) {
+    const int sk = blockIdx.z;
+    k_lo = g.K * sk / g.SK / BK * BK;
+    k_hi = (sk == g.SK - 1) ? g.K : g.K * (sk + 1) / g.SK / BK * BK;
+    if (k_lo >= k_hi) return;
+  }
+  const long ntiles = (k_hi - k_lo + BK - 1) / BK;
+
+  f32x16 acc0 = {}, acc1 = {};
+  float ra[16], rb[16], rb2[16];
+
+  // staging roles (per K-tile):
+  //  A !TA ([M][K] contig K): TM*BK floats — TM=32: threads 0..63 cover
+  //    (m = t&31, kb = ((t>>5)&1)*16); TM=64: threads 0..127,
+  //    (m = t&63, kb = ((t>>6)&1)*16)
+  //  A TA ([K][M] contig M): thread (k = t&31, mb = (t>>5)*16 while
+  //    mb < TM) — read16 zero-fills past M
+  //  B !TB ([K][N] contig N): TN=256: (k = t&31, nb = (t>>5)*32), two
+  //    chunks; TN=128: (k = t&31, nb = (t>>5)*16), one chunk
+  //  B TB ([N][K] contig K): TN=256: (n = t, kb 0/16), two chunks over
+  //    k; TN=128: (n = t&127, kb = (t>>7)*16), one chunk
+  auto a_load = [&](long k0) {
+    if (!TA) {
+      const int m = TM == 32 ? (t & 31) : (t & 63);
+      const int kb = TM == 32 ? ((t >> 5) & 1) * 16 : ((t >> 6) & 1) * 16;
+      const bool mine = TM == 32 ? t < 64 : t < 128;
+      if (mine) read16(g.A, m0 + m, k0 + kb, g.lda, g.M, g.K, g.av, ra);
+    } else {
+      const int k = t & 31;
+      const int mb = (t >> 5) * 16;
+      if (mb < TM)
+        read16(g.A, k0 + k, m0 + mb, g.lda, g.K, g.M, g.av, ra);
+    }
+  };
+  auto a_write = [&](float* A_) {
+    if (!TA) {
+      const int m = TM == 32 ? (t & 31) : (t & 63);
+      const int kb = TM == 32 ? ((t >> 5) & 1) * 16 : ((t >> 6) & 1) * 16;
+      const bool mine = TM == 32 ? t < 64 : t < 128;
+      if (mine) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) A_[(kb + j) * AS_S + m] = ra[j];
+      }
+    } else {
+      const int k = t & 31;
+      const int mb = (t >> 5) * 16;
+      if (mb < TM) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) A_[k * AS_S + mb + j] = ra[j];
+      }
+    }
+  };
+  auto b_load = [&](long k0) {
+    if (!TB) {
+      const int k = t & 31;
+      if (TN == 256) {
+        const int nb = (t >> 5) * 32;
+        read16(g.B, k0 + k, n0 + nb, g.ldb, g.K, g.N, g.bv, rb);
+        read16(g.B, k0 + k, n0 + nb + 16, g.ldb, g.K, g.N, g.bv, rb2);
+      } else {
+        const int nb = (t >> 5) * 16;
+        read16(g.B, k0 + k, n0 + nb, g.ldb, g.K, g.N, g.bv, rb);
+      }
+    } else {
+      if (TN == 256) {
+        read16(g.B, n0 + t, k0, g.ldb, g.N, g.K, g.bv, rb);
+        read16(g.B, n0 + t, k0 + 16, g.ldb, g.N, g.K, g.bv, rb2);
+      } else {
+        const int n = t & 127;
+        const int kb = (t >> 7) * 16;
+        read16(g.B, n0 + n, k0 + kb, g.ldb, g.N, g.K, g.bv, rb);
+      }
+    }
+  };
+  auto b_write = [&](float* B_) {
+    if (!TB) {
+      const int k = t & 31;
+      if (TN == 256) {
+        const int nb = (t >> 5) * 32;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          B_[k * BS_S + nb + j] = rb[j];
+          B_[k * BS_S + nb + 16 + j] = rb2[j];
+        }
+      } else {
+        const int nb = (t >> 5) * 16;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) B_[k * BS_S + nb + j] = rb[j];
+      }
+    } else {
+      if (TN == 256) {
+#pragma unroll
+        for (int j = 0; j < 16; ++j) {
+          B_[j * BS_S + t] = rb[j];
+          B_[(16 + j) * BS_S + t] = rb2[j];
+        }
+      } else {
+        const int n = t & 127;
+        const int kb = (t >> 7) * 16;
+#pragma unroll
+        for (int j = 0; j < 16; ++j) B_[(kb + j) * BS_S + n] = rb[j];
+      }
+    }
+  };
+
+  a_load(k_lo);
+  b_load(k_lo);
+  a_write(As(0));
+  b_write(Bs(0));
+  __syncthreads();
+
+  int cur = 0;
+  for (long tt = 0; tt < ntiles; ++tt) {
+    if (tt + 1 < ntiles) {
+      a_load(k_lo + (tt + 1) * BK);
+      b_load(k_lo + (tt + 1) * BK);
+    }
+    {
+      const float* Ab = As(cur);
+      const float* Bb = Bs(cur);
+#pragma unroll
+      for (int kk = 0; kk < BK; kk += 2) {
+        const int krow = kk + ksel;
+        const float a0 = Ab[krow * AS_S + wr * 32 + row_in];
+        const float b0 = Bb[krow * BS_S + wc * 64 + row_in];
+        const float b1 = Bb[krow * BS_S + wc * 64 + 32 + row_in];
+        acc0 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc1, 0, 0, 0);
+      }
+    }
+    if (tt + 1 < ntiles) {
+      a_write(As(cur ^ 1));
+      b_write(Bs(cur ^ 1));
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  auto epi_tile = [&](const f32x16& a, int tj) {
+    const long col = n0 + wc * 64 + tj * 32 + row_in;
+    if (col >= g.N) return;
+    long col_base = 0;
+    bool col_ok = true;
+    if (!SPLITK && g.spad > 0) {
+      const long n = col / g.spad;
+      const long sp = col - n * g.spad;
+      col_ok = sp < g.S;
+      long pix = sp;
+      if (g.OWo > 0) {
+        const int oh = (int)(sp / g.OWo);
+        const int ow = (int)(sp - (long)oh * g.OWo);
+        pix = ((long)oh * g.osh) * g.Wd + (long)ow * g.osw;
+      }
+      col_base = n * g.n_stride + pix;
+    }
+    if (!col_ok) return;
+    const float cbias =
+        (!SPLITK && g.bias && g.bias_per_col) ? g.bias[col] : 0.f;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const long row =
+          m0 + wr * 32 + ((r & 3) + 8 * (r >> 2) + 4 * ksel);
+      if (row >= g.M) continue;
+      float v = g.alpha * a[r];
+      if (SPLITK) {
+        g.slab[((long)blockIdx.z * g.M + row) * g.N + col] = v;
+        continue;
+      }
+      if (g.bias) v += g.bias_per_col ? cbias : g.bias[row];
+      if (g.relu) v = fmaxf(v, 0.f);
+      const long off =
+          g.spad > 0 ? col_base + row * g.Srow : row * g.ldc + col;
+      if (g.beta != 0.f) v += g.beta * g.C[off];
+      g.C[off] = v;
+    }
+  };
+  epi_tile(acc0, 0);
+  epi_tile(acc1, 1);
+}
+
+// ==================================================================
 // k_gemm2 — glds (global_load_lds) staged fp32 MFMA GEMM, BK=64.
 //
 // Round-2 load-path redesign (cdna_hip_programming.md "glds vs register
@@ -627,6 +848,23 @@ __global__ void k_splitk_reduce_v4(const float* __restrict__ slab, long MN4,
   }
 }
 
+template <int TM, bool SPLITK>
+static void launch_slim(bool transA, bool transB, dim3 grid, dim3 block,
+                        hipStream_t s, const GemmArgs& g) {
+  if (!transA && !transB)
+    hipLaunchKernelGGL((k_gemm_slim<false, false, TM, SPLITK>), grid, block,
+                       0, s, g);
+  else if (!transA && transB)
+    hipLaunchKernelGGL((k_gemm_slim<false, true, TM, SPLITK>), grid, block,
+                       0, s, g);
+  else if (transA && !transB)
+    hipLaunchKernelGGL((k_gemm_slim<true, false, TM, SPLITK>), grid, block,
+                       0, s, g);
+  else
+    hipLaunchKernelGGL((k_gemm_slim<true, true, TM, SPLITK>), grid, block,
+                       0, s, g);
+}
+
 void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
           float alpha, const float* A, long lda, const float* B, long ldb,
           float beta, float* C, long ldc, const GemmEpi* epi,
@@ -662,7 +900,19 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   }
   if (aview) g.av = *aview;
   if (bview) g.bv = *bview;
-  const long tm = (M + BM - 1) / BM, tn = (N + BN - 1) / BN;
+  // small-M slim tile (32x256 / 64x128): every MFMA row a real output row
+  // — the 128^2 tile idles most rows for Cout<=64 layers (inception
+  // branches, stage-1 convs).  fp32 only (the bf16 kernel keeps its own
+  // kwaves path).  CAFFE_GEMM_SLIM=0 reverts.
+  static const int slim_on = [] {
+    const char* e = getenv("CAFFE_GEMM_SLIM");
+    return e ? atoi(e) : 1;
+  }();
+  int TMv = 128;
+  if (slim_on && !Engine::get().gemm_bf16 && M <= 64 && N > 128)
+    TMv = M <= 32 ? 32 : 64;
+  const int TNv = TMv == 128 ? 128 : (TMv == 32 ? 256 : 128);
+  const long tm = (M + TMv - 1) / TMv, tn = (N + TNv - 1) / TNv;
   g.tn = tn;
   g.tiles = tm * tn;
   // split-K when the output grid cannot fill the chip (wgrad shapes);
@@ -704,6 +954,10 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     g.SK = SK;
     if (Engine::get().gemm_bf16) {
       gemm_launch_bf16(s, transA, transB, grid, block, g, true);
+    } else if (TMv == 32) {
+      launch_slim<32, true>(transA, transB, grid, block, s, g);
+    } else if (TMv == 64) {
+      launch_slim<64, true>(transA, transB, grid, block, s, g);
     } else if (!transA && !transB)
       hipLaunchKernelGGL((k_gemm_f32<false, false, true>), grid, block, 0,
                          s, g);
@@ -731,6 +985,14 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   }
   if (Engine::get().gemm_bf16) {
     gemm_launch_bf16(s, transA, transB, grid, block, g, false);
+    return;
+  }
+  if (TMv == 32) {
+    launch_slim<32, false>(transA, transB, grid, block, s, g);
+    return;
+  }
+  if (TMv == 64) {
+    launch_slim<64, false>(transA, transB, grid, block, s, g);
     return;
   }
   // glds v2 path (BK=64, 1 block/CU): NN/TN with plain or channel-view
