@@ -339,7 +339,8 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
 // (register T14 pipeline, one barrier per K-tile, same epilogue/split-K
 // slabs); at TM=64 the LDS footprint (50 KB) admits 3 blocks/CU.
 template <bool TA, bool TB, int TM, bool SPLITK>
-__launch_bounds__(256, 2) __global__ void k_gemm_slim(GemmArgs g) {
+__launch_bounds__(256, TM == 64 ? 3 : 2) __global__
+    void k_gemm_slim(GemmArgs g) {
   constexpr int TN = TM == 32 ? 256 : 128;
   constexpr int AS_S = TM + 1;  // +1 pad: conflict-free b32 banking
   constexpr int BS_S = TN + 1;
@@ -909,7 +910,7 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     return e ? atoi(e) : 1;
   }();
   int TMv = 128;
-  if (slim_on && !Engine::get().gemm_bf16 && M <= 64 && N > 128)
+  if (slim_on && !Engine::get().gemm_bf16 && M <= 64 && N > 64)
     TMv = M <= 32 ? 32 : 64;
   const int TNv = TMv == 128 ? 128 : (TMv == 32 ? 256 : 128);
   const long tm = (M + TMv - 1) / TMv, tn = (N + TNv - 1) / TNv;
