@@ -171,22 +171,16 @@ void axpy(hipStream_t s, long n, float a, const float* x, float* y) {
 }
 
 // ------------------------------------------------------------ pooling
-// 3D launch (x: output pixels of one (n,c) plane, y: C, z: N): the flat
-// grid-stride form spent 3 integer divisions per element — on GoogLeNet's
-// 16 pools that was the largest HBM-class kernel cost (4.9 ms/step)
-__global__ void k_pool_max_fwd(const float* __restrict__ x, int C, int H,
-                               int W, int kh, int kw, int ph, int pw,
+__global__ void k_pool_max_fwd(const float* __restrict__ x, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
                                int sh, int sw, int OH, int OW,
                                float* __restrict__ y, int* __restrict__ mask) {
-  const long nc = (long)blockIdx.z * C + blockIdx.y;
-  const float* xp = x + nc * H * W;
-  float* yp = y + nc * OH * OW;
-  int* mp = mask + nc * OH * OW;
-  const int S = OH * OW;
-  for (int sp = blockIdx.x * blockDim.x + threadIdx.x; sp < S;
-       sp += gridDim.x * blockDim.x) {
-    const int oh = sp / OW;          // single division per element
-    const int ow = sp - oh * OW;
+  const long total = (long)N * C * OH * OW;
+  GRID_STRIDE(idx, total) {
+    const int ow = (int)(idx % OW);
+    const int oh = (int)((idx / OW) % OH);
+    const long nc = idx / ((long)OW * OH);
+    const float* xp = x + nc * H * W;
     int hs = oh * sh - ph, ws = ow * sw - pw;
     const int he = min(hs + kh, H), we = min(ws + kw, W);
     hs = max(hs, 0);
@@ -201,8 +195,8 @@ __global__ void k_pool_max_fwd(const float* __restrict__ x, int C, int H,
           bi = h * W + w;
         }
       }
-    yp[sp] = best;
-    mp[sp] = bi;
+    y[idx] = best;
+    mask[idx] = bi;
   }
 }
 void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
@@ -210,36 +204,35 @@ void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
                   int OW, float* y, int* mask) {
   const long total = (long)N * C * OH * OW;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 4.0 * total * (kh * kw + 2));
-  const int bx = (int)std::min<long>(((long)OH * OW + TPB - 1) / TPB, 8);
-  hipLaunchKernelGGL(k_pool_max_fwd, dim3(bx, C, N), dim3(TPB), 0, s, x, C,
-                     H, W, kh, kw, ph, pw, sh, sw, OH, OW, y, mask);
+  hipLaunchKernelGGL(k_pool_max_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y,
+                     mask);
 }
 
 // deterministic gather backward: each input element scans the <= ceil(k/s)^2
 // windows that can contain it and sums where mask points at it
 __global__ void k_pool_max_bwd(const float* __restrict__ dy,
-                               const int* __restrict__ mask, int C, int H,
-                               int W, int kh, int kw, int ph, int pw,
+                               const int* __restrict__ mask, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
                                int sh, int sw, int OH, int OW,
                                float* __restrict__ dx) {
-  const long nc = (long)blockIdx.z * C + blockIdx.y;
-  const float* dyp = dy + nc * OH * OW;
-  const int* mp = mask + nc * OH * OW;
-  float* dxp = dx + nc * H * W;
-  const int S = H * W;
-  for (int hw = blockIdx.x * blockDim.x + threadIdx.x; hw < S;
-       hw += gridDim.x * blockDim.x) {
-    const int h = hw / W;
-    const int w = hw - h * W;
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int me = h * W + w;
     const int ph0 = (h + ph < kh) ? 0 : (h + ph - kh) / sh + 1;
     const int ph1 = min((h + ph) / sh + 1, OH);
     const int pw0 = (w + pw < kw) ? 0 : (w + pw - kw) / sw + 1;
     const int pw1 = min((w + pw) / sw + 1, OW);
     float acc = 0.f;
+    const float* dyp = dy + nc * OH * OW;
+    const int* mp = mask + nc * OH * OW;
     for (int a = ph0; a < ph1; ++a)
       for (int b = pw0; b < pw1; ++b)
-        if (mp[a * OW + b] == hw) acc += dyp[a * OW + b];
-    dxp[hw] = acc;
+        if (mp[a * OW + b] == me) acc += dyp[a * OW + b];
+    dx[idx] = acc;
   }
 }
 void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
@@ -247,23 +240,21 @@ void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
                   int sh, int sw, int OH, int OW, float* dx) {
   const long total = (long)N * C * H * W;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 12.0 * total);
-  const int bx = (int)std::min<long>(((long)H * W + TPB - 1) / TPB, 8);
-  hipLaunchKernelGGL(k_pool_max_bwd, dim3(bx, C, N), dim3(TPB), 0, s, dy,
-                     mask, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
+  hipLaunchKernelGGL(k_pool_max_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, dy, mask, N, C, H, W, kh, kw, ph, pw, sh, sw, OH,
+                     OW, dx);
 }
 
-__global__ void k_pool_ave_fwd(const float* __restrict__ x, int C, int H,
-                               int W, int kh, int kw, int ph, int pw,
+__global__ void k_pool_ave_fwd(const float* __restrict__ x, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
                                int sh, int sw, int OH, int OW,
                                float* __restrict__ y) {
-  const long nc = (long)blockIdx.z * C + blockIdx.y;
-  const float* xp = x + nc * H * W;
-  float* yp = y + nc * OH * OW;
-  const int S = OH * OW;
-  for (int sp = blockIdx.x * blockDim.x + threadIdx.x; sp < S;
-       sp += gridDim.x * blockDim.x) {
-    const int oh = sp / OW;
-    const int ow = sp - oh * OW;
+  const long total = (long)N * C * OH * OW;
+  GRID_STRIDE(idx, total) {
+    const int ow = (int)(idx % OW);
+    const int oh = (int)((idx / OW) % OH);
+    const long nc = idx / ((long)OW * OH);
+    const float* xp = x + nc * H * W;
     int hs = oh * sh - ph, ws = ow * sw - pw;
     int he = min(hs + kh, H + ph), we = min(ws + kw, W + pw);
     const int ps = (he - hs) * (we - ws);  // padded size (:201)
@@ -274,7 +265,7 @@ __global__ void k_pool_ave_fwd(const float* __restrict__ x, int C, int H,
     float acc = 0.f;
     for (int h = hs; h < he; ++h)
       for (int w = ws; w < we; ++w) acc += xp[h * W + w];
-    yp[sp] = acc / ps;
+    y[idx] = acc / ps;
   }
 }
 void pool_ave_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
@@ -282,28 +273,25 @@ void pool_ave_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
                   int OW, float* y) {
   const long total = (long)N * C * OH * OW;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 4.0 * total * (kh * kw + 1));
-  const int bx = (int)std::min<long>(((long)OH * OW + TPB - 1) / TPB, 8);
-  hipLaunchKernelGGL(k_pool_ave_fwd, dim3(bx, C, N), dim3(TPB), 0, s, x, C,
-                     H, W, kh, kw, ph, pw, sh, sw, OH, OW, y);
+  hipLaunchKernelGGL(k_pool_ave_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y);
 }
 
-__global__ void k_pool_ave_bwd(const float* __restrict__ dy, int C, int H,
-                               int W, int kh, int kw, int ph, int pw,
+__global__ void k_pool_ave_bwd(const float* __restrict__ dy, int N, int C,
+                               int H, int W, int kh, int kw, int ph, int pw,
                                int sh, int sw, int OH, int OW,
                                float* __restrict__ dx) {
-  const long nc = (long)blockIdx.z * C + blockIdx.y;
-  const float* dyp = dy + nc * OH * OW;
-  float* dxp = dx + nc * H * W;
-  const int S = H * W;
-  for (int hw = blockIdx.x * blockDim.x + threadIdx.x; hw < S;
-       hw += gridDim.x * blockDim.x) {
-    const int h = hw / W;
-    const int w = hw - h * W;
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
     const int ph0 = (h + ph < kh) ? 0 : (h + ph - kh) / sh + 1;
     const int ph1 = min((h + ph) / sh + 1, OH);
     const int pw0 = (w + pw < kw) ? 0 : (w + pw - kw) / sw + 1;
     const int pw1 = min((w + pw) / sw + 1, OW);
     float acc = 0.f;
+    const float* dyp = dy + nc * OH * OW;
     for (int a = ph0; a < ph1; ++a)
       for (int b = pw0; b < pw1; ++b) {
         int hs = a * sh - ph, ws = b * sw - pw;
@@ -311,7 +299,7 @@ __global__ void k_pool_ave_bwd(const float* __restrict__ dy, int C, int H,
         const int ps = (he - hs) * (we - ws);
         acc += dyp[a * OW + b] / ps;
       }
-    dxp[hw] = acc;
+    dx[idx] = acc;
   }
 }
 void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
@@ -319,9 +307,8 @@ void pool_ave_bwd(hipStream_t s, const float* dy, int N, int C, int H, int W,
                   int OW, float* dx) {
   const long total = (long)N * C * H * W;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 8.0 * total);
-  const int bx = (int)std::min<long>(((long)H * W + TPB - 1) / TPB, 8);
-  hipLaunchKernelGGL(k_pool_ave_bwd, dim3(bx, C, N), dim3(TPB), 0, s, dy, C,
-                     H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
+  hipLaunchKernelGGL(k_pool_ave_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
+                     s, dy, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, dx);
 }
 
 // y = x * a[c] (+ b[c]) — Scale layer forward (b = bias or null) and its
