@@ -665,31 +665,7 @@ __global__ void k_bn_bwd_stats_v4(const f4* __restrict__ x4,
   };
   double s_dy = 0, s_dyxn = 0, t_dy = 0, t_dyxn = 0;
   int i = threadIdx.x;
-  for (; i + 3 * B < span; i += 4 * B) {  // 8 loads (4 pairs) in flight
-    const long o0 = addr(i), o1 = addr(i + B);
-    const long o2 = addr(i + 2 * B), o3 = addr(i + 3 * B);
-    const f4 xv0 = x4[o0], dv0 = dy4[o0];
-    const f4 xv1 = x4[o1], dv1 = dy4[o1];
-    const f4 xv2 = x4[o2], dv2 = dy4[o2];
-    const f4 xv3 = x4[o3], dv3 = dy4[o3];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float xn0 = (xv0[j] - m) * inv;
-      const float xn1 = (xv1[j] - m) * inv;
-      const float xn2 = (xv2[j] - m) * inv;
-      const float xn3 = (xv3[j] - m) * inv;
-      double d0 = dv0[j], d1 = dv1[j], d2 = dv2[j], d3 = dv3[j];
-      if (frelu && xn0 * sc + bi <= 0.f) d0 = 0.0;
-      if (frelu && xn1 * sc + bi <= 0.f) d1 = 0.0;
-      if (frelu && xn2 * sc + bi <= 0.f) d2 = 0.0;
-      if (frelu && xn3 * sc + bi <= 0.f) d3 = 0.0;
-      s_dy += d0 + d2;
-      s_dyxn += d0 * (double)xn0 + d2 * (double)xn2;
-      t_dy += d1 + d3;
-      t_dyxn += d1 * (double)xn1 + d3 * (double)xn3;
-    }
-  }
-  for (; i + B < span; i += 2 * B) {
+  for (; i + B < span; i += 2 * B) {  // 4 loads (2 pairs) in flight
     const long o0 = addr(i), o1 = addr(i + B);
     const f4 xv0 = x4[o0], dv0 = dy4[o0];
     const f4 xv1 = x4[o1], dv1 = dy4[o1];
@@ -839,69 +815,40 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
 }
 
 // ------------------------------------------------------------ LRN
-// one thread per (n, 4 spatial positions); running cross-channel window
-// (lrn_layer.cu:9-60 shape, restated).  The spatial axis is vectorized
-// with DWORD-aligned float4 loads (S is odd for every AlexNet/GoogLeNet
-// LRN — 55^2/27^2/28^2 — so 16B alignment cannot be assumed; CDNA allows
-// dword-granular dwordx4) — 4 independent channel walks per thread was
-// the missing memory-level parallelism (AlexNet LRN was 5.9 ms/step).
-using f4u = __attribute__((ext_vector_type(4), aligned(4))) float;
+// one thread per (n, s); running cross-channel window (lrn_layer.cu:9-60
+// shape, restated)
 __global__ void k_lrn_fwd(const float* __restrict__ x, int N, int C, long S,
                           int size, float aos, float beta, float k,
                           float* __restrict__ scale, float* __restrict__ y) {
-  const long S4 = (S + 3) / 4;
-  const long total = (long)N * S4;
+  const long total = (long)N * S;
   const int pre = (size - 1) / 2;
   GRID_STRIDE(idx, total) {
-    const int n = (int)(idx / S4);
-    const long sp = (idx - (long)n * S4) * 4;
-    const int nv = (int)min((long)4, S - sp);  // valid lanes
+    const int n = (int)(idx / S);
+    const long sp = idx - (long)n * S;
     const float* xp = x + (long)n * C * S + sp;
     float* scp = scale + (long)n * C * S + sp;
     float* yp = y + (long)n * C * S + sp;
-    f4u acc = {0.f, 0.f, 0.f, 0.f};
-    auto ld = [&](long off) -> f4u {
-      if (nv == 4) return *(const f4u*)(xp + off);
-      f4u v = {0.f, 0.f, 0.f, 0.f};
-      for (int j = 0; j < nv; ++j) v[j] = xp[off + j];
-      return v;
-    };
+    float acc = 0.f;
     for (int c = 0; c < size - pre && c < C; ++c) {
-      const f4u v = ld((long)c * S);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) acc[j] += v[j] * v[j];
+      const float v = xp[(long)c * S];
+      acc += v * v;
     }
     for (int c = 0; c < C; ++c) {
       if (c > 0) {
         const int head = c + size - 1 - pre;
         if (head < C) {
-          const f4u v = ld((long)head * S);
-#pragma unroll
-          for (int j = 0; j < 4; ++j) acc[j] += v[j] * v[j];
+          const float v = xp[(long)head * S];
+          acc += v * v;
         }
         const int tail = c - 1 - pre;
         if (tail >= 0) {
-          const f4u v = ld((long)tail * S);
-#pragma unroll
-          for (int j = 0; j < 4; ++j) acc[j] -= v[j] * v[j];
+          const float v = xp[(long)tail * S];
+          acc -= v * v;
         }
       }
-      const f4u xv = ld((long)c * S);
-      f4u sc, yv;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        sc[j] = k + aos * acc[j];
-        yv[j] = xv[j] * __powf(sc[j], -beta);
-      }
-      if (nv == 4) {
-        *(f4u*)(scp + (long)c * S) = sc;
-        *(f4u*)(yp + (long)c * S) = yv;
-      } else {
-        for (int j = 0; j < nv; ++j) {
-          scp[(long)c * S + j] = sc[j];
-          yp[(long)c * S + j] = yv[j];
-        }
-      }
+      const float sc = k + aos * acc;
+      scp[(long)c * S] = sc;
+      yp[(long)c * S] = xp[(long)c * S] * __powf(sc, -beta);
     }
   }
 }
@@ -910,9 +857,8 @@ void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
              float* y) {
   const long S = (long)H * W;
   PerfScope perf(PERF_CLASS("lrn"), s, 0, 12.0 * N * C * S);
-  hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks(N * ((S + 3) / 4))), dim3(TPB),
-                     0, s, x, N, C, S, size, alpha / size, beta, k, scale,
-                     y);
+  hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, N,
+                     C, S, size, alpha / size, beta, k, scale, y);
 }
 
 __global__ void k_lrn_bwd(const float* __restrict__ x,
@@ -921,48 +867,33 @@ __global__ void k_lrn_bwd(const float* __restrict__ x,
                           const float* __restrict__ scale, int N, int C,
                           long S, int size, float cr, float beta,
                           float* __restrict__ dx) {
-  const long S4 = (S + 3) / 4;
-  const long total = (long)N * S4;
+  const long total = (long)N * S;
   const int pre = (size - 1) / 2;
   GRID_STRIDE(idx, total) {
-    const int n = (int)(idx / S4);
-    const long sp = (idx - (long)n * S4) * 4;
-    const int nv = (int)min((long)4, S - sp);
+    const int n = (int)(idx / S);
+    const long sp = idx - (long)n * S;
     const long base = (long)n * C * S + sp;
-    auto ld = [&](const float* p, long off) -> f4u {
-      if (nv == 4) return *(const f4u*)(p + base + off);
-      f4u v = {0.f, 0.f, 0.f, 0.f};
-      for (int j = 0; j < nv; ++j) v[j] = p[base + off + j];
-      return v;
-    };
     // ratio(c) = dy*y/scale; window for dx[c]: cc in [c-(size-1-pre), c+pre]
-    f4u acc = {0.f, 0.f, 0.f, 0.f};
+    float acc = 0.f;
     const int lo0 = -(size - 1 - pre);
-    auto ratio_add = [&](int cc, float sgn) {
-      const long off = (long)cc * S;
-      const f4u dv = ld(dy, off), yv = ld(y, off), sv = ld(scale, off);
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[j] += sgn * dv[j] * yv[j] / (nv == 4 || j < nv ? sv[j] : 1.f);
-    };
     for (int cc = lo0; cc <= pre - 1; ++cc)
-      if (cc >= 0 && cc < C) ratio_add(cc, 1.f);
+      if (cc >= 0 && cc < C) {
+        const long i = base + (long)cc * S;
+        acc += dy[i] * y[i] / scale[i];
+      }
     for (int c = 0; c < C; ++c) {
       const int head = c + pre;
-      if (head >= 0 && head < C) ratio_add(head, 1.f);
-      const long off = (long)c * S;
-      const f4u dv = ld(dy, off), xv = ld(x, off), sv = ld(scale, off);
-      f4u out;
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        out[j] = dv[j] * __powf(nv == 4 || j < nv ? sv[j] : 1.f, -beta) -
-                 cr * xv[j] * acc[j];
-      if (nv == 4)
-        *(f4u*)(dx + base + off) = out;
-      else
-        for (int j = 0; j < nv; ++j) dx[base + off + j] = out[j];
+      if (head >= 0 && head < C) {
+        const long i = base + (long)head * S;
+        acc += dy[i] * y[i] / scale[i];
+      }
+      const long i = base + (long)c * S;
+      dx[i] = dy[i] * __powf(scale[i], -beta) - cr * x[i] * acc;
       const int tail = c + lo0;
-      if (tail >= 0 && tail < C) ratio_add(tail, -1.f);
+      if (tail >= 0 && tail < C) {
+        const long j = base + (long)tail * S;
+        acc -= dy[j] * y[j] / scale[j];
+      }
     }
   }
 }
@@ -971,9 +902,9 @@ void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
              float alpha, float beta, float* dx) {
   const long S = (long)H * W;
   PerfScope perf(PERF_CLASS("lrn"), s, 0, 20.0 * N * C * S);
-  hipLaunchKernelGGL(k_lrn_bwd, dim3(nblocks(N * ((S + 3) / 4))), dim3(TPB),
-                     0, s, x, y, dy, scale, N, C, S, size,
-                     2.f * alpha * beta / size, beta, dx);
+  hipLaunchKernelGGL(k_lrn_bwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, y,
+                     dy, scale, N, C, S, size, 2.f * alpha * beta / size,
+                     beta, dx);
 }
 
 // ------------------------------------------------------------ softmax
