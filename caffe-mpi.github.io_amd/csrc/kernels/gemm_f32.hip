@@ -909,9 +909,11 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     const char* e = getenv("CAFFE_GEMM_SLIM");
     return e ? atoi(e) : 1;
   }();
+  // the slim tile also serves bf16 mode: M<=64 contractions run the fp32
+  // slim kernel (better utilization than the bf16 kwaves path AND exact
+  // arithmetic — AMP-style per-shape heterogeneity)
   int TMv = 128;
-  if (slim_on && !Engine::get().gemm_bf16 && M <= 64 && N > 64)
-    TMv = M <= 32 ? 32 : 64;
+  if (slim_on && M <= 64 && N > 64) TMv = M <= 32 ? 32 : 64;
   const int TNv = TMv == 128 ? 128 : (TMv == 32 ? 256 : 128);
   const long tm = (M + TMv - 1) / TMv, tn = (N + TNv - 1) / TNv;
   g.tn = tn;
@@ -953,12 +955,12 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
         10, sizeof(float) * (size_t)SK * M * N);
     g.slab = slab;
     g.SK = SK;
-    if (Engine::get().gemm_bf16) {
-      gemm_launch_bf16(s, transA, transB, grid, block, g, true);
-    } else if (TMv == 32) {
+    if (TMv == 32) {
       launch_slim<32, true>(transA, transB, grid, block, s, g);
     } else if (TMv == 64) {
       launch_slim<64, true>(transA, transB, grid, block, s, g);
+    } else if (Engine::get().gemm_bf16) {
+      gemm_launch_bf16(s, transA, transB, grid, block, g, true);
     } else if (!transA && !transB)
       hipLaunchKernelGGL((k_gemm_f32<false, false, true>), grid, block, 0,
                          s, g);
@@ -984,16 +986,16 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
                        MN, SK, C);
     return;
   }
-  if (Engine::get().gemm_bf16) {
-    gemm_launch_bf16(s, transA, transB, grid, block, g, false);
-    return;
-  }
   if (TMv == 32) {
     launch_slim<32, false>(transA, transB, grid, block, s, g);
     return;
   }
   if (TMv == 64) {
     launch_slim<64, false>(transA, transB, grid, block, s, g);
+    return;
+  }
+  if (Engine::get().gemm_bf16) {
+    gemm_launch_bf16(s, transA, transB, grid, block, g, false);
     return;
   }
   // glds v2 path (BK=64, 1 block/CU): NN/TN with plain or channel-view

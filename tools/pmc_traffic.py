@@ -28,6 +28,9 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 # kernel-name prefix -> perf class (keep in sync with csrc perf classes)
 CLASSES = [
     ("k_gemm_f32", "gemm"),
+    ("k_gemm_slim", "gemm"),
+    ("k_gemm_bf16", "gemm"),
+    ("k_gemm2", "gemm"),
     ("k_splitk_reduce", "gemm"),      # part of the split-K GEMM episode
     ("k_weight_flip", "gemm"),
     ("k_bn_", "bn"),
