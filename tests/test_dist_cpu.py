@@ -32,7 +32,7 @@ def _free_port():
     return port
 
 
-def run_dist(nproc, args):
+def run_dist(nproc, args, worker=WORKER):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
     env["MASTER_PORT"] = str(_free_port())
@@ -41,7 +41,7 @@ def run_dist(nproc, args):
         env_r = dict(env, RANK=str(rank), WORLD_SIZE=str(nproc),
                      LOCAL_RANK=str(rank))
         procs.append(subprocess.Popen(
-            [sys.executable, WORKER] + args, env=env_r,
+            [sys.executable, worker] + args, env=env_r,
             stdout=subprocess.PIPE, stderr=subprocess.STDOUT, cwd=REPO))
     outs = []
     for p in procs:

@@ -191,6 +191,29 @@ def test_overflow_values():
         assert relerr(data[j], raw) < 1e-6
 
 
+def test_dist_lmdb_shard_equivalence(db):
+    # world-2 LMDB-sharded training (gloo callback comm) == 1-rank on the
+    # doubled batch: the per-iteration record SET is identical (rank r owns
+    # (iter*B+j)*2+r), so averaged gradients match — the reference's
+    # effective-batch equivalence extended to the LMDB feed.  Full-image
+    # crops keep the rank-keyed crop RNG out of the comparison.
+    import subprocess
+    import sys as _sys
+    from test_dist_cpu import parse_params, run_dist
+
+    worker = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                          "_dist_lmdb_worker.py")
+    out2 = run_dist(2, [db, "4", "6"], worker=worker)
+    out1 = subprocess.run(
+        [_sys.executable, worker, db, "8", "6"],
+        env=dict(os.environ, RANK="0", WORLD_SIZE="1"), cwd=REPO,
+        capture_output=True, text=True, timeout=600)
+    assert out1.returncode == 0, out1.stdout + out1.stderr
+    p2 = parse_params(out2[0])
+    p1 = parse_params(out1.stdout)
+    assert np.abs(p2 - p1).max() < 1e-4 * max(1.0, np.abs(p1).max())
+
+
 @pytest.mark.gpu
 def test_lmdb_gpu_transform_parity(db):
     # GPU path (pinned upload + k_transform_u8) against the CPU transform
