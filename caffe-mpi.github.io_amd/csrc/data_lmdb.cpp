@@ -13,6 +13,7 @@
 // != 0 the wraparound rotates ownership across epochs, which keeps every
 // record in use — the reference's full-cycle guarantee).
 #include <condition_variable>
+#include <functional>
 #include <fstream>
 #include <mutex>
 #include <thread>
@@ -61,6 +62,14 @@ static Datum decode_datum(const uint8_t* p, size_t n) {
 
 struct LmdbFeed {
   LmdbReader reader;
+  // generalized source hooks (ImageData reuses the whole feed machinery):
+  //  provider: fills one record's uint8 image + label (default: LMDB
+  //    record at `idx` decoded as a Datum)
+  //  index_of: maps (iter, j) to a record index (default: the rank-cycle
+  //    shard; ImageData reads the FULL list per rank like the reference,
+  //    with an optional per-epoch keyed shuffle)
+  std::function<void(long idx, uint8_t* dst, float* label)> provider;
+  std::function<long(uint64_t iter, int j)> index_of;
   long nrec = 0;
   int dc = 0, dh = 0, dw = 0;   // datum dims
   int crop = 0;                  // 0 = full image
@@ -122,16 +131,22 @@ struct LmdbFeed {
         h_splitmix64(E.seed ^ 0x17DBull ^ ((uint64_t)E.rank << 40) ^ iter);
     for (int j = 0; j < batch; ++j) {
       const long g =
-          (long)((((iter * (uint64_t)batch + j) * (uint64_t)E.world) +
-                  E.rank) %
-                 (uint64_t)nrec);
-      auto rec = reader.at(g);
-      Datum d = decode_datum(rec.first, rec.second);
-      CHECK_EQ_(d.c, dc);
-      CHECK_EQ_(d.h, dh);
-      CHECK_EQ_(d.w, dw);
-      memcpy(s.data.data() + (size_t)j * imsz, d.data, imsz);
-      s.labels[j] = (float)d.label;
+          index_of ? index_of(iter, j)
+                   : (long)((((iter * (uint64_t)batch + j) *
+                              (uint64_t)E.world) +
+                             E.rank) %
+                            (uint64_t)nrec);
+      if (provider) {
+        provider(g, s.data.data() + (size_t)j * imsz, &s.labels[j]);
+      } else {
+        auto rec = reader.at(g);
+        Datum d = decode_datum(rec.first, rec.second);
+        CHECK_EQ_(d.c, dc);
+        CHECK_EQ_(d.h, dh);
+        CHECK_EQ_(d.w, dw);
+        memcpy(s.data.data() + (size_t)j * imsz, d.data, imsz);
+        s.labels[j] = (float)d.label;
+      }
       int ho = 0, wo = 0, mir = 0;
       if (phase == Phase::TRAIN) {  // random crop + mirror (train)
         const uint64_t hj = h_splitmix64(key ^ (uint64_t)(3 * j + 1));
@@ -217,6 +232,10 @@ struct LmdbFeed {
   }
 };
 
+// shared transform_param parsing (Data + ImageData): crop/mirror/scale
+// + mean_value or mean_file (binaryproto, per-pixel)
+static void setup_transform(LmdbFeed& F, const PMsgPtr& tp);
+
 void DataLayer::setup_lmdb(const std::string& source) {
   feed_.reset(new LmdbFeed());
   LmdbFeed& F = *feed_;
@@ -230,7 +249,14 @@ void DataLayer::setup_lmdb(const std::string& source) {
   F.dw = d0.w;
   F.batch = batch_;
   F.phase = phase_;
-  auto tp = param_->sub("transform_param");
+  setup_transform(F, param_->sub("transform_param"));
+  C_ = F.dc;
+  H_ = F.outH;
+  W_ = F.outW;
+  F.start_worker();
+}
+
+static void setup_transform(LmdbFeed& F, const PMsgPtr& tp) {
   F.crop = tp ? (int)tp->inum("crop_size", 0) : 0;
   F.mirror = tp && tp->boolean("mirror", false);
   F.scale = tp ? (float)tp->num("scale", 1.0) : 1.f;
@@ -258,10 +284,6 @@ void DataLayer::setup_lmdb(const std::string& source) {
       F.mean_per_pixel = false;
     }
   }
-  C_ = F.dc;
-  H_ = F.outH;
-  W_ = F.outW;
-  F.start_worker();
 }
 
 void DataLayer::forward_lmdb_cpu(const std::vector<Blob*>& top) {
@@ -327,5 +349,189 @@ void DataLayer::forward_lmdb_gpu(const std::vector<Blob*>& top) {
                            E.stream));
   HIP_CHECK(hipEventRecord(F.ev[p], E.stream));
 }
+
+
+// ================================================================
+// ImageData layer (reference src/caffe/layers/image_data_layer.cpp):
+// image_data_param { source: listfile ("path label" per line),
+// batch_size, new_height/new_width (bilinear resize), shuffle,
+// root_folder } + the shared transform_param.  The image codecs are
+// PPM (P6) and PGM (P5) — binary netpbm, the formats decodable without
+// an image library (none exists in this environment; JPEG/PNG records
+// fail with a clear message).  Like the reference, every rank reads the
+// FULL list; shuffle is a per-epoch Fisher-Yates keyed by
+// (seed, rank, epoch) so runs are reproducible and ranks decorrelate.
+namespace {
+
+struct PnmImage {
+  int c = 0, h = 0, w = 0;
+  std::vector<uint8_t> px;  // CHW
+};
+
+PnmImage load_pnm(const std::string& path) {
+  std::ifstream f(path, std::ios::binary);
+  CHECK_(f.good()) << "cannot open image " << path;
+  std::string magic;
+  f >> magic;
+  CHECK_(magic == "P6" || magic == "P5")
+      << "unsupported image format '" << magic << "' in " << path
+      << " (only binary PPM/PGM decodable in this environment — "
+         "no JPEG/PNG codec in the image)";
+  auto next_int = [&]() {
+    int v;
+    // skip whitespace/comments (netpbm allows # comments in the header)
+    while (true) {
+      int ch = f.peek();
+      if (ch == '#') {
+        std::string line;
+        std::getline(f, line);
+      } else if (isspace(ch)) {
+        f.get();
+      } else {
+        break;
+      }
+    }
+    f >> v;
+    return v;
+  };
+  PnmImage im;
+  im.w = next_int();
+  im.h = next_int();
+  const int maxv = next_int();
+  CHECK_EQ_(maxv, 255) << "only 8-bit netpbm supported: " << path;
+  f.get();  // single whitespace before raster
+  im.c = magic == "P6" ? 3 : 1;
+  std::vector<uint8_t> raw((size_t)im.w * im.h * im.c);
+  f.read((char*)raw.data(), (long)raw.size());
+  CHECK_(f.good()) << "truncated image " << path;
+  // interleaved HWC -> planar CHW (the engine/caffe layout)
+  im.px.resize(raw.size());
+  for (int y = 0; y < im.h; ++y)
+    for (int x = 0; x < im.w; ++x)
+      for (int ch = 0; ch < im.c; ++ch)
+        im.px[((size_t)ch * im.h + y) * im.w + x] =
+            raw[((size_t)y * im.w + x) * im.c + ch];
+  return im;
+}
+
+// bilinear uint8 resize, CHW (the reference resizes via cv::resize
+// INTER_LINEAR when new_height/new_width are set)
+void resize_bilinear(const PnmImage& src, int nh, int nw,
+                     std::vector<uint8_t>& dst) {
+  dst.resize((size_t)src.c * nh * nw);
+  const float sy = (float)src.h / nh, sx = (float)src.w / nw;
+  for (int c = 0; c < src.c; ++c) {
+    const uint8_t* sp = src.px.data() + (size_t)c * src.h * src.w;
+    uint8_t* dp = dst.data() + (size_t)c * nh * nw;
+    for (int y = 0; y < nh; ++y) {
+      const float fy = (y + 0.5f) * sy - 0.5f;
+      int y0 = (int)floorf(fy);
+      const float wy = fy - y0;
+      y0 = std::max(0, std::min(src.h - 1, y0));
+      const int y1 = std::min(src.h - 1, y0 + 1);
+      for (int x = 0; x < nw; ++x) {
+        const float fx = (x + 0.5f) * sx - 0.5f;
+        int x0 = (int)floorf(fx);
+        const float wx = fx - x0;
+        x0 = std::max(0, std::min(src.w - 1, x0));
+        const int x1 = std::min(src.w - 1, x0 + 1);
+        const float v =
+            (1 - wy) * ((1 - wx) * sp[y0 * src.w + x0] +
+                        wx * sp[y0 * src.w + x1]) +
+            wy * ((1 - wx) * sp[y1 * src.w + x0] +
+                  wx * sp[y1 * src.w + x1]);
+        dp[y * nw + x] = (uint8_t)(v + 0.5f);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+class ImageDataLayer : public DataLayer {
+ public:
+  using DataLayer::DataLayer;
+  void LayerSetUp(const std::vector<Blob*>&,
+                  const std::vector<Blob*>&) override {
+    auto ip = param_->sub("image_data_param");
+    CHECK_(ip) << "ImageData layer needs image_data_param";
+    batch_ = (int)ip->inum("batch_size", 1);
+    const std::string root = ip->str("root_folder", "");
+    const std::string list = ip->str("source");
+    std::ifstream f(list);
+    CHECK_(f.good()) << "cannot read image list " << list;
+    std::string path;
+    long label;
+    while (f >> path >> label) lines_.emplace_back(root + path, label);
+    CHECK_GT_((long)lines_.size(), 0) << "empty image list " << list;
+    nh_ = (int)ip->inum("new_height", 0);
+    nw_ = (int)ip->inum("new_width", 0);
+    shuffle_ = ip->boolean("shuffle", false);
+    PnmImage first = load_pnm(lines_[0].first);
+    const int dh = nh_ > 0 ? nh_ : first.h;
+    const int dw = nw_ > 0 ? nw_ : first.w;
+
+    feed_.reset(new LmdbFeed());
+    LmdbFeed& F = *feed_;
+    F.nrec = (long)lines_.size();
+    F.dc = first.c;
+    F.dh = dh;
+    F.dw = dw;
+    F.batch = batch_;
+    F.phase = phase_;
+    setup_transform(F, param_->sub("transform_param"));
+    C_ = F.dc;
+    H_ = F.outH;
+    W_ = F.outW;
+    // full list per rank (reference semantics); optional keyed shuffle
+    F.index_of = [this](uint64_t iter, int j) {
+      const uint64_t pos = iter * (uint64_t)batch_ + j;
+      const long n = (long)lines_.size();
+      const long epoch = (long)(pos / (uint64_t)n);
+      const long i = (long)(pos % (uint64_t)n);
+      if (!shuffle_) return i;
+      if (epoch != perm_epoch_) reshuffle(epoch);
+      return perm_[i];
+    };
+    F.provider = [this, &F](long idx, uint8_t* dst, float* label) {
+      PnmImage im = load_pnm(lines_[idx].first);
+      CHECK_EQ_(im.c, F.dc) << lines_[idx].first;
+      if (nh_ > 0 || nw_ > 0) {
+        std::vector<uint8_t> rs;
+        resize_bilinear(im, F.dh, F.dw, rs);
+        memcpy(dst, rs.data(), rs.size());
+      } else {
+        CHECK_EQ_(im.h, F.dh) << lines_[idx].first;
+        CHECK_EQ_(im.w, F.dw) << lines_[idx].first;
+        memcpy(dst, im.px.data(), im.px.size());
+      }
+      *label = (float)lines_[idx].second;
+    };
+    F.start_worker();
+  }
+
+ private:
+  void reshuffle(long epoch) {
+    Engine& E = Engine::get();
+    const long n = (long)lines_.size();
+    perm_.resize(n);
+    for (long i = 0; i < n; ++i) perm_[i] = i;
+    uint64_t key = h_splitmix64(E.seed ^ 0x1A6Eull ^
+                                ((uint64_t)E.rank << 40) ^
+                                (uint64_t)epoch);
+    for (long i = n - 1; i > 0; --i) {  // keyed Fisher-Yates
+      key = h_splitmix64(key);
+      std::swap(perm_[i], perm_[key % (uint64_t)(i + 1)]);
+    }
+    perm_epoch_ = epoch;
+  }
+  std::vector<std::pair<std::string, long>> lines_;
+  std::vector<long> perm_;
+  long perm_epoch_ = -1;
+  int nh_ = 0, nw_ = 0;
+  bool shuffle_ = false;
+};
+
+REGISTER_LAYER("ImageData", ImageDataLayer)
 
 }  // namespace camd
