@@ -905,15 +905,15 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // — the 128^2 tile idles most rows for Cout<=64 layers (inception
   // branches, stage-1 convs).  fp32 only (the bf16 kernel keeps its own
   // kwaves path).  CAFFE_GEMM_SLIM=0 reverts.
-  static const int slim_on = [] {
-    const char* e = getenv("CAFFE_GEMM_SLIM");
-    return e ? atoi(e) : 1;
+  static const int slim_maxm = [] {
+    const char* e = getenv("CAFFE_GEMM_SLIM");  // 0 = off, else max M
+    return e ? atoi(e) : 64;
   }();
   // the slim tile also serves bf16 mode: M<=64 contractions run the fp32
   // slim kernel (better utilization than the bf16 kwaves path AND exact
   // arithmetic — AMP-style per-shape heterogeneity)
   int TMv = 128;
-  if (slim_on && M <= 64 && N > 64) TMv = M <= 32 ? 32 : 64;
+  if (M <= slim_maxm && N > 64) TMv = M <= 32 ? 32 : 64;
   const int TNv = TMv == 128 ? 128 : (TMv == 32 ? 256 : 128);
   const long tm = (M + TMv - 1) / TMv, tn = (N + TNv - 1) / TNv;
   g.tn = tn;
