@@ -162,24 +162,13 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
       {
         const float* Ab = As(cur);
         const float* Bb = Bs(cur);
-        // explicit one-step LDS rotation: kk+2's operands are read while
-        // kk's MFMAs occupy the matrix pipe, breaking the ds_read->mfma
-        // dependency the in-order wave would otherwise stall on
-        const int r0 = ksel;
-        float a0 = Ab[r0 * LDS_S + wr * 64 + row_in];
-        float a1 = Ab[r0 * LDS_S + wr * 64 + 32 + row_in];
-        float b0 = Bb[r0 * LDS_S + wc * 64 + row_in];
-        float b1 = Bb[r0 * LDS_S + wc * 64 + 32 + row_in];
 #pragma unroll
         for (int kk = 0; kk < BK; kk += 2) {
-          float na0, na1, nb0, nb1;
-          if (kk + 2 < BK) {
-            const int nrow = kk + 2 + ksel;
-            na0 = Ab[nrow * LDS_S + wr * 64 + row_in];
-            na1 = Ab[nrow * LDS_S + wr * 64 + 32 + row_in];
-            nb0 = Bb[nrow * LDS_S + wc * 64 + row_in];
-            nb1 = Bb[nrow * LDS_S + wc * 64 + 32 + row_in];
-          }
+          const int krow = kk + ksel;
+          const float a0 = Ab[krow * LDS_S + wr * 64 + row_in];
+          const float a1 = Ab[krow * LDS_S + wr * 64 + 32 + row_in];
+          const float b0 = Bb[krow * LDS_S + wc * 64 + row_in];
+          const float b1 = Bb[krow * LDS_S + wc * 64 + 32 + row_in];
           acc00 =
               __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc00, 0, 0, 0);
           acc01 =
@@ -188,12 +177,6 @@ __launch_bounds__(256, 2) __global__ void k_gemm_f32(GemmArgs g) {
               __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc10, 0, 0, 0);
           acc11 =
               __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc11, 0, 0, 0);
-          if (kk + 2 < BK) {
-            a0 = na0;
-            a1 = na1;
-            b0 = nb0;
-            b1 = nb1;
-          }
         }
       }
       // single barrier per K-tile: writing buf[cur^1] here is safe — its
