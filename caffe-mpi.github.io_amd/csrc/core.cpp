@@ -7,6 +7,11 @@ Engine& Engine::get() {
   return e;
 }
 
+const char*& error_context() {
+  thread_local const char* ctx = nullptr;
+  return ctx;
+}
+
 void Engine::set_mode_gpu(int dev) {
   mode = Mode::GPU;
   device = dev;

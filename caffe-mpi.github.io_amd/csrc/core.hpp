@@ -33,6 +33,12 @@
 namespace camd {
 
 // ---------------------------------------------------------------- logging
+// error-context breadcrumb: Net sets this to "layer 'conv1' (Convolution)
+// Reshape" etc. before dispatching into layer code, so a CHECK failure
+// deep in Blob/kernel plumbing names the layer it fired under (round-1
+// review: a bare "CHECK failed: (d) >= (0)" was undiagnosable)
+const char*& error_context();
+
 struct FatalStream {
   std::ostringstream ss;
   const char* file;
@@ -40,6 +46,11 @@ struct FatalStream {
   FatalStream(const char* f, int l) : file(f), line(l) {}
   [[noreturn]] ~FatalStream() noexcept(false) {
     std::string msg = ss.str();
+    if (error_context()) {
+      msg += " [in ";
+      msg += error_context();
+      msg += "]";
+    }
     fprintf(stderr, "[caffe_amd FATAL %s:%d] %s\n", file, line, msg.c_str());
     throw std::runtime_error(msg);
   }

@@ -127,6 +127,9 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
         top.push_back(nb.get());
       }
     }
+    const std::string ctx_s =
+        "layer '" + layer->name() + "' (" + layer->type() + ") SetUp";
+    error_context() = ctx_s.c_str();
     if (batch_override > 0 && layer->type() == "Data") {
       // per-rank batch override (reference divides the prototxt batch
       // across GPUs, parallel.cpp:284-348)
@@ -136,6 +139,7 @@ void Net::init(const PMsgPtr& msg, int batch_override) {
     } else {
       layer->SetUp(bottom, top);
     }
+    error_context() = nullptr;
     // loss weights
     auto lw = lm->nums("loss_weight");
     for (size_t i = 0; i < top.size(); ++i) {
@@ -304,7 +308,13 @@ void Net::setup_arena() {
 
 void Net::Forward() {
   for (size_t i = 0; i < layers_.size(); ++i)
-    layers_[i]->Forward(bottoms_[i], tops_[i]);
+    {
+      const std::string ctx_s = "layer '" + layers_[i]->name() + "' (" +
+                                layers_[i]->type() + ") Forward";
+      error_context() = ctx_s.c_str();
+      layers_[i]->Forward(bottoms_[i], tops_[i]);
+      error_context() = nullptr;
+    }
 }
 
 void Net::Backward(ReduceHook* hook) {
@@ -314,7 +324,13 @@ void Net::Backward(ReduceHook* hook) {
   hipEvent_t last_ev = nullptr;
   for (int i = (int)layers_.size() - 1; i >= 0; --i) {
     if (layer_need_bwd_[i])
-      layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+      {
+        const std::string ctx_s = "layer '" + layers_[i]->name() + "' (" +
+                                  layers_[i]->type() + ") Backward";
+        error_context() = ctx_s.c_str();
+        layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+        error_context() = nullptr;
+      }
     if (hook) {
       // emit this layer's params (consecutive in params_, ascending offset)
       const size_t start = pidx;
@@ -474,14 +490,26 @@ void Net::time_layers(int iters) {
   for (int it = 0; it < iters; ++it) {
     for (size_t i = 0; i < L; ++i) {
       auto t0 = tick();
+      {
+      const std::string ctx_s = "layer '" + layers_[i]->name() + "' (" +
+                                layers_[i]->type() + ") Forward";
+      error_context() = ctx_s.c_str();
       layers_[i]->Forward(bottoms_[i], tops_[i]);
+      error_context() = nullptr;
+    }
       auto t1 = tick();
       fwd_ms[i] += std::chrono::duration<double, std::milli>(t1 - t0).count();
     }
     for (size_t i = L; i-- > 0;) {
       if (!layer_need_bwd_[i]) continue;
       auto t0 = tick();
-      layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+      {
+        const std::string ctx_s = "layer '" + layers_[i]->name() + "' (" +
+                                  layers_[i]->type() + ") Backward";
+        error_context() = ctx_s.c_str();
+        layers_[i]->Backward(tops_[i], prop_down_[i], bottoms_[i]);
+        error_context() = nullptr;
+      }
       auto t1 = tick();
       bwd_ms[i] += std::chrono::duration<double, std::milli>(t1 - t0).count();
     }

@@ -87,3 +87,19 @@ layer { name: "r" type: "ReLU" bottom: "mid" top: "out" }""")
             net_from_text(full[:cut])
         except Exception:
             pass  # clean failure is the contract
+
+
+def test_error_names_the_layer():
+    # round-2 diagnostics: a CHECK failing deep in blob plumbing carries a
+    # "[in layer '<name>' (<type>) <stage>]" breadcrumb (the round-1
+    # review hit a bare "CHECK failed: (d) >= (0)" with no context)
+    import caffe_amd as ca
+    from engine_util import input_net, net_from_text
+    ca.set_mode("cpu")
+    body = """layer { name: "badconv" type: "Convolution" bottom: "in0"
+  top: "out" convolution_param { num_output: 4 kernel_size: 9 } }"""
+    try:
+        net_from_text(input_net([(1, 3, 5, 5)], body))
+        raise AssertionError("expected a shape failure")
+    except ca.CaffeError as e:
+        assert "badconv" in str(e), str(e)
