@@ -303,6 +303,91 @@ int caffe_net_blob_set(caffe_net_t n, const char* name, int diff,
   API_CATCH
 }
 
+// ---- pycaffe-shim surface (round 2): zero-copy CPU pointers with the
+// SyncedMemory head semantics pycaffe relied on (mutable access marks the
+// host copy dirty; the next GPU use re-syncs), plus layer enumeration so
+// `net.blobs` / `net.params` / `net.layers` can be dict-shaped.
+float* caffe_net_blob_cpu_ptr(caffe_net_t n, const char* name, int diff,
+                              int writable) {
+  try {
+    Blob* b = N(n)->blob_by_name(name);
+    CHECK_(b) << "no blob " << name;
+    Engine::get().sync();
+    if (writable)
+      return diff ? b->mutable_cpu_diff() : b->mutable_cpu_data();
+    return const_cast<float*>(diff ? b->cpu_diff() : b->cpu_data());
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
+
+int caffe_net_num_layers(caffe_net_t n) {
+  return (int)N(n)->layers().size();
+}
+
+// newline-joined blob names in topological order
+int caffe_net_blob_names(caffe_net_t n, char* out, int cap) {
+  API_TRY
+  std::string joined;
+  for (auto& nm : N(n)->blob_names()) {
+    if (!joined.empty()) joined += "\n";
+    joined += nm;
+  }
+  snprintf(out, cap, "%s", joined.c_str());
+  return 0;
+  API_CATCH
+}
+
+int caffe_net_layer_info(caffe_net_t n, int idx, char* name_out,
+                         int name_cap, char* type_out, int type_cap,
+                         int* num_blobs_out) {
+  API_TRY
+  auto& ls = N(n)->layers();
+  CHECK_LT_(idx, (int)ls.size());
+  snprintf(name_out, name_cap, "%s", ls[idx]->name().c_str());
+  snprintf(type_out, type_cap, "%s", ls[idx]->type().c_str());
+  *num_blobs_out = (int)ls[idx]->blobs().size();
+  return 0;
+  API_CATCH
+}
+
+static Layer* find_layer(caffe_net_t n, const char* lname) {
+  for (auto& l : N(n)->layers())
+    if (l->name() == lname) return l.get();
+  CAMD_FATAL << "no layer " << lname;
+}
+
+int caffe_net_layer_blob_shape(caffe_net_t n, const char* lname, int bidx,
+                               int* shape_out, int max_dims,
+                               int* ndims_out) {
+  API_TRY
+  Layer* l = find_layer(n, lname);
+  CHECK_LT_(bidx, (int)l->blobs().size());
+  const auto& s = l->blobs()[bidx]->shape();
+  *ndims_out = (int)s.size();
+  for (int i = 0; i < (int)s.size() && i < max_dims; ++i)
+    shape_out[i] = s[i];
+  return 0;
+  API_CATCH
+}
+
+float* caffe_net_layer_blob_cpu_ptr(caffe_net_t n, const char* lname,
+                                    int bidx, int diff, int writable) {
+  try {
+    Layer* l = find_layer(n, lname);
+    CHECK_LT_(bidx, (int)l->blobs().size());
+    Blob* b = l->blobs()[bidx].get();
+    Engine::get().sync();
+    if (writable)
+      return diff ? b->mutable_cpu_diff() : b->mutable_cpu_data();
+    return const_cast<float*>(diff ? b->cpu_diff() : b->cpu_data());
+  } catch (const std::exception& e) {
+    g_err = e.what();
+    return nullptr;
+  }
+}
+
 int caffe_net_num_params(caffe_net_t n) {
   return (int)N(n)->learnable_params().size();
 }

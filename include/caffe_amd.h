@@ -110,6 +110,24 @@ int caffe_net_blob_set(caffe_net_t n, const char* name, int diff,
 
 /* learnable params (Net::learnable_params, net.cpp:1350): idx ordering is
  * the arena (backward-completion) order */
+/* pycaffe-shim surface: zero-copy CPU pointers with SyncedMemory head
+ * semantics (mutable access dirties the host copy; next GPU use re-syncs
+ * — the contract python/caffe/_caffe.cpp exposed through Blob.data /
+ * Blob.diff), plus layer enumeration for net.blobs/net.params/net.layers
+ * dict shapes. */
+float* caffe_net_blob_cpu_ptr(caffe_net_t n, const char* name, int diff,
+                              int writable);
+int caffe_net_num_layers(caffe_net_t n);
+int caffe_net_blob_names(caffe_net_t n, char* out, int cap);
+int caffe_net_layer_info(caffe_net_t n, int idx, char* name_out,
+                         int name_cap, char* type_out, int type_cap,
+                         int* num_blobs_out);
+int caffe_net_layer_blob_shape(caffe_net_t n, const char* lname, int bidx,
+                               int* shape_out, int max_dims,
+                               int* ndims_out);
+float* caffe_net_layer_blob_cpu_ptr(caffe_net_t n, const char* lname,
+                                    int bidx, int diff, int writable);
+
 int caffe_net_num_params(caffe_net_t n);
 int caffe_net_param_info(caffe_net_t n, int idx, char* layer_name_out,
                          int name_cap, int* blob_idx_out, long* count_out);
