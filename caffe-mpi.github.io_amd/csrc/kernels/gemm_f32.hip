@@ -27,7 +27,8 @@ constexpr int LDS_S = BM + 1;  // LDS row stride (floats), BM == BN
 // bf16 tensor-core variant (gemm_bf16.hip): same call surface, selected
 // by Engine::gemm_bf16 (mixed precision — fp32 storage, bf16 MFMA)
 void gemm_launch_bf16(hipStream_t s, bool transA, bool transB, dim3 grid,
-                      dim3 block, const GemmArgs& g, bool splitk);
+                      dim3 block, const GemmArgs& g, bool splitk,
+                      bool wide);
 
 // ---- staging: global -> regs (load) and regs -> LDS (write)
 // A tile is held in LDS as As[k][m] (k-major); B as Bs[k][n].
@@ -914,7 +915,16 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // arithmetic — AMP-style per-shape heterogeneity)
   int TMv = 128;
   if (M <= slim_maxm && N > 64) TMv = M <= 32 ? 32 : 64;
-  const int TNv = TMv == 128 ? 128 : (TMv == 32 ? 256 : 128);
+  // bf16 wide tile (256x256): quadruple MACs per staged byte — the bf16
+  // kernel is staging-bound (see gemm_bf16.hip)
+  bool bf16_wide = false;
+  if (Engine::get().gemm_bf16 && TMv == 128 && M > 128 && N > 128) {
+    bf16_wide = true;
+    TMv = 256;
+  }
+  const int TNv = TMv == 256 ? 256
+                 : TMv == 128 ? 128
+                 : (TMv == 32 ? 256 : 128);
   const long tm = (M + TMv - 1) / TMv, tn = (N + TNv - 1) / TNv;
   g.tn = tn;
   g.tiles = tm * tn;
@@ -960,7 +970,7 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     } else if (TMv == 64) {
       launch_slim<64, true>(transA, transB, grid, block, s, g);
     } else if (Engine::get().gemm_bf16) {
-      gemm_launch_bf16(s, transA, transB, grid, block, g, true);
+      gemm_launch_bf16(s, transA, transB, grid, block, g, true, bf16_wide);
     } else if (!transA && !transB)
       hipLaunchKernelGGL((k_gemm_f32<false, false, true>), grid, block, 0,
                          s, g);
@@ -995,7 +1005,7 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
     return;
   }
   if (Engine::get().gemm_bf16) {
-    gemm_launch_bf16(s, transA, transB, grid, block, g, false);
+    gemm_launch_bf16(s, transA, transB, grid, block, g, false, bf16_wide);
     return;
   }
   // glds v2 path (BK=64, 1 block/CU): NN/TN with plain or channel-view
