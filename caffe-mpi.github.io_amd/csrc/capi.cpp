@@ -46,6 +46,20 @@ int caffe_set_mode(int mode, int device) {
   API_CATCH
 }
 
+// multi-node uid bootstrap (csrc/bootstrap.cpp) — exposed so the TCP
+// exchange is testable without GPUs or multiple boxes
+int caffe_uid_serve(const uint8_t* uid, int port, int nclients) {
+  API_TRY
+  return uid_serve(uid, 128, port, nclients);
+  API_CATCH
+}
+int caffe_uid_fetch(uint8_t* out, const char* host, int port,
+                    int timeout_s) {
+  API_TRY
+  return uid_fetch(out, 128, host, port, timeout_s);
+  API_CATCH
+}
+
 // data-stream iteration counter (the LMDB cursor / synthetic stream
 // position; Solver::Step drives it during training — tests reset it)
 int caffe_set_data_iter(uint64_t iter) {

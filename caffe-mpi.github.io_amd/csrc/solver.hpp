@@ -25,6 +25,11 @@ struct Comm {
 };
 
 void rccl_unique_id(void* out_bytes128);
+// multi-node uid bootstrap over TCP (csrc/bootstrap.cpp — the reference's
+// Clusters/MPI_Bcast replacement, clusters.cpp + parallel.cpp:42-45)
+int uid_serve(const void* payload, int len, int port, int nclients);
+int uid_fetch(void* out, int len, const char* host, int port,
+              int timeout_s);
 int rccl_selftest();  // world-1 RCCL linkage/call smoke (GPU mode)
 std::unique_ptr<Comm> make_rccl_comm(int rank, int world,
                                      const void* uid_bytes128);

@@ -155,6 +155,25 @@ static long train_batch_size(const PMsgPtr& sp) {
   return 0;
 }
 
+// multi-node configuration from the environment (the reference's
+// Clusters shim, clusters.cpp: node_rank/node_count + MPI_Bcast of the
+// ncclUniqueId; here the id travels over TCP — csrc/bootstrap.cpp):
+// CAFFE_NNODES, CAFFE_NODE_RANK, MASTER_ADDR, MASTER_PORT.
+struct NodeEnv {
+  int nnodes = 1, node_rank = 0, port = 29500;
+  std::string master = "127.0.0.1";
+};
+static NodeEnv node_env() {
+  NodeEnv e;
+  if (const char* v = getenv("CAFFE_NNODES")) e.nnodes = atoi(v);
+  if (const char* v = getenv("CAFFE_NODE_RANK")) e.node_rank = atoi(v);
+  if (const char* v = getenv("MASTER_ADDR")) e.master = v;
+  if (const char* v = getenv("MASTER_PORT")) e.port = atoi(v);
+  CHECK_GT_(e.nnodes, 0);
+  CHECK_LT_(e.node_rank, e.nnodes);
+  return e;
+}
+
 static int run_train_rank(std::map<std::string, std::string> flags, int dev,
                           int rank, int world,
                           const std::string& rdv_dir) {
@@ -163,42 +182,67 @@ static int run_train_rank(std::map<std::string, std::string> flags, int dev,
     E.set_mode_gpu(dev);
   else
     E.mode = Mode::CPU;
-  E.rank = rank;  // fillers + synthetic data use seed+rank
+  const NodeEnv ne = node_env();
+  // ONE communicator spans every GPU of every node: global rank =
+  // node_rank * local_world + local_rank (parallel.cpp:166-169)
+  const int local_rank = rank, local_world = world;
+  const int grank = ne.node_rank * local_world + local_rank;
+  const int gworld = ne.nnodes * local_world;
+  E.rank = grank;  // fillers + synthetic/LMDB sharding use seed+rank
+  E.world = gworld;
   install_signal_handlers(flags);
   auto sp = parse_prototxt_file(flags["solver"]);
   int batch_override = 0;
-  if (world > 1) {
+  if (local_world > 1) {
+    // the reference divides the prototxt batch among THIS node's solvers
+    // (parallel.cpp:284-348); multi-node multiplies the effective batch
     const long total = train_batch_size(sp);
     if (total > 0)
-      batch_override =
-          (int)(total / world + (rank < total % world ? 1 : 0));
+      batch_override = (int)(total / local_world +
+                             (local_rank < total % local_world ? 1 : 0));
   }
   Solver solver(sp, batch_override);
   solver.set_action_request(&action_request);
-  if (rank != 0) solver.set_snapshot_enabled(false);
-  if (world > 1) {
+  if (grank != 0) solver.set_snapshot_enabled(false);
+  if (gworld > 1) {
     uint8_t uid[128];
-    const std::string uid_path = rdv_dir + "/rccl_uid";
-    if (rank == 0) {
-      rccl_unique_id(uid);
-      const std::string tmp = uid_path + ".tmp";
-      FILE* f = fopen(tmp.c_str(), "wb");
-      CHECK_(f) << "cannot write " << tmp;
-      fwrite(uid, 1, sizeof(uid), f);
-      fclose(f);
-      CHECK_EQ_(rename(tmp.c_str(), uid_path.c_str()), 0);
-    } else {
-      bool got = false;
-      for (int i = 0; i < 6000 && !got; ++i) {  // up to 60 s
-        if (FILE* f = fopen(uid_path.c_str(), "rb")) {
-          got = fread(uid, 1, sizeof(uid), f) == sizeof(uid);
-          fclose(f);
-        }
-        if (!got) usleep(10000);
+    if (ne.nnodes > 1) {
+      // TCP bootstrap: global rank 0 creates + serves the id; every
+      // other rank (any node) fetches from MASTER_ADDR:MASTER_PORT
+      if (grank == 0) {
+        rccl_unique_id(uid);
+        CHECK_EQ_(uid_serve(uid, sizeof(uid), ne.port, gworld - 1), 0)
+            << "uid bootstrap serve failed on port " << ne.port;
+      } else {
+        CHECK_EQ_(uid_fetch(uid, sizeof(uid), ne.master.c_str(), ne.port,
+                            120),
+                  0)
+            << "uid bootstrap fetch from " << ne.master << ":" << ne.port
+            << " failed";
       }
-      CHECK_(got) << "rank " << rank << ": no RCCL uid rendezvous";
+    } else {
+      const std::string uid_path = rdv_dir + "/rccl_uid";
+      if (rank == 0) {
+        rccl_unique_id(uid);
+        const std::string tmp = uid_path + ".tmp";
+        FILE* f = fopen(tmp.c_str(), "wb");
+        CHECK_(f) << "cannot write " << tmp;
+        fwrite(uid, 1, sizeof(uid), f);
+        fclose(f);
+        CHECK_EQ_(rename(tmp.c_str(), uid_path.c_str()), 0);
+      } else {
+        bool got = false;
+        for (int i = 0; i < 6000 && !got; ++i) {  // up to 60 s
+          if (FILE* f = fopen(uid_path.c_str(), "rb")) {
+            got = fread(uid, 1, sizeof(uid), f) == sizeof(uid);
+            fclose(f);
+          }
+          if (!got) usleep(10000);
+        }
+        CHECK_(got) << "rank " << rank << ": no RCCL uid rendezvous";
+      }
     }
-    solver.set_comm(make_rccl_comm(rank, world, uid));
+    solver.set_comm(make_rccl_comm(grank, gworld, uid));
     solver.bcast_weights();  // initial weight bcast, parallel.cpp:208-227
   }
   if (flags.count("snapshot") && !flags["snapshot"].empty())
@@ -223,12 +267,12 @@ static int run_train_rank(std::map<std::string, std::string> flags, int dev,
       fclose(f);
     }
   }
-  if (rank == 0 &&
+  if (E.rank == 0 &&
       solver.param()->boolean("snapshot_after_train", true))
     solver.Snapshot();
   if (solver.early_exit())
     fprintf(stderr, "Optimization stopped early.\n");
-  if (rank == 0) fprintf(stderr, "Optimization Done.\n");
+  if (E.rank == 0) fprintf(stderr, "Optimization Done.\n");
   return 0;
 }
 

@@ -37,6 +37,13 @@ int caffe_set_compute(const char* dtype);
 int caffe_set_rank_world(int rank, int world);
 /* data-stream position (LMDB cursor analog; Solver::Step drives it). */
 int caffe_set_data_iter(uint64_t iter);
+/* multi-node ncclUniqueId bootstrap over TCP (the reference's
+ * Clusters/MPI_Bcast replacement, clusters.cpp + parallel.cpp:42-45):
+ * global rank 0 serves the 128-byte id to world-1 clients; the `caffe`
+ * CLI drives this from CAFFE_NNODES/CAFFE_NODE_RANK/MASTER_ADDR/PORT. */
+int caffe_uid_serve(const uint8_t* uid, int port, int nclients);
+int caffe_uid_fetch(uint8_t* out, const char* host, int port,
+                    int timeout_s);
 /* Caffe::set_random_seed (common.cpp); rank offsets seed like
  * parallel.cpp:179-187 */
 int caffe_set_random_seed(uint64_t seed);
