@@ -199,11 +199,60 @@ __global__ void k_pool_max_fwd(const float* __restrict__ x, int N, int C,
     mask[idx] = bi;
   }
 }
+// 3x3 specialization: the generic kernel's runtime-bounded window loop
+// keeps only ~2 loads in flight per wave (SQ: 76-78% parked on the
+// GoogLeNet pools); a fully-unrolled predicated 3x3 issues all 9 loads
+// before any wait.  Same first-max-wins order (fixed (h,w) ascending).
+__global__ void k_pool_max_fwd3(const float* __restrict__ x, int N, int C,
+                                int H, int W, int ph, int pw, int sh,
+                                int sw, int OH, int OW,
+                                float* __restrict__ y,
+                                int* __restrict__ mask) {
+  const long total = (long)N * C * OH * OW;
+  GRID_STRIDE(idx, total) {
+    const int ow = (int)(idx % OW);
+    const int oh = (int)((idx / OW) % OH);
+    const long nc = idx / ((long)OW * OH);
+    const float* xp = x + nc * H * W;
+    const int hs = oh * sh - ph, ws = ow * sw - pw;
+    float v[9];
+#pragma unroll
+    for (int a = 0; a < 3; ++a)
+#pragma unroll
+      for (int b = 0; b < 3; ++b) {
+        const int h = hs + a, w = ws + b;
+        const bool ok = h >= 0 && h < H && w >= 0 && w < W;
+        v[a * 3 + b] = ok ? xp[h * W + w] : -3.402823466e38f;
+      }
+    float best = -3.402823466e38f;
+    int bi = -1;
+#pragma unroll
+    for (int a = 0; a < 3; ++a)
+#pragma unroll
+      for (int b = 0; b < 3; ++b) {
+        const int h = hs + a, w = ws + b;
+        const bool ok = h >= 0 && h < H && w >= 0 && w < W;
+        if (ok && v[a * 3 + b] > best) {  // strict >: first max wins
+          best = v[a * 3 + b];
+          bi = h * W + w;
+        }
+      }
+    y[idx] = best;
+    mask[idx] = bi;
+  }
+}
+
 void pool_max_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
                   int kh, int kw, int ph, int pw, int sh, int sw, int OH,
                   int OW, float* y, int* mask) {
   const long total = (long)N * C * OH * OW;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 4.0 * total * (kh * kw + 2));
+  if (kh == 3 && kw == 3) {
+    hipLaunchKernelGGL(k_pool_max_fwd3, dim3(nblocks(total, 2)), dim3(TPB),
+                       0, s, x, N, C, H, W, ph, pw, sh, sw, OH, OW, y,
+                       mask);
+    return;
+  }
   hipLaunchKernelGGL(k_pool_max_fwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, x, N, C, H, W, kh, kw, ph, pw, sh, sw, OH, OW, y,
                      mask);
@@ -235,11 +284,59 @@ __global__ void k_pool_max_bwd(const float* __restrict__ dy,
     dx[idx] = acc;
   }
 }
+// 3x3 bwd specialization: for k=3 an input element sits in at most 3x3
+// candidate windows — unroll them with predicates so all 18 mask/dy
+// loads issue before any wait (the generic loop was latency-serialized,
+// 78% parked).  Ascending (a, b) — fixed, deterministic order.
+__global__ void k_pool_max_bwd3(const float* __restrict__ dy,
+                                const int* __restrict__ mask, int N, int C,
+                                int H, int W, int ph, int pw, int sh,
+                                int sw, int OH, int OW,
+                                float* __restrict__ dx) {
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int me = h * W + w;
+    const int ph0 = (h + ph < 3) ? 0 : (h + ph - 3) / sh + 1;
+    const int ph1 = min((h + ph) / sh + 1, OH);
+    const int pw0 = (w + pw < 3) ? 0 : (w + pw - 3) / sw + 1;
+    const int pw1 = min((w + pw) / sw + 1, OW);
+    const float* dyp = dy + nc * OH * OW;
+    const int* mp = mask + nc * OH * OW;
+    bool hit[9];
+    float d[9];
+#pragma unroll
+    for (int da = 0; da < 3; ++da)
+#pragma unroll
+      for (int db = 0; db < 3; ++db) {
+        const int a = ph0 + da, b = pw0 + db;
+        const bool ok = a < ph1 && b < pw1;
+        const int off = ok ? a * OW + b : 0;
+        hit[da * 3 + db] = ok && mp[off] == me;
+        d[da * 3 + db] = dyp[off];
+      }
+    float acc = 0.f;
+#pragma unroll
+    for (int i = 0; i < 9; ++i)
+      if (hit[i]) acc += d[i];  // select, not multiply: a stray NaN in a
+                                // NON-matching window must not leak in
+    dx[idx] = acc;
+  }
+}
+
 void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
                   int C, int H, int W, int kh, int kw, int ph, int pw,
                   int sh, int sw, int OH, int OW, float* dx) {
   const long total = (long)N * C * H * W;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 12.0 * total);
+  if (kh == 3 && kw == 3) {
+    hipLaunchKernelGGL(k_pool_max_bwd3, dim3(nblocks(total, 2)), dim3(TPB),
+                       0, s, dy, mask, N, C, H, W, ph, pw, sh, sw, OH, OW,
+                       dx);
+    return;
+  }
   hipLaunchKernelGGL(k_pool_max_bwd, dim3(nblocks(total, 2)), dim3(TPB), 0,
                      s, dy, mask, N, C, H, W, kh, kw, ph, pw, sh, sw, OH,
                      OW, dx);
