@@ -254,3 +254,36 @@ net_param {{
     s = ca.Solver(text=text)
     s.step(10)
     assert np.isfinite(s.loss())
+
+
+def test_cli_caffe_test_on_lmdb(db, tmp_path):
+    # the `caffe test` subcommand scoring a TEST-phase LMDB net (CPU):
+    # reference tools/caffe.cpp:241 test() — forward test_iter batches,
+    # print averaged scores
+    import subprocess
+    model = tmp_path / "net.prototxt"
+    model.write_text(f"""name: "t"
+layer {{
+  name: "data" type: "Data" top: "data" top: "label"
+  include {{ phase: TEST }}
+  data_param {{ source: "{db}" batch_size: 5 backend: LMDB }}
+  transform_param {{ scale: 0.02 }}
+}}
+layer {{ name: "ip" type: "InnerProduct" bottom: "data" top: "fc"
+  inner_product_param {{ num_output: 10
+    weight_filler {{ type: "gaussian" std: 0.1 }} }} }}
+layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "fc" bottom: "label"
+  top: "loss" }}
+layer {{ name: "acc" type: "Accuracy" bottom: "fc" bottom: "label"
+  top: "accuracy" include {{ phase: TEST }} }}
+""")
+    caffe_bin = os.path.join(REPO, "caffe-mpi.github.io_amd", "caffe")
+    r = subprocess.run([caffe_bin, "test", f"-model={model}",
+                        "-iterations=3"], capture_output=True, text=True,
+                       timeout=300, cwd=REPO)
+    assert r.returncode == 0, r.stdout + r.stderr
+    out = r.stdout + r.stderr
+    assert "loss" in out and "acc" in out, out
+    import re
+    acc = float(re.search(r"acc = ([\d.]+)", out).group(1))
+    assert 0.0 <= acc <= 1.0
