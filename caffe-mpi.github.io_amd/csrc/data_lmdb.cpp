@@ -93,7 +93,10 @@ struct LmdbFeed {
   std::condition_variable cv;
   std::thread worker;
   bool stop = false;
-  uint64_t want = ~0ull;  // iteration the worker should fill next
+  uint64_t want = ~0ull;     // iteration the worker should fill next
+  uint64_t filling = ~0ull;  // iteration the worker is CURRENTLY filling
+                             // (guards get() from racing an in-flight
+                             // fill on the same slot)
 
   // GPU upload staging: PINNED double buffers (parity = iter & 1) so the
   // H2D copies are truly async (no pageable staging), with an
@@ -175,10 +178,12 @@ struct LmdbFeed {
         if (stop) return;
         const uint64_t it = want;
         want = ~0ull;
+        filling = it;
         Slot& s = slots[it & 1];
         lk.unlock();
         fill_slot(s, it);
         lk.lock();
+        filling = ~0ull;
         cv.notify_all();
       }
     });
@@ -189,8 +194,12 @@ struct LmdbFeed {
   Slot& get(uint64_t iter) {
     std::unique_lock<std::mutex> lk(mu);
     Slot& s = slots[iter & 1];
-    // wait out the worker if it is mid-fill on this slot
-    cv.wait(lk, [&] { return want == ~0ull || stop; });
+    // wait out the worker completely (assigned OR mid-fill): without the
+    // `filling` guard an inline fill here could race an in-flight worker
+    // fill of the SAME slot
+    cv.wait(lk, [&] {
+      return (want == ~0ull && filling == ~0ull) || stop;
+    });
     if (!(s.ready && s.iter == iter)) {
       lk.unlock();
       fill_slot(s, iter);
