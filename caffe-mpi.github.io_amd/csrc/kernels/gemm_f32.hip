@@ -914,7 +914,15 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // slim kernel (better utilization than the bf16 kwaves path AND exact
   // arithmetic — AMP-style per-shape heterogeneity)
   int TMv = 128;
-  if (M <= slim_maxm && N > 64) TMv = M <= 32 ? 32 : 64;
+  if (M <= slim_maxm && N > 64) {
+    TMv = M <= 32 ? 32 : 64;
+  } else if (!Engine::get().gemm_bf16 && N > 64 && M > 128 && M <= 448 &&
+             M % 128 >= 1 && M % 128 <= 64) {
+    // remainder rule: a 128-tile grid wastes (128 - M%128) rows of its
+    // last tile row (e.g. GoogLeNet conv2 M=192: 25% idle MFMA rows);
+    // 64-row tiles fit these exactly at the cost of one extra B pass
+    TMv = 64;
+  }
   // bf16 wide tile (256x256): quadruple MACs per staged byte — the bf16
   // kernel is staging-bound (see gemm_bf16.hip)
   bool bf16_wide = false;
