@@ -93,39 +93,11 @@ __device__ __forceinline__ void read16(const float* __restrict__ P, long r,
       return;
     }
     if (v.OW >= 16) {
-      // at most ONE row wrap inside the 16-chunk
+      // at most ONE row wrap inside the 16-chunk: all 16 loads become
+      // independent (the sequential walk would chain their addresses)
       const int jw = v.OW - ow;       // first j on the next output row
       const int w0 = w;               // input col at j=0
       const int wreset = -v.pw + kj;  // input col after the wrap
-      // interior two-segment VECTOR path (s1 only): the chunk is two
-      // contiguous runs — jw floats at (h, w0) and 16-jw at (h+1,
-      // wreset).  Load 16 from each base with unaligned float4s (the
-      // overread stays inside the tensor: every device allocation
-      // carries a 64 B = 16-float tail) and select per lane — the
-      // 16-scalar gather this replaces was the staging hot spot for
-      // OW in [16, 2*16) shapes (every other chunk wraps at OW=28).
-      if (v.sw == 1 && v.sh == 1 && smax >= 16 && h >= 0 && h + 1 < v.H &&
-          w0 >= 0 && w0 + jw - 1 < v.W && wreset >= 0 &&
-          wreset + (15 - jw) < v.W) {
-        using f4u = __attribute__((ext_vector_type(4), aligned(4))) float;
-        const float* p1 = xp + h * v.W + w0;
-        const float* p2 = xp + (h + 1) * v.W + wreset;
-        float s1[16], s2[16];
-#pragma unroll
-        for (int q = 0; q < 4; ++q) {
-          const f4u a = *(const f4u*)(p1 + 4 * q);
-          const f4u b = *(const f4u*)(p2 + 4 * q);
-#pragma unroll
-          for (int e = 0; e < 4; ++e) {
-            s1[4 * q + e] = a[e];
-            s2[4 * q + e] = b[e];
-          }
-        }
-#pragma unroll
-        for (int j = 0; j < 16; ++j)
-          out[j] = j < jw ? s1[j] : s2[(j - jw) & 15];
-        return;
-      }
 #pragma unroll
       for (int j = 0; j < 16; ++j) {
         const bool wrapped = j >= jw;
