@@ -383,10 +383,19 @@ void Solver::TestAll(long iters) {
   // (reference Solver::Test, solver.cpp:439-540; cross-rank SharedScores
   // aggregation is multi-node machinery — single-node here)
   std::map<std::string, double> scores;
+  // the data feeds key on Engine::data_iter: advance it per TEST forward
+  // so successive test iterations see successive batches (with LMDB this
+  // walks the test set in order from record 0 — a fixed eval set; the
+  // training stream is unaffected, Step re-pins data_iter each train
+  // iteration).  Without this every test iteration repeated ONE batch.
+  Engine& E = Engine::get();
+  const uint64_t saved_iter = E.data_iter;
   for (long it = 0; it < iters; ++it) {
+    E.data_iter = (uint64_t)it;
     tn->Forward();
     for (auto& kv : tn->scores()) scores[kv.first] += kv.second;
   }
+  E.data_iter = saved_iter;
   for (auto& kv : scores) {
     const double v = kv.second / iters;
     fprintf(stderr, "[caffe_amd] Test net output: %s = %g\n",

@@ -409,6 +409,9 @@ int main(int argc, char** argv) {
                             : 50;
       std::map<std::string, double> scores;
       for (int i = 0; i < iters; ++i) {
+        // advance the data-stream cursor per iteration (successive test
+        // batches; with LMDB this walks the set from record 0)
+        Engine::get().data_iter = (uint64_t)i;
         net.Forward();
         for (auto& kv : net.scores()) scores[kv.first] += kv.second;
       }
