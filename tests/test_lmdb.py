@@ -336,3 +336,33 @@ layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "fc" bottom: "label"
     # our own 3-batch losses must not all be equal)
     assert len({round(v, 6) for v in losses}) > 1, losses
     assert abs(l3 - l1) > 1e-6, (l1, l3)
+
+
+def test_mean_file_binaryproto(db, tmp_path):
+    # transform_param.mean_file: a BlobProto (binaryproto) full-size mean,
+    # subtracted in SOURCE coordinates — encoded here independently
+    # (field 7 shape{dim}, field 5 packed float data)
+    from make_lmdb import varint
+    rng = np.random.default_rng(77)
+    mean = rng.uniform(0, 255, (C, H, W)).astype(np.float32)
+    shape_msg = b""
+    for d in (C, H, W):
+        shape_msg += b"\x08" + varint(d)
+    blob = b"\x3a" + varint(len(shape_msg)) + shape_msg  # field 7 (shape)
+    data = mean.ravel().tobytes()
+    blob += b"\x2a" + varint(len(data)) + data           # field 5 packed
+    mf = tmp_path / "mean.binaryproto"
+    mf.write_bytes(blob)
+    ca.set_mode("cpu")
+    net = net_from_text(f"""name: "t"
+layer {{
+  name: "data" type: "Data" top: "data" top: "label"
+  data_param {{ source: "{db}" batch_size: 2 backend: LMDB }}
+  transform_param {{ mean_file: "{mf}" scale: 0.5 }}
+}}
+""")
+    net.forward()
+    out = np.asarray(net.blob("data")).reshape(2, C, H, W)
+    for j in range(2):
+        exp = (expected_image(j) - mean) * 0.5
+        assert relerr(out[j], exp) < 1e-5, j

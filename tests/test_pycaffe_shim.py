@@ -108,3 +108,34 @@ snapshot_prefix: "{tmp}/s"
         s2 = caffe.SGDSolver(os.path.join(tmp, "s.prototxt"))
         s2.restore(state)
         assert s2.iter == 6
+
+
+import pytest
+
+
+@pytest.mark.gpu
+def test_pycaffe_gpu_solver():
+    # the classic pycaffe GPU flow end-to-end on the engine
+    caffe.set_device(0)
+    caffe.set_mode_gpu()
+    caffe.set_random_seed(21)
+    with tempfile.TemporaryDirectory() as tmp:
+        netp = write(tmp, "n.prototxt", NET.replace(
+            "force_backward: true\n", ""))
+        s = caffe.SGDSolver(write(tmp, "s.prototxt", f"""net: "{netp}"
+base_lr: 0.05
+lr_policy: "fixed"
+momentum: 0.9
+random_seed: 3
+snapshot_prefix: "{tmp}/s"
+"""))
+        rng = np.random.default_rng(5)
+        s.net.blobs["data"].data[...] = rng.standard_normal((4, 3, 8, 8))
+        s.net.blobs["label"].data[...] = [0, 1, 2, 3]
+        l0 = None
+        for _ in range(8):
+            s.step(1)
+            l = float(s.net.blobs["loss"].data.ravel()[0])
+            l0 = l if l0 is None else l0
+        assert np.isfinite(l) and l < l0
+    caffe.set_mode_cpu()
