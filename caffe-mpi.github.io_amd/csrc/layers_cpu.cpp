@@ -332,6 +332,12 @@ void InnerProductLayer::LayerSetUp(const std::vector<Blob*>& bottom,
   CHECK_(ip);
   Nout_ = (int)ip->inum("num_output");
   bias_ = ip->boolean("bias_term", true);
+  // flatten-from-axis-1, W stored [Nout][K] — any other axis or the
+  // transposed-weight layout must fail loudly, not silently mis-multiply
+  CHECK_EQ_(ip->inum("axis", 1), 1)
+      << "only inner_product_param.axis 1 is implemented";
+  CHECK_(!ip->boolean("transpose", false))
+      << "inner_product_param.transpose is not implemented";
   K_ = bottom[0]->count(1);
   if (blobs_.empty()) {
     blobs_.emplace_back(new Blob({Nout_, (int)K_}));
@@ -862,6 +868,14 @@ void DropoutLayer::Backward_cpu(const std::vector<Blob*>& top,
 // ---------------------------------------------------------------- Concat
 void ConcatLayer::Reshape(const std::vector<Blob*>& bottom,
                           const std::vector<Blob*>& top) {
+  // channel concat only (axis 1, the GoogLeNet/inception form): any other
+  // concat_param.axis must fail loudly, not silently concat channels
+  if (auto cp = param_->sub("concat_param")) {
+    CHECK_EQ_(cp->inum("axis", 1), 1)
+        << "only channel concat (axis 1) is implemented";
+    CHECK_(!cp->has("concat_dim") || cp->inum("concat_dim") == 1)
+        << "only channel concat (concat_dim 1) is implemented";
+  }
   int C = 0;
   for (auto* b : bottom) C += b->channels();
   top[0]->Reshape({bottom[0]->num(), C, bottom[0]->height(),
@@ -949,6 +963,9 @@ void SplitLayer::Backward_gpu(const std::vector<Blob*>& top,
 // ---------------------------------------------------------------- Softmax
 void SoftmaxLayer::Reshape(const std::vector<Blob*>& bottom,
                            const std::vector<Blob*>& top) {
+  if (auto sp = param_->sub("softmax_param"))
+    CHECK_EQ_(sp->inum("axis", 1), 1)
+        << "only softmax over axis 1 (channels) is implemented";
   top[0]->ReshapeLike(*bottom[0]);
   outer_ = bottom[0]->num();
   C_ = bottom[0]->channels();
