@@ -103,3 +103,27 @@ def test_error_names_the_layer():
         raise AssertionError("expected a shape failure")
     except ca.CaffeError as e:
         assert "badconv" in str(e), str(e)
+
+
+def test_unimplemented_options_fail_loudly():
+    # silently mis-computing an unsupported prototxt option is worse than
+    # aborting: axis-2 softmax, transposed IP weights, axis-0 concat
+    import caffe_amd as ca
+    from engine_util import input_net, net_from_text
+    ca.set_mode("cpu")
+    cases = [
+        """layer { name: "s" type: "Softmax" bottom: "in0" top: "out"
+  softmax_param { axis: 2 } }""",
+        """layer { name: "ip" type: "InnerProduct" bottom: "in0"
+  top: "out" inner_product_param { num_output: 3 transpose: true
+    weight_filler { type: "xavier" } } }""",
+        """layer { name: "ip" type: "InnerProduct" bottom: "in0"
+  top: "out" inner_product_param { num_output: 3 axis: 2
+    weight_filler { type: "xavier" } } }""",
+    ]
+    for body in cases:
+        try:
+            net_from_text(input_net([(2, 3, 4, 4)], body))
+            raise AssertionError("expected rejection: " + body[:40])
+        except ca.CaffeError as e:
+            assert "implement" in str(e), str(e)
