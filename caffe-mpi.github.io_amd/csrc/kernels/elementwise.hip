@@ -344,11 +344,56 @@ __global__ void k_pool_max_bwd3(const float* __restrict__ dy,
   }
 }
 
+// 3x3 STRIDE-1 form (the 13 GoogLeNet inception pools): window bounds
+// become max/min — no per-element divisions at all beyond the index
+// decompose (the generic form pays five)
+__global__ void k_pool_max_bwd3s1(const float* __restrict__ dy,
+                                  const int* __restrict__ mask, int N,
+                                  int C, int H, int W, int ph, int pw,
+                                  int OH, int OW, float* __restrict__ dx) {
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int me = h * W + w;
+    const int ph0 = max(h + ph - 2, 0);
+    const int ph1 = min(h + ph + 1, OH);
+    const int pw0 = max(w + pw - 2, 0);
+    const int pw1 = min(w + pw + 1, OW);
+    const float* dyp = dy + nc * OH * OW;
+    const int* mp = mask + nc * OH * OW;
+    bool hit[9];
+    float d[9];
+#pragma unroll
+    for (int da = 0; da < 3; ++da)
+#pragma unroll
+      for (int db = 0; db < 3; ++db) {
+        const int a = ph0 + da, b = pw0 + db;
+        const bool ok = a < ph1 && b < pw1;
+        const int off = ok ? a * OW + b : 0;
+        hit[da * 3 + db] = ok && mp[off] == me;
+        d[da * 3 + db] = dyp[off];
+      }
+    float acc = 0.f;
+#pragma unroll
+    for (int i = 0; i < 9; ++i)
+      if (hit[i]) acc += d[i];
+    dx[idx] = acc;
+  }
+}
+
 void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
                   int C, int H, int W, int kh, int kw, int ph, int pw,
                   int sh, int sw, int OH, int OW, float* dx) {
   const long total = (long)N * C * H * W;
   PerfScope perf(PERF_CLASS("pool"), s, 0, 12.0 * total);
+  if (kh == 3 && kw == 3 && sh == 1 && sw == 1) {
+    hipLaunchKernelGGL(k_pool_max_bwd3s1, dim3(nblocks(total, 2)),
+                       dim3(TPB), 0, s, dy, mask, N, C, H, W, ph, pw, OH,
+                       OW, dx);
+    return;
+  }
   if (kh == 3 && kw == 3) {
     hipLaunchKernelGGL(k_pool_max_bwd3, dim3(nblocks(total, 2)), dim3(TPB),
                        0, s, dy, mask, N, C, H, W, ph, pw, sh, sw, OH, OW,
@@ -930,40 +975,58 @@ void bn_bwd_apply(hipStream_t s, const float* x, const float* dy,
 }
 
 // ------------------------------------------------------------ LRN
-// one thread per (n, s); running cross-channel window (lrn_layer.cu:9-60
-// shape, restated)
+// TWO (n, s) positions per thread, far apart so both streams stay
+// coalesced: the single running cross-channel window was a serial
+// load->accumulate chain (SQ: 66% parked) — interleaving two independent
+// chains doubles the loads in flight (lrn_layer.cu:9-60 restated)
 __global__ void k_lrn_fwd(const float* __restrict__ x, int N, int C, long S,
                           int size, float aos, float beta, float k,
                           float* __restrict__ scale, float* __restrict__ y) {
   const long total = (long)N * S;
+  const long half = (total + 1) / 2;
   const int pre = (size - 1) / 2;
-  GRID_STRIDE(idx, total) {
-    const int n = (int)(idx / S);
-    const long sp = idx - (long)n * S;
-    const float* xp = x + (long)n * C * S + sp;
-    float* scp = scale + (long)n * C * S + sp;
-    float* yp = y + (long)n * C * S + sp;
-    float acc = 0.f;
+  GRID_STRIDE(i, half) {
+    const long idx0 = i;
+    const long idx1 = i + half;
+    const bool two = idx1 < total;
+    const int n0 = (int)(idx0 / S);
+    const int n1 = two ? (int)(idx1 / S) : n0;
+    const long b0 = (long)n0 * C * S + (idx0 - (long)n0 * S);
+    const long b1 = two ? (long)n1 * C * S + (idx1 - (long)n1 * S) : b0;
+    float a0 = 0.f, a1 = 0.f;
     for (int c = 0; c < size - pre && c < C; ++c) {
-      const float v = xp[(long)c * S];
-      acc += v * v;
+      const float v0 = x[b0 + (long)c * S];
+      const float v1 = x[b1 + (long)c * S];
+      a0 += v0 * v0;
+      a1 += v1 * v1;
     }
     for (int c = 0; c < C; ++c) {
       if (c > 0) {
         const int head = c + size - 1 - pre;
         if (head < C) {
-          const float v = xp[(long)head * S];
-          acc += v * v;
+          const float v0 = x[b0 + (long)head * S];
+          const float v1 = x[b1 + (long)head * S];
+          a0 += v0 * v0;
+          a1 += v1 * v1;
         }
         const int tail = c - 1 - pre;
         if (tail >= 0) {
-          const float v = xp[(long)tail * S];
-          acc -= v * v;
+          const float v0 = x[b0 + (long)tail * S];
+          const float v1 = x[b1 + (long)tail * S];
+          a0 -= v0 * v0;
+          a1 -= v1 * v1;
         }
       }
-      const float sc = k + aos * acc;
-      scp[(long)c * S] = sc;
-      yp[(long)c * S] = xp[(long)c * S] * __powf(sc, -beta);
+      const long o0 = b0 + (long)c * S;
+      const long o1 = b1 + (long)c * S;
+      const float s0 = k + aos * a0;
+      const float s1 = k + aos * a1;
+      scale[o0] = s0;
+      y[o0] = x[o0] * __powf(s0, -beta);
+      if (two) {
+        scale[o1] = s1;
+        y[o1] = x[o1] * __powf(s1, -beta);
+      }
     }
   }
 }
@@ -972,8 +1035,9 @@ void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
              float* y) {
   const long S = (long)H * W;
   PerfScope perf(PERF_CLASS("lrn"), s, 0, 12.0 * N * C * S);
-  hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, N,
-                     C, S, size, alpha / size, beta, k, scale, y);
+  hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks((N * S + 1) / 2)), dim3(TPB),
+                     0, s, x, N, C, S, size, alpha / size, beta, k, scale,
+                     y);
 }
 
 __global__ void k_lrn_bwd(const float* __restrict__ x,
