@@ -60,6 +60,25 @@ def test_conv_bf16(cfg):
     assert _relerr(net.param(0, diff=True), dw_ref) < BTOL, "dW"
 
 
+def test_conv_bf16_grouped():
+    # AlexNet-style grouped conv under bf16 compute (group GEMM slices)
+    N, C, H, W, Co, k, p, grp = 2, 8, 14, 14, 16, 5, 2, 2
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w = (rng.standard_normal((Co, C // grp, k, k)) * 0.2).astype(np.float32)
+    body = f"""layer {{ name: "conv" type: "Convolution" bottom: "in0"
+  top: "out" convolution_param {{ num_output: {Co} kernel_size: {k}
+  pad: {p} group: {grp} bias_term: false }} }}"""
+    y_ref = orc.conv_fwd(x, w, None, pad=(p, p), stride=(1, 1), group=grp)
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer("gpu", [(N, C, H, W)], body, [x], params=[w],
+                       top_diff=dy)
+    assert _relerr(y, y_ref) < BTOL
+    dx_ref, dw_ref, _ = orc.conv_bwd(x, w, dy, pad=(p, p), stride=(1, 1),
+                                     group=grp, want_db=False)
+    assert _relerr(net.blob("in0", diff=True), dx_ref) < BTOL
+    assert _relerr(net.param(0, diff=True), dw_ref) < BTOL
+
+
 @pytest.mark.parametrize("shape", [(64, 64, 64), (130, 1000, 2048),
                                    (64, 147, 40000)])  # last: split-K
 def test_ip_bf16(shape):
