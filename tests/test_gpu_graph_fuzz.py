@@ -108,31 +108,39 @@ def gen_graph(rng):
     return head + "\n".join(lines), (n, c, hw)
 
 
+@pytest.mark.parametrize("dtype", ["f32", "bf16"])
 @pytest.mark.parametrize("seed", [5, 19, 42, 77, 101, 137])
-def test_gpu_graph_fusion_parity(seed):
+def test_gpu_graph_fusion_parity(seed, dtype):
+    # tol: exact-class for f32, the bf16 mixed-precision class otherwise
+    tol = 1e-3 if dtype == "f32" else 4e-2
     rng = np.random.default_rng(seed)
     text, (n, c, hw) = gen_graph(rng)
     x = rng.standard_normal((n, c, hw, hw)).astype(np.float32)
     labels = rng.integers(0, 4, n).astype(np.float32)
     results = {}
-    for mode in ("cpu", "gpu"):
-        ca.set_mode(mode)
-        ca.set_random_seed(1000 + seed)
-        net = net_from_text(text)
-        net.set_blob("in0", x)
-        net.set_blob("in1", labels)
-        net.forward()
-        net.backward()
-        results[mode] = (
-            float(np.asarray(net.blob("loss")).ravel()[0]),
-            np.asarray(net.blob("in0", diff=True)).copy(),
-            [np.asarray(net.param(i, diff=True)).copy()
-             for i in range(net.num_params())])
+    try:
+        for mode in ("cpu", "gpu"):
+            ca.set_mode(mode)
+            if mode == "gpu":
+                ca.set_compute(dtype)
+            ca.set_random_seed(1000 + seed)
+            net = net_from_text(text)
+            net.set_blob("in0", x)
+            net.set_blob("in1", labels)
+            net.forward()
+            net.backward()
+            results[mode] = (
+                float(np.asarray(net.blob("loss")).ravel()[0]),
+                np.asarray(net.blob("in0", diff=True)).copy(),
+                [np.asarray(net.param(i, diff=True)).copy()
+                 for i in range(net.num_params())])
+    finally:
+        ca.set_compute("f32")
     lc, dc, pc = results["cpu"]
     lg, dg, pg = results["gpu"]
-    assert abs(lg - lc) < 1e-3 * max(1.0, abs(lc)), (lc, lg)
+    assert abs(lg - lc) < tol * max(1.0, abs(lc)), (lc, lg)
     scale = max(np.abs(dc).max(), 1e-3)
-    assert np.abs(dg - dc).max() < 1e-3 * scale
+    assert np.abs(dg - dc).max() < tol * scale
     for i, (a, b) in enumerate(zip(pc, pg)):
         s = max(np.abs(a).max(), 1e-3)
-        assert np.abs(b - a).max() < 1e-3 * s, f"param {i}"
+        assert np.abs(b - a).max() < tol * s, f"param {i}"
