@@ -366,3 +366,40 @@ layer {{
     for j in range(2):
         exp = (expected_image(j) - mean) * 0.5
         assert relerr(out[j], exp) < 1e-5, j
+
+
+def test_iter_size_equivalence_on_lmdb(db, tmp_path):
+    # iter_size 2 @ batch 4 == batch 8: the sub-passes advance the data
+    # cursor (data_iter = iter*iter_size + sub), so both runs consume the
+    # SAME record set per update — gradients and params must match
+    def train(iter_size, batch, tag):
+        ca.set_mode("cpu")
+        ca.set_rank_world(0, 1)
+        ca.set_data_iter(0)
+        text = f"""base_lr: 0.05
+lr_policy: "fixed"
+momentum: 0.9
+random_seed: 6
+iter_size: {iter_size}
+snapshot_prefix: "{tmp_path}/{tag}"
+net_param {{
+  name: "n"
+  layer {{ name: "data" type: "Data" top: "data" top: "label"
+    data_param {{ source: "{db}" batch_size: {batch} backend: LMDB }}
+    transform_param {{ scale: 0.0078125 mean_value: 128 }} }}
+  layer {{ name: "ip" type: "InnerProduct" bottom: "data" top: "fc"
+    inner_product_param {{ num_output: 10
+      weight_filler {{ type: "gaussian" std: 0.05 }} }} }}
+  layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "fc"
+    bottom: "label" top: "loss" }}
+}}
+"""
+        s = ca.Solver(text=text)
+        s.step(4)
+        return [np.asarray(s.net.param(i)).copy()
+                for i in range(s.net.num_params())]
+
+    acc = train(2, 4, "a")
+    ref = train(1, 8, "b")
+    for a, b in zip(acc, ref):
+        assert relerr(a, b) < 1e-5
