@@ -147,3 +147,32 @@ def test_imagedata_gpu(dataset):
         net.forward()
         outs[mode] = np.asarray(net.blob("data")).copy()
     assert relerr(outs["gpu"], outs["cpu"]) < 1e-6
+
+
+def test_imagedata_solver_training(dataset, tmp_path):
+    # ImageData through the real solver loop (prefetch worker + shuffle)
+    d, imgs = dataset
+    ca.set_mode("cpu")
+    text = f"""base_lr: 0.05
+lr_policy: "fixed"
+momentum: 0.9
+random_seed: 2
+snapshot_prefix: "{tmp_path}/s"
+net_param {{
+  name: "n"
+  layer {{ name: "data" type: "ImageData" top: "data" top: "label"
+    image_data_param {{ source: "{d}/list.txt" root_folder: "{d}/"
+      batch_size: 6 shuffle: true }}
+    transform_param {{ scale: 0.0078125 mean_value: 128 crop_size: 8
+      mirror: true }} }}
+  layer {{ name: "ip" type: "InnerProduct" bottom: "data" top: "fc"
+    inner_product_param {{ num_output: 4
+      weight_filler {{ type: "gaussian" std: 0.05 }} }} }}
+  layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "fc"
+    bottom: "label" top: "loss" }}
+}}
+"""
+    s = ca.Solver(text=text)
+    s.step(12)  # several epochs over 9 images
+    import numpy as np
+    assert np.isfinite(s.loss())
