@@ -940,10 +940,13 @@ void gemm(hipStream_t s, bool transA, bool transB, long M, long N, long K,
   // plain-C calls only (no epilogue) — fused-epilogue GEMMs have huge N
   int SK = 1;
   // split-K block target: enough tiles*SK to oversubscribe the 256 CUs
-  // ~2x at 2 waves/SIMD occupancy (A/B-able via CAFFE_SK_TARGET)
+  // (A/B-able via CAFFE_SK_TARGET).  Round 2: 2048 re-measured BEST after
+  // the slim tiles landed (RN50 1996 -> 2013, GoogLeNet 3223 -> 3248;
+  // 4096 regresses) — the slim wgrad grids have more, smaller tiles, so
+  // deeper K-slicing now pays where it was neutral in round 1.
   static const long sk_target = [] {
     const char* e = getenv("CAFFE_SK_TARGET");
-    return e ? atol(e) : 1024L;
+    return e ? atol(e) : 2048L;
   }();
   if (!epi && beta == 0.f && g.tiles < sk_target && K > 4 * BK) {
     SK = (int)std::min<long>(
