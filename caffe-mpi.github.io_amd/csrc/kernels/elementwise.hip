@@ -344,6 +344,47 @@ __global__ void k_pool_max_bwd3(const float* __restrict__ dy,
   }
 }
 
+// 3x3 STRIDE-2 form (the ResNet/GoogLeNet/AlexNet downsampling pools):
+// at stride 2 an input pixel overlaps at most 2x2 output windows, so the
+// gather loop is 2x2 — 4 dy + 4 mask loads instead of the generic form's
+// 9+9 — and the stride divisions become shifts
+__global__ void k_pool_max_bwd3s2(const float* __restrict__ dy,
+                                  const int* __restrict__ mask, int N,
+                                  int C, int H, int W, int ph, int pw,
+                                  int OH, int OW, float* __restrict__ dx) {
+  const long total = (long)N * C * H * W;
+  GRID_STRIDE(idx, total) {
+    const int w = (int)(idx % W);
+    const int h = (int)((idx / W) % H);
+    const long nc = idx / ((long)W * H);
+    const int me = h * W + w;
+    const int th = h + ph, tw = w + pw;  // >= 0
+    const int ph0 = (th < 3) ? 0 : ((th - 3) >> 1) + 1;
+    const int ph1 = min((th >> 1) + 1, OH);
+    const int pw0 = (tw < 3) ? 0 : ((tw - 3) >> 1) + 1;
+    const int pw1 = min((tw >> 1) + 1, OW);
+    const float* dyp = dy + nc * OH * OW;
+    const int* mp = mask + nc * OH * OW;
+    bool hit[4];
+    float d[4];
+#pragma unroll
+    for (int da = 0; da < 2; ++da)
+#pragma unroll
+      for (int db = 0; db < 2; ++db) {
+        const int a = ph0 + da, b = pw0 + db;
+        const bool ok = a < ph1 && b < pw1;
+        const int off = ok ? a * OW + b : 0;
+        hit[da * 2 + db] = ok && mp[off] == me;
+        d[da * 2 + db] = dyp[off];
+      }
+    float acc = 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      if (hit[i]) acc += d[i];  // select, not multiply (NaN containment)
+    dx[idx] = acc;
+  }
+}
+
 // 3x3 STRIDE-1 form (the 13 GoogLeNet inception pools): window bounds
 // become max/min — no per-element divisions at all beyond the index
 // decompose (the generic form pays five)
@@ -390,6 +431,12 @@ void pool_max_bwd(hipStream_t s, const float* dy, const int* mask, int N,
   PerfScope perf(PERF_CLASS("pool"), s, 0, 12.0 * total);
   if (kh == 3 && kw == 3 && sh == 1 && sw == 1) {
     hipLaunchKernelGGL(k_pool_max_bwd3s1, dim3(nblocks(total, 2)),
+                       dim3(TPB), 0, s, dy, mask, N, C, H, W, ph, pw, OH,
+                       OW, dx);
+    return;
+  }
+  if (kh == 3 && kw == 3 && sh == 2 && sw == 2) {
+    hipLaunchKernelGGL(k_pool_max_bwd3s2, dim3(nblocks(total, 2)),
                        dim3(TPB), 0, s, dy, mask, N, C, H, W, ph, pw, OH,
                        OW, dx);
     return;

@@ -97,7 +97,8 @@ def check_pool(mode, pool="MAX", N=2, C=4, H=13, W=13, k=3, s=2, p=0):
         _ = mask
 
 
-def check_pool_bwd(mode, pool="MAX", N=2, C=4, H=13, W=13, k=3, s=2):
+def check_pool_bwd(mode, pool="MAX", N=2, C=4, H=13, W=13, k=3, s=2,
+                   p=0):
     # pool preceded by a 1x1 conv so prop_down[0] is true
     x = rng.standard_normal((N, C, H, W)).astype(np.float32)
     w = np.eye(C, dtype=np.float32).reshape(C, C, 1, 1).copy()
@@ -113,12 +114,12 @@ layer {{
   type: "Pooling"
   bottom: "mid"
   top: "out"
-  pooling_param {{ pool: {pool} kernel_size: {k} stride: {s} }}
+  pooling_param {{ pool: {pool} kernel_size: {k} stride: {s} pad: {p} }}
 }}"""
     if pool == "MAX":
-        y_ref, mask = orc.pool_max_fwd(x, k, k, 0, 0, s, s)
+        y_ref, mask = orc.pool_max_fwd(x, k, k, p, p, s, s)
     else:
-        y_ref = orc.pool_ave_fwd(x, k, k, 0, 0, s, s)
+        y_ref = orc.pool_ave_fwd(x, k, k, p, p, s, s)
     dy = rng.standard_normal(y_ref.shape).astype(np.float32)
     net, y = run_layer(mode, [(N, C, H, W)], body, [x], params=[w],
                        top_diff=dy)
@@ -126,7 +127,7 @@ layer {{
     if pool == "MAX":
         dx_ref = orc.pool_max_bwd(dy, mask, H, W)
     else:
-        dx_ref = orc.pool_ave_bwd(dy, H, W, k, k, 0, 0, s, s)
+        dx_ref = orc.pool_ave_bwd(dy, H, W, k, k, p, p, s, s)
     # dx lands in "mid"'s diff == conv's top diff; identity conv passes it on
     assert relerr(net.blob("mid", diff=True), dx_ref) < TOL
 
@@ -451,6 +452,9 @@ ALL_CHECKS = {
     "pool_max": lambda m: check_pool(m, "MAX"),
     "pool_ave": lambda m: check_pool(m, "AVE"),
     "pool_max_bwd": lambda m: check_pool_bwd(m, "MAX"),
+    # 3x3 s2 pad1 (ResNet pool1 shape class -> k_pool_max_bwd3s2 padded)
+    "pool_max_bwd_p1": lambda m: check_pool_bwd(m, "MAX", H=14, W=14,
+                                                p=1),
     "pool_ave_bwd": lambda m: check_pool_bwd(m, "AVE"),
     "bn": check_bn,
     "bn_large_s": lambda m: check_bn(m, N=3, C=4, H=17, W=17),
