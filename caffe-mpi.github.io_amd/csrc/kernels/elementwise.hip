@@ -1080,11 +1080,86 @@ __global__ void k_lrn_fwd(const float* __restrict__ x, int N, int C, long S,
     }
   }
 }
+// size-SPECIALIZED forward (local_size is 5 in every model-zoo LRN): a
+// compile-time register ring of the last SZ window terms makes the
+// running-window update ONE global load per channel — the tail term and
+// the center value both come out of the ring instead of being re-read
+// from HBM (3 loads/channel -> 1).  Same two-position interleave.
+template <int SZ>
+__global__ void k_lrn_fwd_t(const float* __restrict__ x, int N, int C,
+                            long S, float aos, float beta, float k,
+                            float* __restrict__ scale,
+                            float* __restrict__ y) {
+  constexpr int pre = (SZ - 1) / 2;       // window [c-pre, c+SZ-1-pre]
+  const long total = (long)N * S;
+  const long half = (total + 1) / 2;
+  GRID_STRIDE(i, half) {
+    const long idx0 = i;
+    const long idx1 = i + half;
+    const bool two = idx1 < total;
+    const int n0 = (int)(idx0 / S);
+    const int n1 = two ? (int)(idx1 / S) : n0;
+    const long b0 = (long)n0 * C * S + (idx0 - (long)n0 * S);
+    const long b1 = two ? (long)n1 * C * S + (idx1 - (long)n1 * S) : b0;
+    // ring[j] at top of iter c holds channel c-1-pre+j's raw value /
+    // square = window(c-1); out-of-range channels are zero
+    float q0[SZ], q1[SZ], v0[SZ], v1[SZ];
+    float a0 = 0.f, a1 = 0.f;
+#pragma unroll
+    for (int j = 0; j < SZ; ++j) q0[j] = q1[j] = v0[j] = v1[j] = 0.f;
+#pragma unroll
+    for (int j = 0; j + pre + 1 < SZ; ++j)  // channels 0..SZ-2-pre
+      if (j < C) {
+        const float u0 = x[b0 + (long)j * S];
+        const float u1 = x[b1 + (long)j * S];
+        v0[j + pre + 1] = u0;
+        q0[j + pre + 1] = u0 * u0;
+        a0 += u0 * u0;
+        v1[j + pre + 1] = u1;
+        q1[j + pre + 1] = u1 * u1;
+        a1 += u1 * u1;
+      }
+    for (int c = 0; c < C; ++c) {
+      const int head = c + SZ - 1 - pre;
+      float h0 = 0.f, h1 = 0.f;
+      if (head < C) {
+        h0 = x[b0 + (long)head * S];
+        h1 = x[b1 + (long)head * S];
+      }
+      a0 += h0 * h0 - q0[0];
+      a1 += h1 * h1 - q1[0];
+#pragma unroll
+      for (int j = 0; j + 1 < SZ; ++j) {
+        q0[j] = q0[j + 1]; q1[j] = q1[j + 1];
+        v0[j] = v0[j + 1]; v1[j] = v1[j + 1];
+      }
+      q0[SZ - 1] = h0 * h0; q1[SZ - 1] = h1 * h1;
+      v0[SZ - 1] = h0;      v1[SZ - 1] = h1;
+      const long o0 = b0 + (long)c * S;
+      const float s0 = k + aos * a0;
+      scale[o0] = s0;
+      y[o0] = v0[pre] * __powf(s0, -beta);  // ring slot pre == channel c
+      if (two) {
+        const long o1 = b1 + (long)c * S;
+        const float s1 = k + aos * a1;
+        scale[o1] = s1;
+        y[o1] = v1[pre] * __powf(s1, -beta);
+      }
+    }
+  }
+}
+
 void lrn_fwd(hipStream_t s, const float* x, int N, int C, int H, int W,
              int size, float alpha, float beta, float k, float* scale,
              float* y) {
   const long S = (long)H * W;
   PerfScope perf(PERF_CLASS("lrn"), s, 0, 12.0 * N * C * S);
+  if (size == 5) {
+    hipLaunchKernelGGL(k_lrn_fwd_t<5>, dim3(nblocks((N * S + 1) / 2)),
+                       dim3(TPB), 0, s, x, N, C, S, alpha / size, beta, k,
+                       scale, y);
+    return;
+  }
   hipLaunchKernelGGL(k_lrn_fwd, dim3(nblocks((N * S + 1) / 2)), dim3(TPB),
                      0, s, x, N, C, S, size, alpha / size, beta, k, scale,
                      y);
@@ -1126,11 +1201,94 @@ __global__ void k_lrn_bwd(const float* __restrict__ x,
     }
   }
 }
+// size-SPECIALIZED backward: register rings of the last SZ-1 window
+// ratios (dy*y/scale) AND of the diagonal term dy*scale^-beta — each
+// channel's (dy, y, scale) triple is loaded ONCE, as the window head,
+// instead of three times (head ratio, tail ratio, diagonal): 9 loads +
+// 2 divisions per channel -> 4 loads + 1 division.  Two (n, s) positions
+// per thread as in the forward (the single running chain serializes).
+template <int SZ>
+__global__ void k_lrn_bwd_t(const float* __restrict__ x,
+                            const float* __restrict__ y,
+                            const float* __restrict__ dy,
+                            const float* __restrict__ scale, int N, int C,
+                            long S, float cr, float beta,
+                            float* __restrict__ dx) {
+  constexpr int pre = (SZ - 1) / 2;
+  constexpr int ctr = SZ - 1 - pre;  // center channel's ring slot (-lo0)
+  const long total = (long)N * S;
+  const long half = (total + 1) / 2;
+  GRID_STRIDE(i, half) {
+    const long idx0 = i;
+    const long idx1 = i + half;
+    const bool two = idx1 < total;
+    const int n0 = (int)(idx0 / S);
+    const int n1 = two ? (int)(idx1 / S) : n0;
+    const long b0 = (long)n0 * C * S + (idx0 - (long)n0 * S);
+    const long b1 = two ? (long)n1 * C * S + (idx1 - (long)n1 * S) : b0;
+    // rings hold channels [c-ctr, c+pre-1] at top of iter c
+    float rr0[SZ - 1], rr1[SZ - 1], tt0[SZ - 1], tt1[SZ - 1];
+    float a0 = 0.f, a1 = 0.f;
+#pragma unroll
+    for (int j = 0; j < SZ - 1; ++j) rr0[j] = rr1[j] = tt0[j] = tt1[j] = 0.f;
+#pragma unroll
+    for (int ch = 0; ch < pre; ++ch)    // prologue: channels 0..pre-1
+      if (ch < C) {
+        const long i0 = b0 + (long)ch * S, i1 = b1 + (long)ch * S;
+        const float d0 = dy[i0], d1 = dy[i1];
+        const float s0 = scale[i0], s1 = scale[i1];
+        const float r0 = d0 * y[i0] / s0, r1 = d1 * y[i1] / s1;
+        rr0[ch + ctr] = r0;
+        tt0[ch + ctr] = d0 * __powf(s0, -beta);
+        a0 += r0;
+        rr1[ch + ctr] = r1;
+        tt1[ch + ctr] = d1 * __powf(s1, -beta);
+        a1 += r1;
+      }
+    for (int c = 0; c < C; ++c) {
+      const int head = c + pre;
+      float rh0 = 0.f, rh1 = 0.f, th0 = 0.f, th1 = 0.f;
+      if (head < C) {
+        const long i0 = b0 + (long)head * S, i1 = b1 + (long)head * S;
+        const float d0 = dy[i0], d1 = dy[i1];
+        const float s0 = scale[i0], s1 = scale[i1];
+        rh0 = d0 * y[i0] / s0;
+        th0 = d0 * __powf(s0, -beta);
+        rh1 = d1 * y[i1] / s1;
+        th1 = d1 * __powf(s1, -beta);
+      }
+      a0 += rh0;
+      a1 += rh1;
+      const long o0 = b0 + (long)c * S;
+      dx[o0] = tt0[ctr] - cr * x[o0] * a0;
+      if (two) {
+        const long o1 = b1 + (long)c * S;
+        dx[o1] = tt1[ctr] - cr * x[o1] * a1;
+      }
+      a0 -= rr0[0];
+      a1 -= rr1[0];
+#pragma unroll
+      for (int j = 0; j + 1 < SZ - 1; ++j) {
+        rr0[j] = rr0[j + 1]; rr1[j] = rr1[j + 1];
+        tt0[j] = tt0[j + 1]; tt1[j] = tt1[j + 1];
+      }
+      rr0[SZ - 2] = rh0; rr1[SZ - 2] = rh1;
+      tt0[SZ - 2] = th0; tt1[SZ - 2] = th1;
+    }
+  }
+}
+
 void lrn_bwd(hipStream_t s, const float* x, const float* y, const float* dy,
              const float* scale, int N, int C, int H, int W, int size,
              float alpha, float beta, float* dx) {
   const long S = (long)H * W;
   PerfScope perf(PERF_CLASS("lrn"), s, 0, 20.0 * N * C * S);
+  if (size == 5) {
+    hipLaunchKernelGGL(k_lrn_bwd_t<5>, dim3(nblocks((N * S + 1) / 2)),
+                       dim3(TPB), 0, s, x, y, dy, scale, N, C, S,
+                       2.f * alpha * beta / size, beta, dx);
+    return;
+  }
   hipLaunchKernelGGL(k_lrn_bwd, dim3(nblocks(N * S)), dim3(TPB), 0, s, x, y,
                      dy, scale, N, C, S, size, 2.f * alpha * beta / size,
                      beta, dx);

@@ -460,6 +460,10 @@ ALL_CHECKS = {
     "bn_large_s": lambda m: check_bn(m, N=3, C=4, H=17, W=17),
     "relu": check_relu,
     "lrn": check_lrn,
+    # C smaller than the window (prologue/head guards of the size-5
+    # register-ring kernels) and a non-5 size (generic fallback)
+    "lrn_smallc": lambda m: check_lrn(m, C=3),
+    "lrn_size3": lambda m: check_lrn(m, size=3),
     "softmaxloss": check_softmaxloss,
     "eltwise_concat": check_eltwise_concat,
     "dropout": check_dropout,
