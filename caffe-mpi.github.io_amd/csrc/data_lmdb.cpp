@@ -55,6 +55,9 @@ static Datum decode_datum(const uint8_t* p, size_t n) {
     }
   }
   CHECK_GT_(d.c, 0);
+  CHECK_(d.data != nullptr)
+      << "Datum record has no uint8 `data` field (float_data Datums are "
+         "not supported by this feed)";
   CHECK_EQ_((long)d.len, (long)d.c * d.h * d.w)
       << "Datum data size mismatch";
   return d;
