@@ -176,3 +176,36 @@ net_param {{
     s.step(12)  # several epochs over 9 images
     import numpy as np
     assert np.isfinite(s.loss())
+
+
+def test_header_comments(tmp_path):
+    # netpbm allows '#' comments anywhere in the header (load_pnm skips)
+    a = np.random.default_rng(7).integers(0, 256, (3, 4, 5), dtype=np.uint8)
+    hwc = np.transpose(a, (1, 2, 0)).reshape(-1)
+    with open(tmp_path / "c.ppm", "wb") as f:
+        f.write(b"P6\n# a comment\n5 # trailing\n4\n# another\n255\n")
+        f.write(hwc.tobytes())
+    (tmp_path / "list.txt").write_text("c.ppm 0\n")
+    ca.set_mode("cpu")
+    net = net_from_text(img_net(str(tmp_path), batch=1))
+    net.forward()
+    data = np.asarray(net.blob("data")).reshape(3, 4, 5)
+    assert relerr(data, a.astype(np.float32)) < 1e-6
+
+
+@pytest.mark.parametrize("case", ["ascii_magic", "deep_maxval", "truncated"])
+def test_bad_image_errors(tmp_path, case):
+    # decoder failures must be LOUD with the offending path in the message
+    pth = tmp_path / "bad.ppm"
+    if case == "ascii_magic":
+        pth.write_bytes(b"P3\n2 2\n255\n0 0 0 0 0 0 0 0 0 0 0 0\n")
+    elif case == "deep_maxval":
+        pth.write_bytes(b"P6\n2 2\n65535\n" + bytes(24))
+    else:
+        pth.write_bytes(b"P6\n4 4\n255\n" + bytes(5))  # 48 expected
+    (tmp_path / "list.txt").write_text("bad.ppm 0\n")
+    ca.set_mode("cpu")
+    with pytest.raises(Exception) as e:
+        net = net_from_text(img_net(str(tmp_path), batch=1))
+        net.forward()
+    assert "bad.ppm" in str(e.value)

@@ -403,3 +403,26 @@ net_param {{
     ref = train(1, 8, "b")
     for a, b in zip(acc, ref):
         assert relerr(a, b) < 1e-5
+
+
+def test_corrupt_lmdb_errors(tmp_path):
+    # reader failures at setup must be loud and name the problem:
+    # (a) wrong magic, (b) file shorter than the two meta pages
+    bad = tmp_path / "bad_lmdb"
+    bad.mkdir()
+    (bad / "data.mdb").write_bytes(b"\x00" * 8192)
+    net_text = f"""name: "t"
+layer {{ name: "data" type: "Data" top: "data" top: "label"
+  data_param {{ source: "{bad}" batch_size: 2 }} }}
+"""
+    ca.set_mode("cpu")
+    with pytest.raises(Exception) as e:
+        net_from_text(net_text)
+    assert "not an LMDB file" in str(e.value)
+
+    short = tmp_path / "short_lmdb"
+    short.mkdir()
+    (short / "data.mdb").write_bytes(b"\x01\x02")
+    net_text2 = net_text.replace(str(bad), str(short))
+    with pytest.raises(Exception):
+        net_from_text(net_text2)
