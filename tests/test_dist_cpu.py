@@ -106,3 +106,16 @@ def test_two_ranks_sharded_match_combined_batch():
     p2 = parse_params(double[0])
     p1 = parse_params(single[0])
     assert np.allclose(p1, p2, rtol=1e-4, atol=1e-5), np.abs(p1 - p2).max()
+
+
+def test_four_ranks_sharded_match_combined_batch():
+    # the driver's N=4/8 shape: more ranks, smaller per-rank batch — the
+    # bucketed all-reduce averages over 4 members and must still equal
+    # the combined-batch gradient
+    quad = run_dist(4, ["--iters", "3", "--rank-data", "shard",
+                        "--batch", "4"])
+    single = run_dist(1, ["--iters", "3", "--rank-data", "combined",
+                          "--batch", "16"])
+    p4 = parse_params(quad[0])
+    p1 = parse_params(single[0])
+    assert np.allclose(p1, p4, rtol=1e-4, atol=1e-5), np.abs(p1 - p4).max()
