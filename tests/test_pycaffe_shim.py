@@ -139,3 +139,30 @@ snapshot_prefix: "{tmp}/s"
             l0 = l if l0 is None else l0
         assert np.isfinite(l) and l < l0
     caffe.set_mode_cpu()
+
+
+def test_sgdsolver_snapshot_restore_roundtrip():
+    # solver.snapshot() / solver.restore(state): params + iter survive the
+    # round trip through a fresh solver (pycaffe SGDSolver surface)
+    caffe.set_mode_cpu()
+    with tempfile.TemporaryDirectory() as tmp:
+        sol = f"""base_lr: 0.05
+lr_policy: "fixed"
+momentum: 0.9
+random_seed: 11
+snapshot_prefix: "{tmp}/s"
+net_param {{ {NET[NET.index('name'):]} }}
+"""
+        s = caffe.SGDSolver(write(tmp, "s.prototxt", sol))
+        s.step(3)
+        assert s.iter == 3
+        w = s.net.params["conv1"][0].data.copy()
+        s.snapshot()
+        state = os.path.join(tmp, "s_iter_3.solverstate")
+        assert os.path.exists(state)
+        s2 = caffe.SGDSolver(write(tmp, "s2.prototxt", sol))
+        s2.restore(state)
+        assert s2.iter == 3
+        assert np.array_equal(s2.net.params["conv1"][0].data, w)
+        s2.step(1)  # training continues from the restored state
+        assert s2.iter == 4
