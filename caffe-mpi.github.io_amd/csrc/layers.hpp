@@ -19,8 +19,29 @@ class Layer {
   virtual ~Layer() = default;
 
   void SetUp(const std::vector<Blob*>& bottom, const std::vector<Blob*>& top) {
+    check_blob_counts(bottom, top);
     LayerSetUp(bottom, top);
     Reshape(bottom, top);
+  }
+  // blob-count contract (reference layer.hpp CheckBlobCounts /
+  // ExactNumBottomBlobs analog): enforced before any blob is touched, so
+  // a miswired prototxt (e.g. a typo'd `bottom:` key leaving a layer with
+  // no input) fails with a message naming the layer instead of an
+  // out-of-range access
+  void check_blob_counts(const std::vector<Blob*>& bottom,
+                         const std::vector<Blob*>& top) const {
+    CHECK_GE_((int)bottom.size(), min_bottom_blobs())
+        << type() << " layer '" << name() << "': needs at least "
+        << min_bottom_blobs() << " bottom blob(s), got " << bottom.size();
+    CHECK_LE_((int)bottom.size(), max_bottom_blobs())
+        << type() << " layer '" << name() << "': takes at most "
+        << max_bottom_blobs() << " bottom blob(s), got " << bottom.size();
+    CHECK_GE_((int)top.size(), min_top_blobs())
+        << type() << " layer '" << name() << "': needs at least "
+        << min_top_blobs() << " top blob(s), got " << top.size();
+    CHECK_LE_((int)top.size(), max_top_blobs())
+        << type() << " layer '" << name() << "': takes at most "
+        << max_top_blobs() << " top blob(s), got " << top.size();
   }
   virtual void LayerSetUp(const std::vector<Blob*>&,
                           const std::vector<Blob*>&) {}
@@ -66,6 +87,9 @@ class Layer {
 
   virtual bool auto_top_blobs() const { return false; }
   virtual int min_top_blobs() const { return 1; }
+  virtual int max_top_blobs() const { return 1; }
+  virtual int min_bottom_blobs() const { return 1; }
+  virtual int max_bottom_blobs() const { return 1; }
 
   // loss weight per top (loss layers default 1 for top 0)
   float loss(int i) const { return i < (int)loss_.size() ? loss_[i] : 0.f; }
@@ -141,6 +165,9 @@ class DataLayer : public Layer {  // synthetic or LMDB source (§8a a12,
   void Backward_gpu(const std::vector<Blob*>&, const std::vector<bool>&,
                     const std::vector<Blob*>&) override {}
   int min_top_blobs() const override { return 2; }
+  int max_top_blobs() const override { return 2; }
+  int min_bottom_blobs() const override { return 0; }
+  int max_bottom_blobs() const override { return 0; }
 
   int batch_ = 0, C_ = 3, H_ = 224, W_ = 224;
   uint64_t iter_ = 0;
@@ -346,6 +373,8 @@ class ReLULayer : public Layer {
 class EltwiseLayer : public Layer {
  public:
   using Layer::Layer;
+  int min_bottom_blobs() const override { return 2; }
+  int max_bottom_blobs() const override { return 4096; }
   void LayerSetUp(const std::vector<Blob*>&,
                   const std::vector<Blob*>&) override;
   void Reshape(const std::vector<Blob*>& b,
@@ -419,6 +448,7 @@ class DropoutLayer : public Layer {
 class ConcatLayer : public Layer {
  public:
   using Layer::Layer;
+  int max_bottom_blobs() const override { return 4096; }
   void Reshape(const std::vector<Blob*>&, const std::vector<Blob*>&) override;
   void Forward_cpu(const std::vector<Blob*>&,
                    const std::vector<Blob*>&) override;
@@ -433,6 +463,7 @@ class ConcatLayer : public Layer {
 class SplitLayer : public Layer {
  public:
   using Layer::Layer;
+  int max_top_blobs() const override { return 4096; }
   void Reshape(const std::vector<Blob*>&, const std::vector<Blob*>&) override;
   void Forward_cpu(const std::vector<Blob*>&,
                    const std::vector<Blob*>&) override;
@@ -467,6 +498,8 @@ class SoftmaxLayer : public Layer {
 class SoftmaxWithLossLayer : public Layer {
  public:
   using Layer::Layer;
+  int min_bottom_blobs() const override { return 2; }
+  int max_bottom_blobs() const override { return 2; }
   void LayerSetUp(const std::vector<Blob*>&,
                   const std::vector<Blob*>&) override;
   void Reshape(const std::vector<Blob*>&, const std::vector<Blob*>&) override;
@@ -486,6 +519,8 @@ class SoftmaxWithLossLayer : public Layer {
 class AccuracyLayer : public Layer {  // CPU-resident (reference: CPU only)
  public:
   using Layer::Layer;
+  int min_bottom_blobs() const override { return 2; }
+  int max_bottom_blobs() const override { return 2; }
   void Reshape(const std::vector<Blob*>&, const std::vector<Blob*>&) override;
   void Forward_cpu(const std::vector<Blob*>&,
                    const std::vector<Blob*>&) override;

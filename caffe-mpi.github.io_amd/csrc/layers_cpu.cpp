@@ -1107,6 +1107,9 @@ void AccuracyLayer::Forward_cpu(const std::vector<Blob*>& bottom,
 class InputLayer : public Layer {
  public:
   using Layer::Layer;
+  int min_bottom_blobs() const override { return 0; }
+  int max_bottom_blobs() const override { return 0; }
+  int max_top_blobs() const override { return 4096; }
   void Reshape(const std::vector<Blob*>&,
                const std::vector<Blob*>& top) override {
     auto ip = param_->sub("input_param");
