@@ -201,6 +201,7 @@ void ConvolutionLayer::LayerSetUp(const std::vector<Blob*>& bottom,
   auto cp = param_->sub("convolution_param");
   CHECK_(cp) << "Convolution layer needs convolution_param";
   Cout_ = (int)cp->inum("num_output");
+  CHECK_GT_(Cout_, 0) << "convolution_param.num_output must be positive";
   const int k = (int)cp->inum("kernel_size", 0);
   kh_ = k ? k : (int)cp->inum("kernel_h");
   kw_ = k ? k : (int)cp->inum("kernel_w");
@@ -331,6 +332,7 @@ void InnerProductLayer::LayerSetUp(const std::vector<Blob*>& bottom,
   auto ip = param_->sub("inner_product_param");
   CHECK_(ip);
   Nout_ = (int)ip->inum("num_output");
+  CHECK_GT_(Nout_, 0) << "inner_product_param.num_output must be positive";
   bias_ = ip->boolean("bias_term", true);
   // flatten-from-axis-1, W stored [Nout][K] — any other axis or the
   // transposed-weight layout must fail loudly, not silently mis-multiply
@@ -877,7 +879,16 @@ void ConcatLayer::Reshape(const std::vector<Blob*>& bottom,
         << "only channel concat (concat_dim 1) is implemented";
   }
   int C = 0;
-  for (auto* b : bottom) C += b->channels();
+  for (auto* b : bottom) {
+    // every non-concat axis must match (reference concat_layer.cpp:46-52)
+    CHECK_EQ_(b->num(), bottom[0]->num())
+        << "Concat bottoms disagree on num";
+    CHECK_EQ_(b->height(), bottom[0]->height())
+        << "Concat bottoms disagree on height";
+    CHECK_EQ_(b->width(), bottom[0]->width())
+        << "Concat bottoms disagree on width";
+    C += b->channels();
+  }
   top[0]->Reshape({bottom[0]->num(), C, bottom[0]->height(),
                    bottom[0]->width()});
 }
@@ -1027,6 +1038,9 @@ void SoftmaxWithLossLayer::Reshape(const std::vector<Blob*>& bottom,
   outer_ = bottom[0]->num();
   C_ = bottom[0]->channels();
   inner_ = (int)bottom[0]->count(2);
+  // one label per prediction row (reference softmax_loss_layer.cpp:57)
+  CHECK_EQ_(bottom[1]->count(), (long)outer_ * inner_)
+      << "SoftmaxWithLoss label count must equal outer*inner";
   prob_.ReshapeLike(*bottom[0]);
   top[0]->Reshape({1});
 }
@@ -1069,10 +1083,13 @@ void SoftmaxWithLossLayer::Backward_cpu(const std::vector<Blob*>& top,
 }
 
 // ---------------------------------------------------------------- Accuracy
-void AccuracyLayer::Reshape(const std::vector<Blob*>&,
+void AccuracyLayer::Reshape(const std::vector<Blob*>& bottom,
                             const std::vector<Blob*>& top) {
   auto ap = param_->sub("accuracy_param");
   top_k_ = ap ? (int)ap->inum("top_k", 1) : 1;
+  CHECK_EQ_(bottom[1]->count(),
+            (long)bottom[0]->num() * bottom[0]->count(2))
+      << "Accuracy label count must equal outer*inner";
   top[0]->Reshape({1});
 }
 
