@@ -88,3 +88,43 @@ layer {{ name: "loss" type: "SoftmaxWithLoss" bottom: "fc"
             s2.step(1)
         except Exception:
             pass
+
+
+def test_lmdb_mutations(tmp_path):
+    # corrupted LMDB databases: the mmap B+tree walker must raise or read
+    # garbage-but-bounded, never fault (page offsets are CHECKed against
+    # the map)
+    import subprocess
+    import sys
+
+    from engine_util import REPO
+    sys.path.insert(0, os.path.join(REPO, "tools"))
+    from make_lmdb import make_lmdb
+
+    db = str(tmp_path / "db")
+    make_lmdb(db, 24, 3, 8, 8, 3)
+    raw = open(os.path.join(db, "data.mdb"), "rb").read()
+    rng = np.random.default_rng(23)
+    ca.set_mode("cpu")
+    for t in range(50):
+        b = bytearray(raw)
+        mode = t % 3
+        if mode == 0:
+            for _ in range(int(rng.integers(1, 10))):
+                b[int(rng.integers(0, len(b)))] = int(rng.integers(0, 256))
+        elif mode == 1:
+            b = b[:int(rng.integers(4096, len(b)))]  # keep meta pages
+        else:  # B+tree area flips
+            for _ in range(int(rng.integers(1, 6))):
+                b[int(rng.integers(8192, len(b)))] = int(rng.integers(0, 256))
+        bd = tmp_path / f"bad{t}"
+        bd.mkdir(exist_ok=True)
+        (bd / "data.mdb").write_bytes(bytes(b))
+        try:
+            net = net_from_text(f"""name: "t"
+layer {{ name: "data" type: "Data" top: "data" top: "label"
+  data_param {{ source: "{bd}" batch_size: 4 }} }}
+""")
+            net.forward()
+        except Exception:
+            pass
