@@ -171,3 +171,23 @@ def test_cli_train_missing_solver_flag():
                          timeout=60)
     assert out.returncode == 1
     assert "solver" in out.stderr
+
+
+def test_cli_bad_invocations():
+    # every bad invocation must exit 1 with a message — never crash
+    # (signal) or hang
+    import subprocess
+    cases = [
+        ([], b"usage"),
+        (["bogus"], b"unknown command"),
+        (["train"], b""),                       # no -solver
+        (["train", "-solver=/does/not/exist.prototxt"], b""),
+        (["test", "-model=/does/not/exist.prototxt"], b""),
+        (["time"], b""),                        # no -model
+    ]
+    for args, expect in cases:
+        r = subprocess.run([CAFFE] + args, capture_output=True, timeout=60)
+        assert r.returncode == 1, (args, r.returncode,
+                                   r.stderr.decode()[-200:])
+        if expect:
+            assert expect in r.stderr.lower(), (args, r.stderr[-200:])
