@@ -412,6 +412,12 @@ void PoolingLayer::LayerSetUp(const std::vector<Blob*>&,
   const int p = (int)pp->inum("pad", 0);
   ph_ = pp->has("pad_h") ? (int)pp->inum("pad_h") : p;
   pw_ = pp->has("pad_w") ? (int)pp->inum("pad_w") : p;
+  if (!global_) {
+    // reference pooling_layer.cpp CHECK_LT(pad, kernel): a window fully
+    // inside the padding would MAX over nothing
+    CHECK_LT_(ph_, kh_) << "pooling pad_h must be < kernel_h";
+    CHECK_LT_(pw_, kw_) << "pooling pad_w must be < kernel_w";
+  }
 }
 
 void PoolingLayer::Reshape(const std::vector<Blob*>& bottom,
@@ -425,6 +431,12 @@ void PoolingLayer::Reshape(const std::vector<Blob*>& bottom,
     kw_ = W_;
   }
   pool_out_dim(H_, W_, kh_, kw_, ph_, pw_, sh_, sw_, &OH_, &OW_);
+  CHECK_GT_(OH_, 0) << "pooling '" << name_ << "': kernel " << kh_ << "x"
+                    << kw_ << " larger than padded input " << H_ << "x"
+                    << W_;
+  CHECK_GT_(OW_, 0) << "pooling '" << name_ << "': kernel " << kh_ << "x"
+                    << kw_ << " larger than padded input " << H_ << "x"
+                    << W_;
   top[0]->Reshape({N_, C_, OH_, OW_});
   if (max_) mask_.Reshape({N_, C_, OH_, OW_});
 }
