@@ -15,8 +15,11 @@ TOL = 1e-4
 
 
 def relerr(a, b):
-    a = np.asarray(a, np.float64)
-    b = np.asarray(b, np.float64)
+    # ravel both sides: comparisons are elementwise in memory order, and a
+    # shape mismatch must be an error, never a silent numpy broadcast
+    a = np.asarray(a, np.float64).ravel()
+    b = np.asarray(b, np.float64).ravel()
+    assert a.size == b.size, (a.size, b.size)
     denom = max(np.abs(b).max(), 1e-8)
     return np.abs(a - b).max() / denom
 
