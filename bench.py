@@ -370,7 +370,8 @@ def main():
             "dtype": args.dtype,
             "data": "synthetic",
             "config": {
-                "workload": f"{args.model} fp32 training, per-GPU batch "
+                "workload": f"{args.model} {args.dtype} training, "
+                            f"per-GPU batch "
                             f"{args.batch}, synthetic "
                             f"{'x'.join(map(str, shapes[args.model]))}",
                 "model": args.model,
