@@ -409,8 +409,42 @@ def check_accuracy(mode, N=12, C=10, top_k=3):
                - orc.accuracy(pred, label, N, C, 1, 1)) < 1e-6
 
 
+def check_conv_asym(mode, N=2, C=5, H=14, W=11, Co=6, kh=3, kw=5, sh=2,
+                    sw=1, ph=1, pw=2):
+    # asymmetric kernel/stride/pad (kernel_h != kernel_w): exercises the
+    # separate h/w plumbing through im2col / the GemmView walks — a
+    # swapped-axis bug is invisible to every square-shape test
+    x = rng.standard_normal((N, C, H, W)).astype(np.float32)
+    w = (rng.standard_normal((Co, C, kh, kw)) * 0.2).astype(np.float32)
+    body = f"""layer {{
+  name: "conv"
+  type: "Convolution"
+  bottom: "in0"
+  top: "out"
+  convolution_param {{
+    num_output: {Co}
+    kernel_h: {kh} kernel_w: {kw}
+    stride_h: {sh} stride_w: {sw}
+    pad_h: {ph} pad_w: {pw}
+    bias_term: false
+  }}
+}}"""
+    y_ref = orc.conv_fwd(x, w, None, pad=(ph, pw), stride=(sh, sw))
+    dy = rng.standard_normal(y_ref.shape).astype(np.float32)
+    net, y = run_layer(mode, [(N, C, H, W)], body, [x], params=[w],
+                       top_diff=dy)
+    assert relerr(y, y_ref) < TOL, f"asym fwd {relerr(y, y_ref)}"
+    dx_ref, dw_ref, _ = orc.conv_bwd(x, w, dy, pad=(ph, pw),
+                                     stride=(sh, sw), want_db=False)
+    assert relerr(net.blob("in0", diff=True), dx_ref) < TOL, "asym dx"
+    assert relerr(net.param(0, diff=True), dw_ref) < TOL, "asym dw"
+
+
 ALL_CHECKS = {
     "conv_3x3": lambda m: check_conv(m),
+    "conv_asym": check_conv_asym,
+    "conv_asym_tall": lambda m: check_conv_asym(m, kh=5, kw=1, sh=1, sw=3,
+                                                ph=2, pw=0, H=11, W=16),
     "conv_7x7s2": lambda m: check_conv(m, C=3, H=19, W=19, Co=8, k=7, s=2,
                                        p=3),
     "conv_1x1": lambda m: check_conv(m, k=1, p=0, bias=False),
