@@ -445,6 +445,12 @@ ALL_CHECKS = {
     "conv_asym": check_conv_asym,
     "conv_asym_tall": lambda m: check_conv_asym(m, kh=5, kw=1, sh=1, sw=3,
                                                 ph=2, pw=0, H=11, W=16),
+    # OW >= 24 so the GPU routes these through the IMPLICIT im2col views
+    # (s1 branch / the strided conv1-like branch) with kh != kw
+    "conv_asym_implicit": lambda m: check_conv_asym(
+        m, kh=3, kw=5, sh=1, sw=1, ph=1, pw=2, H=12, W=26, Co=8),
+    "conv_asym_strided_view": lambda m: check_conv_asym(
+        m, kh=3, kw=2, sh=2, sw=1, ph=1, pw=0, H=14, W=26, Co=8),
     "conv_7x7s2": lambda m: check_conv(m, C=3, H=19, W=19, Co=8, k=7, s=2,
                                        p=3),
     "conv_1x1": lambda m: check_conv(m, k=1, p=0, bias=False),
