@@ -237,6 +237,28 @@ layer {
     assert relerr(net.blob("fc", diff=True), dx_ref) < TOL
 
 
+def check_softmaxloss_spatial(mode, N=2, C=5, H=3, W=4):
+    # FCN-style spatial loss: inner = H*W > 1 exercises the inner-stride
+    # walk of the softmax/NLL kernels (softmax_layer.cu's inner_num_ path)
+    x = (rng.standard_normal((N, C, H, W)) * 2).astype(np.float32)
+    lab = rng.integers(0, C, (N, H, W)).astype(np.float32)
+    inner = H * W
+    prob = orc.softmax_fwd(x, N, C, inner)
+    loss_ref = orc.softmaxloss_fwd(prob, lab.ravel(), N, C, inner)
+    body = """layer {
+  name: "loss"
+  type: "SoftmaxWithLoss"
+  bottom: "in0"
+  bottom: "in1"
+  top: "out"
+}"""
+    net, y = run_layer(mode, [(N, C, H, W), (N, H, W)], body, [x, lab])
+    assert abs(y.ravel()[0] - loss_ref) < 1e-5 * max(1, abs(loss_ref))
+    net.backward()
+    dx_ref = orc.softmaxloss_bwd(prob, lab.ravel(), N, C, inner)
+    assert relerr(net.blob("in0", diff=True), dx_ref) < TOL
+
+
 def check_eltwise_concat(mode):
     N, C, H, W = 2, 4, 5, 5
     x = rng.standard_normal((N, C, H, W)).astype(np.float32)
@@ -505,6 +527,7 @@ ALL_CHECKS = {
     "lrn_smallc": lambda m: check_lrn(m, C=3),
     "lrn_size3": lambda m: check_lrn(m, size=3),
     "softmaxloss": check_softmaxloss,
+    "softmaxloss_spatial": check_softmaxloss_spatial,
     "eltwise_concat": check_eltwise_concat,
     "dropout": check_dropout,
     "bn_global_stats": check_bn_global,
